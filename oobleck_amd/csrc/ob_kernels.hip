@@ -1,0 +1,776 @@
+// ob_kernels.hip — hand-written CDNA4 (gfx950) HIP kernels for the Oobleck
+// hot path, fp32.  No CUDA shims, no hipify: written for 64-wide wavefronts,
+// MFMA f32 (v_mfma_f32_32x32x2_f32, the exact-f32 matrix path at the 157 TF
+// f32 rate), LDS-staged tiles, coalesced HBM access.
+//
+// Operator content restated from the reference's layer math (HF GPT-2 eager
+// ops the reference fx-splits; see oracle/gpt2_oracle.py for the per-op
+// citations).  This file is the compute substrate the reference runs as
+// torch eager ops inside fx GraphModules
+// (/root/reference/oobleck/execution/layer.py:144-145).
+#include "ob_internal.h"
+
+#include <cmath>
+
+static thread_local char g_err[1024] = {0};
+
+int ob_fail(const char* fmt, ...) {
+  va_list ap;
+  va_start(ap, fmt);
+  vsnprintf(g_err, sizeof(g_err), fmt, ap);
+  va_end(ap);
+  return 1;
+}
+
+extern "C" const char* ob_last_error(void) { return g_err; }
+extern "C" const char* ob_build_arch(void) {
+#if defined(__HIP_DEVICE_COMPILE__)
+  return "device";
+#else
+  return "gfx950";
+#endif
+}
+
+using f32x16 = __attribute__((ext_vector_type(16))) float;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+static inline hipStream_t S(void* s) { return reinterpret_cast<hipStream_t>(s); }
+
+// ---------------------------------------------------------------------------
+// block reduction helpers (wave64 shfl + LDS across the block's waves)
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ int64_t imin64d(int64_t a, int64_t b) {
+  return a < b ? a : b;
+}
+
+__device__ __forceinline__ float wave_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+  return v;  // valid in lane 0
+}
+
+__device__ __forceinline__ float wave_max(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_down(v, off, 64));
+  return v;
+}
+
+// block of 256 threads = 4 waves; result broadcast to all threads.
+__device__ __forceinline__ float block_sum256(float v, float* lds4) {
+  v = wave_sum(v);
+  const int wid = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) lds4[wid] = v;
+  __syncthreads();
+  const float r = lds4[0] + lds4[1] + lds4[2] + lds4[3];
+  __syncthreads();
+  return r;
+}
+
+__device__ __forceinline__ float block_max256(float v, float* lds4) {
+  v = wave_max(v);
+  const int wid = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) lds4[wid] = v;
+  __syncthreads();
+  const float r = fmaxf(fmaxf(lds4[0], lds4[1]), fmaxf(lds4[2], lds4[3]));
+  __syncthreads();
+  return r;
+}
+
+// ---------------------------------------------------------------------------
+// GEMM: C[M,N] (+)= alpha * op(A) @ op(B) (+ bias[n]) (+ R), fp32 MFMA.
+//
+// 128x128 block tile, BK=32, 4 waves; each wave computes a 64x64 sub-tile as
+// 2x2 fragments of v_mfma_f32_32x32x2_f32 (A: lane l holds A[i=l&31][k=l>>5],
+// B[k=l>>5][j=l&31]; C/D: col=l&31, row=(r&3)+8*(r>>2)+4*(l>>5)).
+// LDS tiles are staged k-major As[k][m], Bs[k][n] with a +4-float row pad so
+// both the transpose-staging writes and the fragment reads are conflict-free
+// (bank = (addr/4)%32).
+// ---------------------------------------------------------------------------
+
+#define GEMM_BM 128
+#define GEMM_BN 128
+#define GEMM_BK 32
+#define GEMM_LDT 132  // padded LDS row stride (floats)
+
+// stage dst[k][c] = src[k][c] (src already offset; ld = src row stride).
+// kmax/cmax = valid extents (zero-fill outside).  256 threads.
+__device__ __forceinline__ void stage_direct(float* dst, const float* src,
+                                             int64_t ld, int kmax, int cmax) {
+  const int c4 = (threadIdx.x & 31) * 4;  // 0..124
+  const int k0 = threadIdx.x >> 5;        // 0..7
+#pragma unroll
+  for (int it = 0; it < 4; ++it) {
+    const int k = k0 + it * 8;
+    float4 v = {0.f, 0.f, 0.f, 0.f};
+    if (k < kmax) {
+      if (c4 + 3 < cmax) {
+        v = *reinterpret_cast<const float4*>(src + (int64_t)k * ld + c4);
+      } else {
+        if (c4 + 0 < cmax) v.x = src[(int64_t)k * ld + c4 + 0];
+        if (c4 + 1 < cmax) v.y = src[(int64_t)k * ld + c4 + 1];
+        if (c4 + 2 < cmax) v.z = src[(int64_t)k * ld + c4 + 2];
+        if (c4 + 3 < cmax) v.w = src[(int64_t)k * ld + c4 + 3];
+      }
+    }
+    *reinterpret_cast<float4*>(dst + k * GEMM_LDT + c4) = v;
+  }
+}
+
+// stage dst[k][c] = src[c][k] (transpose; src offset to (c0,k0); ld = src
+// row stride).  Reads float4 along k (coalesced 128B per 8 lanes), writes
+// columns; the +4 row pad makes the column writes conflict-free.
+__device__ __forceinline__ void stage_transpose(float* dst, const float* src,
+                                                int64_t ld, int kmax, int cmax) {
+  const int k4 = (threadIdx.x & 7) * 4;  // 0..28
+  const int c0 = threadIdx.x >> 3;       // 0..31
+#pragma unroll
+  for (int it = 0; it < 4; ++it) {
+    const int c = c0 + it * 32;
+    float4 v = {0.f, 0.f, 0.f, 0.f};
+    if (c < cmax) {
+      if (k4 + 3 < kmax) {
+        v = *reinterpret_cast<const float4*>(src + (int64_t)c * ld + k4);
+      } else {
+        if (k4 + 0 < kmax) v.x = src[(int64_t)c * ld + k4 + 0];
+        if (k4 + 1 < kmax) v.y = src[(int64_t)c * ld + k4 + 1];
+        if (k4 + 2 < kmax) v.z = src[(int64_t)c * ld + k4 + 2];
+        if (k4 + 3 < kmax) v.w = src[(int64_t)c * ld + k4 + 3];
+      }
+    }
+    dst[(k4 + 0) * GEMM_LDT + c] = v.x;
+    dst[(k4 + 1) * GEMM_LDT + c] = v.y;
+    dst[(k4 + 2) * GEMM_LDT + c] = v.z;
+    dst[(k4 + 3) * GEMM_LDT + c] = v.w;
+  }
+}
+
+template <bool TA, bool TB, bool ATOMIC>
+__global__ __launch_bounds__(256, 4) void k_gemm_f32(
+    const float* __restrict__ A, const float* __restrict__ B,
+    float* __restrict__ C, const float* __restrict__ bias,
+    const float* __restrict__ R, int M, int N, int K, int64_t lda, int64_t ldb,
+    int64_t ldc, int64_t sA1, int64_t sA2, int64_t sB1, int64_t sB2,
+    int64_t sC1, int64_t sC2, int n2, float alpha, float beta, int nbn) {
+  __shared__ float As[GEMM_BK * GEMM_LDT];
+  __shared__ float Bs[GEMM_BK * GEMM_LDT];
+
+  const int tile = blockIdx.x;
+  const int bm = tile / nbn, bn = tile % nbn;
+  const int m0 = bm * GEMM_BM, n0 = bn * GEMM_BN;
+
+  const int z = blockIdx.z;
+  const int i1 = z / n2, i2 = z % n2;
+  A += (int64_t)i1 * sA1 + (int64_t)i2 * sA2;
+  B += (int64_t)i1 * sB1 + (int64_t)i2 * sB2;
+  C += (int64_t)i1 * sC1 + (int64_t)i2 * sC2;
+  if (R) R += (int64_t)i1 * sC1 + (int64_t)i2 * sC2;
+
+  // split-K slice (blockIdx.y): [k0, kend)
+  const int splitk = gridDim.y;
+  const int kchunk = ((K + splitk * GEMM_BK - 1) / (splitk * GEMM_BK)) * GEMM_BK;
+  const int kbeg = blockIdx.y * kchunk;
+  const int kend = min(K, kbeg + kchunk);
+  if (kbeg >= kend) return;
+
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int wr = w >> 1, wc = w & 1;
+  const int il = lane & 31, kh = lane >> 5;
+
+  f32x16 acc00 = {}, acc01 = {}, acc10 = {}, acc11 = {};
+
+  for (int kt = kbeg; kt < kend; kt += GEMM_BK) {
+    const int kmax = kend - kt;
+    if (TA)
+      stage_direct(As, A + (int64_t)kt * lda + m0, lda, min(kmax, GEMM_BK),
+                   min(M - m0, GEMM_BM));
+    else
+      stage_transpose(As, A + (int64_t)m0 * lda + kt, lda, min(kmax, GEMM_BK),
+                      min(M - m0, GEMM_BM));
+    if (TB)
+      stage_transpose(Bs, B + (int64_t)n0 * ldb + kt, ldb, min(kmax, GEMM_BK),
+                      min(N - n0, GEMM_BN));
+    else
+      stage_direct(Bs, B + (int64_t)kt * ldb + n0, ldb, min(kmax, GEMM_BK),
+                   min(N - n0, GEMM_BN));
+    __syncthreads();
+
+#pragma unroll
+    for (int kk = 0; kk < GEMM_BK / 2; ++kk) {
+      const float* ar = As + (kk * 2 + kh) * GEMM_LDT + wr * 64;
+      const float* br = Bs + (kk * 2 + kh) * GEMM_LDT + wc * 64;
+      const float a0 = ar[il], a1 = ar[32 + il];
+      const float b0 = br[il], b1 = br[32 + il];
+      acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
+      acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
+      acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
+      acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue
+  const int mw = m0 + wr * 64, nw = n0 + wc * 64;
+#define OB_EPI(ACC, TI, TJ)                                                   \
+  {                                                                           \
+    const int nn = nw + (TJ)*32 + il;                                         \
+    if (nn < N) {                                                             \
+      _Pragma("unroll") for (int r = 0; r < 16; ++r) {                        \
+        const int mm = mw + (TI)*32 + (r & 3) + 8 * (r >> 2) + 4 * kh;        \
+        if (mm < M) {                                                         \
+          float v = alpha * ACC[r];                                           \
+          if (ATOMIC) {                                                       \
+            atomicAdd(&C[(int64_t)mm * ldc + nn], v);                         \
+          } else {                                                            \
+            if (bias) v += bias[nn];                                          \
+            if (R) v += R[(int64_t)mm * ldc + nn];                            \
+            if (beta != 0.f) v += beta * C[(int64_t)mm * ldc + nn];           \
+            C[(int64_t)mm * ldc + nn] = v;                                    \
+          }                                                                   \
+        }                                                                     \
+      }                                                                       \
+    }                                                                         \
+  }
+  OB_EPI(acc00, 0, 0)
+  OB_EPI(acc01, 0, 1)
+  OB_EPI(acc10, 1, 0)
+  OB_EPI(acc11, 1, 1)
+#undef OB_EPI
+}
+
+extern "C" int ob_gemm_f32(int transA, int transB, int64_t M, int64_t N,
+                           int64_t K, float alpha, const void* A, int64_t lda,
+                           int64_t strideA1, int64_t strideA2, const void* B,
+                           int64_t ldb, int64_t strideB1, int64_t strideB2,
+                           float beta, void* C, int64_t ldc, int64_t strideC1,
+                           int64_t strideC2, int64_t n1, int64_t n2,
+                           const void* bias, const void* residual, int atomic,
+                           int splitk, void* stream) {
+  if (M <= 0 || N <= 0 || K <= 0) return ob_fail("gemm: bad dims");
+  if (splitk < 1) splitk = 1;
+  if (splitk > 1 && !atomic)
+    return ob_fail("gemm: splitk>1 requires atomic stores");
+  if (atomic && (bias || residual))
+    return ob_fail("gemm: atomic epilogue excludes bias/residual");
+  const int nbm = (int)((M + GEMM_BM - 1) / GEMM_BM);
+  const int nbn = (int)((N + GEMM_BN - 1) / GEMM_BN);
+  dim3 grid(nbm * nbn, splitk, (unsigned)(n1 * n2));
+  dim3 block(256);
+#define OB_GEMM_LAUNCH(TA_, TB_, AT_)                                       \
+  k_gemm_f32<TA_, TB_, AT_><<<grid, block, 0, S(stream)>>>(                 \
+      (const float*)A, (const float*)B, (float*)C, (const float*)bias,      \
+      (const float*)residual, (int)M, (int)N, (int)K, lda, ldb, ldc,        \
+      strideA1, strideA2, strideB1, strideB2, strideC1, strideC2, (int)n2,  \
+      alpha, beta, nbn)
+  const int sel = (transA ? 4 : 0) | (transB ? 2 : 0) | (atomic ? 1 : 0);
+  switch (sel) {
+    case 0: OB_GEMM_LAUNCH(false, false, false); break;
+    case 1: OB_GEMM_LAUNCH(false, false, true); break;
+    case 2: OB_GEMM_LAUNCH(false, true, false); break;
+    case 3: OB_GEMM_LAUNCH(false, true, true); break;
+    case 4: OB_GEMM_LAUNCH(true, false, false); break;
+    case 5: OB_GEMM_LAUNCH(true, false, true); break;
+    case 6: OB_GEMM_LAUNCH(true, true, false); break;
+    case 7: OB_GEMM_LAUNCH(true, true, true); break;
+  }
+#undef OB_GEMM_LAUNCH
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+// ---------------------------------------------------------------------------
+// LayerNorm forward: one 256-thread block per row.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void k_ln_fwd(const float* __restrict__ x,
+                                                const float* __restrict__ w,
+                                                const float* __restrict__ b,
+                                                float* __restrict__ y,
+                                                float* __restrict__ mean,
+                                                float* __restrict__ rstd,
+                                                int64_t rows, int H, float eps) {
+  __shared__ float lds4[4];
+  for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
+    const float* xr = x + row * H;
+    float s = 0.f, sq = 0.f;
+    for (int c = threadIdx.x; c < H; c += 256) {
+      const float v = xr[c];
+      s += v;
+      sq += v * v;
+    }
+    const float mu = block_sum256(s, lds4) / H;
+    const float var = block_sum256(sq, lds4) / H - mu * mu;
+    const float rs = rsqrtf(var + eps);
+    if (threadIdx.x == 0) {
+      mean[row] = mu;
+      rstd[row] = rs;
+    }
+    float* yr = y + row * H;
+    for (int c = threadIdx.x; c < H; c += 256)
+      yr[c] = (xr[c] - mu) * rs * w[c] + b[c];
+    __syncthreads();
+  }
+}
+
+static inline int64_t imin64(int64_t a, int64_t b) { return a < b ? a : b; }
+
+extern "C" int ob_layernorm_fwd_f32(const void* x, const void* w, const void* b,
+                                    void* y, void* mean, void* rstd,
+                                    int64_t rows, int64_t H, float eps,
+                                    void* stream) {
+  const int grid = (int)imin64(rows, 16384);
+  k_ln_fwd<<<grid, 256, 0, S(stream)>>>((const float*)x, (const float*)w,
+                                        (const float*)b, (float*)y,
+                                        (float*)mean, (float*)rstd, rows,
+                                        (int)H, eps);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+// LayerNorm backward.  CHUNK rows per block; dw/db accumulated in registers
+// per thread-owned columns, one atomicAdd per column per block.
+// dx_i (+)= rstd*( dy_i*w_i − mean_c(dy*w) − xhat_i * mean_c(dy*w*xhat) )
+#define LN_CHUNK 16
+
+template <bool DX_ACCUM>
+__global__ __launch_bounds__(256) void k_ln_bwd(
+    const float* __restrict__ x, const float* __restrict__ w,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const float* __restrict__ dy, float* __restrict__ dx,
+    float* __restrict__ dw, float* __restrict__ db, int64_t rows, int H) {
+  __shared__ float lds4[4];
+  const int64_t r0 = (int64_t)blockIdx.x * LN_CHUNK;
+  // register accumulators for up to 8 columns per thread (H <= 2048 here;
+  // GPT-2 H is 768/1600 — LN is only applied across n_embd).  Fixed-bound
+  // unrolled loops keep the arrays in registers (hipcc sends runtime-indexed
+  // local arrays to scratch).
+  float accw[8] = {0}, accb[8] = {0};
+  const int64_t rend = imin64d(rows, r0 + LN_CHUNK);
+  for (int64_t row = r0; row < rend; ++row) {
+    const float* xr = x + row * H;
+    const float* dyr = dy + row * H;
+    const float mu = mean[row], rs = rstd[row];
+    float s1 = 0.f, s2 = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = threadIdx.x + j * 256;
+      if (c < H) {
+        const float xhat = (xr[c] - mu) * rs;
+        const float dyw = dyr[c] * w[c];
+        s1 += dyw * xhat;
+        s2 += dyw;
+        accw[j] += dyr[c] * xhat;
+        accb[j] += dyr[c];
+      }
+    }
+    const float m1 = block_sum256(s1, lds4) / H;
+    const float m2 = block_sum256(s2, lds4) / H;
+    float* dxr = dx + row * H;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = threadIdx.x + j * 256;
+      if (c < H) {
+        const float xhat = (xr[c] - mu) * rs;
+        const float v = rs * (dyr[c] * w[c] - m2 - xhat * m1);
+        if (DX_ACCUM)
+          dxr[c] += v;
+        else
+          dxr[c] = v;
+      }
+    }
+    __syncthreads();
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int c = threadIdx.x + j * 256;
+    if (c < H) {
+      atomicAdd(&dw[c], accw[j]);
+      atomicAdd(&db[c], accb[j]);
+    }
+  }
+}
+
+extern "C" int ob_layernorm_bwd_f32(const void* x, const void* w,
+                                    const void* mean, const void* rstd,
+                                    const void* dy, void* dx, void* dw,
+                                    void* db, int64_t rows, int64_t H,
+                                    int dx_accum, void* stream) {
+  if (H > 2048) return ob_fail("ln_bwd: H > 2048 unsupported");
+  const int grid = (int)((rows + LN_CHUNK - 1) / LN_CHUNK);
+  if (dx_accum)
+    k_ln_bwd<true><<<grid, 256, 0, S(stream)>>>(
+        (const float*)x, (const float*)w, (const float*)mean,
+        (const float*)rstd, (const float*)dy, (float*)dx, (float*)dw,
+        (float*)db, rows, (int)H);
+  else
+    k_ln_bwd<false><<<grid, 256, 0, S(stream)>>>(
+        (const float*)x, (const float*)w, (const float*)mean,
+        (const float*)rstd, (const float*)dy, (float*)dx, (float*)dw,
+        (float*)db, rows, (int)H);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+// ---------------------------------------------------------------------------
+// Causal softmax (fp32 scores in place).  Row r of the S×S score matrix
+// keeps columns 0..r; masked entries become exactly 0 (matching exp of the
+// finfo.min-masked eager path, which underflows to 0 in fp32).
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void k_softmax_causal_fwd(
+    float* __restrict__ scores, int Sq, float scale) {
+  __shared__ float lds4[4];
+  const int64_t rid = blockIdx.x;        // z * S + row
+  const int row = (int)(rid % Sq);
+  float* p = scores + rid * Sq;
+  const int valid = row + 1;
+  // registers: up to 8 elements per thread (S <= 2048); fixed-bound
+  // unrolled loops keep v[] in registers.
+  float v[8];
+  float mx = -INFINITY;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int c = threadIdx.x + j * 256;
+    v[j] = (c < valid) ? p[c] * scale : -INFINITY;
+    mx = fmaxf(mx, v[j]);
+  }
+  mx = block_max256(mx, lds4);
+  float sum = 0.f;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    v[j] = (v[j] == -INFINITY) ? 0.f : __expf(v[j] - mx);
+    sum += v[j];
+  }
+  sum = block_sum256(sum, lds4);
+  const float inv = 1.f / sum;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int c = threadIdx.x + j * 256;
+    if (c < Sq) p[c] = v[j] * inv;
+  }
+}
+
+extern "C" int ob_softmax_causal_fwd_f32(void* scores, int64_t batch, int64_t Sq,
+                                         float scale, void* stream) {
+  if (Sq > 2048) return ob_fail("softmax: S > 2048 unsupported");
+  k_softmax_causal_fwd<<<(unsigned)(batch * Sq), 256, 0, S(stream)>>>(
+      (float*)scores, (int)Sq, scale);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+// dS = P * (dP − sum_c dP*P), in place on dP.
+__global__ __launch_bounds__(256) void k_softmax_causal_bwd(
+    const float* __restrict__ P, float* __restrict__ dP, int Sq) {
+  __shared__ float lds4[4];
+  const int64_t rid = blockIdx.x;
+  const int row = (int)(rid % Sq);
+  const float* pr = P + rid * Sq;
+  float* dr = dP + rid * Sq;
+  const int valid = row + 1;
+  float pv[8], dv[8];
+  float t = 0.f;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int c = threadIdx.x + j * 256;
+    pv[j] = (c < valid) ? pr[c] : 0.f;
+    dv[j] = (c < valid) ? dr[c] : 0.f;
+    t += pv[j] * dv[j];
+  }
+  t = block_sum256(t, lds4);
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int c = threadIdx.x + j * 256;
+    if (c < Sq) dr[c] = pv[j] * (dv[j] - t);
+  }
+}
+
+extern "C" int ob_softmax_causal_bwd_f32(const void* P, void* dP, int64_t batch,
+                                         int64_t Sq, void* stream) {
+  if (Sq > 2048) return ob_fail("softmax_bwd: S > 2048 unsupported");
+  k_softmax_causal_bwd<<<(unsigned)(batch * Sq), 256, 0, S(stream)>>>(
+      (const float*)P, (float*)dP, (int)Sq);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+// ---------------------------------------------------------------------------
+// gelu_new (transformers NewGELUActivation):
+//   g = 0.5*u*(1 + tanh(k*(u + 0.044715*u^3))), k = sqrt(2/pi)
+//   dg/du = 0.5*(1+tanh(t)) + 0.5*u*(1-tanh(t)^2)*k*(1 + 3*0.044715*u^2)
+// ---------------------------------------------------------------------------
+
+#define GELU_K 0.7978845608028654f
+#define GELU_C 0.044715f
+
+__global__ __launch_bounds__(256) void k_gelu_fwd(const float* __restrict__ u,
+                                                  float* __restrict__ g,
+                                                  int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * 256) {
+    const float x = u[i];
+    const float t = tanhf(GELU_K * (x + GELU_C * x * x * x));
+    g[i] = 0.5f * x * (1.f + t);
+  }
+}
+
+__global__ __launch_bounds__(256) void k_gelu_bwd(const float* __restrict__ u,
+                                                  const float* __restrict__ dg,
+                                                  float* __restrict__ du,
+                                                  int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * 256) {
+    const float x = u[i];
+    const float t = tanhf(GELU_K * (x + GELU_C * x * x * x));
+    const float d = 0.5f * (1.f + t) +
+                    0.5f * x * (1.f - t * t) * GELU_K * (1.f + 3.f * GELU_C * x * x);
+    du[i] = dg[i] * d;
+  }
+}
+
+extern "C" int ob_gelu_fwd_f32(const void* u, void* g, int64_t n, void* stream) {
+  const int grid = (int)imin64((n + 255) / 256, 2048);
+  k_gelu_fwd<<<grid, 256, 0, S(stream)>>>((const float*)u, (float*)g, n);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+extern "C" int ob_gelu_bwd_f32(const void* u, const void* dg, void* du,
+                               int64_t n, void* stream) {
+  const int grid = (int)imin64((n + 255) / 256, 2048);
+  k_gelu_bwd<<<grid, 256, 0, S(stream)>>>((const float*)u, (const float*)dg,
+                                          (float*)du, n);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+// ---------------------------------------------------------------------------
+// column sum: db[n] += sum_m X[m,n]
+// ---------------------------------------------------------------------------
+
+#define CS_ROWS 256
+
+__global__ __launch_bounds__(256) void k_colsum(const float* __restrict__ X,
+                                                float* __restrict__ db,
+                                                int64_t M, int64_t N) {
+  const int64_t c = (int64_t)blockIdx.x * 256 + threadIdx.x;
+  if (c >= N) return;
+  const int64_t r0 = (int64_t)blockIdx.y * CS_ROWS;
+  const int64_t r1 = min(M, r0 + CS_ROWS);
+  float acc = 0.f;
+  for (int64_t r = r0; r < r1; ++r) acc += X[r * N + c];
+  atomicAdd(&db[c], acc);
+}
+
+extern "C" int ob_colsum_f32(const void* X, void* db, int64_t M, int64_t N,
+                             void* stream) {
+  dim3 grid((unsigned)((N + 255) / 256), (unsigned)((M + CS_ROWS - 1) / CS_ROWS));
+  k_colsum<<<grid, 256, 0, S(stream)>>>((const float*)X, (float*)db, M, N);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+// ---------------------------------------------------------------------------
+// Embedding: out[b,s,:] = wte[ids[b,s]] + wpe[s]   (model.py wte+wpe sum)
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void k_embed_fwd(const int64_t* __restrict__ ids,
+                                                   const float* __restrict__ wte,
+                                                   const float* __restrict__ wpe,
+                                                   float* __restrict__ out,
+                                                   int64_t BS, int Sq, int H) {
+  for (int64_t row = blockIdx.x; row < BS; row += gridDim.x) {
+    const int64_t id = ids[row];
+    const int s = (int)(row % Sq);
+    const float* te = wte + id * H;
+    const float* pe = wpe + (int64_t)s * H;
+    float* o = out + row * H;
+    for (int c = threadIdx.x; c < H; c += 256) o[c] = te[c] + pe[c];
+  }
+}
+
+int ob_embed_fwd_f32(const int64_t* ids, const float* wte, const float* wpe,
+                     float* out, int64_t B, int64_t Sq, int64_t H, void* stream) {
+  const int grid = (int)imin64(B * Sq, 16384);
+  k_embed_fwd<<<grid, 256, 0, S(stream)>>>(ids, wte, wpe, out, B * Sq, (int)Sq,
+                                           (int)H);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+__global__ __launch_bounds__(256) void k_embed_bwd(const int64_t* __restrict__ ids,
+                                                   const float* __restrict__ dout,
+                                                   float* __restrict__ dwte,
+                                                   float* __restrict__ dwpe,
+                                                   int64_t BS, int Sq, int H) {
+  for (int64_t row = blockIdx.x; row < BS; row += gridDim.x) {
+    const int64_t id = ids[row];
+    const int s = (int)(row % Sq);
+    const float* d = dout + row * H;
+    for (int c = threadIdx.x; c < H; c += 256) {
+      atomicAdd(&dwte[id * H + c], d[c]);
+      atomicAdd(&dwpe[(int64_t)s * H + c], d[c]);
+    }
+  }
+}
+
+int ob_embed_bwd_f32(const int64_t* ids, const float* dout, float* dwte,
+                     float* dwpe, int64_t B, int64_t Sq, int64_t H,
+                     void* stream) {
+  const int grid = (int)imin64(B * Sq, 16384);
+  k_embed_bwd<<<grid, 256, 0, S(stream)>>>(ids, dout, dwte, dwpe, B * Sq,
+                                           (int)Sq, (int)H);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+// ---------------------------------------------------------------------------
+// Cross-entropy with the GPT-2 shift (modeling_gpt2: loss over
+// logits[:, :-1] vs labels[:, 1:], mean).  One block per row; single online
+// (max,sum) pass over V; loss accumulated with one atomicAdd per valid row.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void k_ce_fwd(const float* __restrict__ logits,
+                                                const int64_t* __restrict__ labels,
+                                                float* __restrict__ lse,
+                                                float* __restrict__ loss,
+                                                int64_t BS, int Sq, int V) {
+  __shared__ float lmax[4], lsum[4];
+  for (int64_t row = blockIdx.x; row < BS; row += gridDim.x) {
+    const float* lr = logits + row * V;
+    float m = -INFINITY, s = 0.f;
+    for (int c = threadIdx.x; c < V; c += 256) {
+      const float x = lr[c];
+      if (x > m) {
+        s = s * __expf(m - x) + 1.f;
+        m = x;
+      } else {
+        s += __expf(x - m);
+      }
+    }
+    // combine across the wave then the block: (m,s) pairs.  A lane/wave that
+    // processed no elements carries (m=-inf, s=0); guard the exp so
+    // exp(-inf - -inf) never produces NaN.
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      const float om = __shfl_down(m, off, 64);
+      const float os = __shfl_down(s, off, 64);
+      const float nm = fmaxf(m, om);
+      const float t1 = (m > -INFINITY) ? s * __expf(m - nm) : 0.f;
+      const float t2 = (om > -INFINITY) ? os * __expf(om - nm) : 0.f;
+      s = t1 + t2;
+      m = nm;
+    }
+    const int wid = threadIdx.x >> 6;
+    if ((threadIdx.x & 63) == 0) {
+      lmax[wid] = m;
+      lsum[wid] = s;
+    }
+    __syncthreads();
+    const float bm = fmaxf(fmaxf(lmax[0], lmax[1]), fmaxf(lmax[2], lmax[3]));
+    float bs = 0.f;
+    #pragma unroll
+    for (int wi = 0; wi < 4; ++wi)
+      bs += (lmax[wi] > -INFINITY) ? lsum[wi] * __expf(lmax[wi] - bm) : 0.f;
+    const float l = bm + __logf(bs);
+    const int s_pos = (int)(row % Sq);
+    if (threadIdx.x == 0) {
+      lse[row] = l;
+      if (s_pos < Sq - 1) {
+        const int64_t lab = labels[row + 1];  // labels[b, s+1]
+        const int64_t B = BS / Sq;
+        const float inv = 1.f / (float)(B * (Sq - 1));
+        atomicAdd(loss, (l - lr[lab]) * inv);
+      }
+    }
+    __syncthreads();
+  }
+}
+
+int ob_ce_fwd_f32(const float* logits, const int64_t* labels, float* lse,
+                  float* loss, int64_t B, int64_t Sq, int64_t V, void* stream) {
+  const int grid = (int)imin64(B * Sq, 16384);
+  k_ce_fwd<<<grid, 256, 0, S(stream)>>>(logits, labels, lse, loss, B * Sq,
+                                        (int)Sq, (int)V);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+// dlogits[m,v] = dloss/(B*(S-1)) * (softmax - onehot(label[m+1]))  for rows
+// with s < S-1, else 0.  In place on the logits stash.
+__global__ __launch_bounds__(256) void k_ce_bwd(float* __restrict__ logits,
+                                                const int64_t* __restrict__ labels,
+                                                const float* __restrict__ lse,
+                                                const float* __restrict__ dloss,
+                                                int64_t BS, int Sq, int V) {
+  const float dl = dloss ? *dloss : 1.f;
+  const int64_t B = BS / Sq;
+  const float scale = dl / (float)(B * (Sq - 1));
+  for (int64_t row = blockIdx.y; row < BS; row += gridDim.y) {
+    float* lr = logits + row * V;
+    const int s_pos = (int)(row % Sq);
+    const float l = lse[row];
+    const bool valid = s_pos < Sq - 1;
+    const int64_t lab = valid ? labels[row + 1] : -1;
+    for (int64_t c = (int64_t)blockIdx.x * 256 + threadIdx.x; c < V;
+         c += (int64_t)gridDim.x * 256) {
+      float g = 0.f;
+      if (valid) {
+        g = scale * (__expf(lr[c] - l) - (c == lab ? 1.f : 0.f));
+      }
+      lr[c] = g;
+    }
+  }
+}
+
+int ob_ce_bwd_f32(float* logits, const int64_t* labels, const float* lse,
+                  const float* dloss, int64_t B, int64_t Sq, int64_t V,
+                  void* stream) {
+  dim3 grid((unsigned)imin64((V + 255) / 256, 256),
+            (unsigned)imin64(B * Sq, 16384));
+  k_ce_bwd<<<grid, 256, 0, S(stream)>>>(logits, labels, lse, dloss, B * Sq,
+                                        (int)Sq, (int)V);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+// ---------------------------------------------------------------------------
+// Fused AdamW over a flat buffer: 7 words/param traffic (r: p,g,m,v;
+// w: p,m,v).  Matches torch.optim.AdamW (decoupled wd, bias correction,
+// eps added after sqrt(v)/sqrt(bc2)).
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void k_adamw(float* __restrict__ p,
+                                               const float* __restrict__ g,
+                                               float* __restrict__ m,
+                                               float* __restrict__ v, int64_t n,
+                                               float wd_factor, float step_size,
+                                               float inv_sqrt_bc2, float beta1,
+                                               float beta2, float eps) {
+  for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * 256) {
+    const float gi = g[i];
+    float mi = beta1 * m[i] + (1.f - beta1) * gi;
+    float vi = beta2 * v[i] + (1.f - beta2) * gi * gi;
+    m[i] = mi;
+    v[i] = vi;
+    const float denom = sqrtf(vi) * inv_sqrt_bc2 + eps;
+    p[i] = p[i] * wd_factor - step_size * mi / denom;
+  }
+}
+
+extern "C" int ob_adamw_step(void* p, const void* g, void* m, void* v,
+                             int64_t n, int32_t step, float lr, float beta1,
+                             float beta2, float eps, float weight_decay,
+                             void* stream) {
+  if (step < 1) return ob_fail("adamw: step must be >= 1");
+  const double bc1 = 1.0 - pow((double)beta1, (double)step);
+  const double bc2 = 1.0 - pow((double)beta2, (double)step);
+  const int grid = (int)imin64((n + 255) / 256, 4096);
+  k_adamw<<<grid, 256, 0, S(stream)>>>(
+      (float*)p, (const float*)g, (float*)m, (float*)v, n,
+      1.f - lr * weight_decay, (float)(lr / bc1), (float)(1.0 / sqrt(bc2)),
+      beta1, beta2, eps);
+  OB_LAUNCH_CHECK();
+  return 0;
+}

@@ -1,0 +1,494 @@
+// ob_layer.hip — C-ABI layer objects for the Oobleck hot path on MI355X.
+//
+// One ob_layer = one fx-shard of the reference's model
+// (/root/reference/oobleck/module/sharding.py:12-47): embedding, one
+// transformer block, or ln_f+lm_head+loss.  Forward/backward orchestrate the
+// hand-written kernels of ob_kernels.hip on the caller's hipStream_t,
+// mirroring the semantics of /root/reference/oobleck/execution/layer.py
+// (forward :144-145, backward :250-260 — grads ACCUMULATE into the bound
+// flat grad buffer) and pipeline.py:169-244 (loss on the last layer).
+//
+// The activation stash is slot-indexed so several microbatches can be in
+// flight, matching deepspeed-style pipe buffers (pipeline.py:556-562).
+#include "ob_internal.h"
+
+#include <cmath>
+#include <cstring>
+
+static inline hipStream_t S(void* s) { return reinterpret_cast<hipStream_t>(s); }
+static inline int64_t i64min(int64_t a, int64_t b) { return a < b ? a : b; }
+static inline int64_t i64max(int64_t a, int64_t b) { return a > b ? a : b; }
+
+// ---------------------------------------------------------------------------
+// shared backward workspace (single driving thread per GPU by ABI contract)
+// ---------------------------------------------------------------------------
+
+struct Workspace {
+  float* dp = nullptr;     // [B*nh, S, S]
+  float* bsh1 = nullptr;   // [B, S, H]
+  float* bsh2 = nullptr;   // [B, S, H]
+  float* dqkv = nullptr;   // [B, S, 3H]
+  float* b4h = nullptr;    // [B, S, 4H]
+  int64_t sz_dp = 0, sz_bsh = 0, sz_bsh2 = 0, sz_dqkv = 0, sz_b4h = 0;
+};
+static Workspace g_ws;
+
+static int ws_ensure(float** buf, int64_t* cur, int64_t need) {
+  if (need <= *cur) return 0;
+  if (*buf) OB_HIP(hipFree(*buf));
+  *buf = nullptr;
+  *cur = 0;
+  OB_HIP(hipMalloc(buf, need * sizeof(float)));
+  *cur = need;
+  return 0;
+}
+
+// ---------------------------------------------------------------------------
+// layer object
+// ---------------------------------------------------------------------------
+
+struct ob_layer {
+  ob_layer_desc d;
+  int B;                    // current microbatch size (<= max_batch)
+  float* params = nullptr;  // caller-owned flat fp32
+  float* grads = nullptr;
+  // stash (extension-owned), one contiguous allocation, slot-strided
+  float* stash = nullptr;
+  int64_t slot_stride = 0;  // floats per slot
+  int64_t* ids = nullptr;   // EMBED: [n_slots, B*S]; FINAL: labels
+  // per-kind offsets into stash (floats, within a slot)
+  int64_t o_x = 0, o_mean1 = 0, o_rstd1 = 0, o_ln1 = 0, o_qkv = 0, o_p = 0,
+          o_attnm = 0, o_hmid = 0, o_mean2 = 0, o_rstd2 = 0, o_ln2 = 0,
+          o_u = 0, o_g = 0, o_logits = 0, o_lse = 0;
+};
+
+// parameter offsets (canonical layout, oracle/gpt2_oracle.py::layer_param_spec)
+struct BlockParams {
+  int64_t ln1_w, ln1_b, w_qkv, b_qkv, w_attnproj, b_attnproj, ln2_w, ln2_b,
+      w_fc, b_fc, w_mlpproj, b_mlpproj, total;
+};
+static BlockParams block_params(int64_t H) {
+  BlockParams p;
+  int64_t o = 0;
+  p.ln1_w = o; o += H;
+  p.ln1_b = o; o += H;
+  p.w_qkv = o; o += H * 3 * H;
+  p.b_qkv = o; o += 3 * H;
+  p.w_attnproj = o; o += H * H;
+  p.b_attnproj = o; o += H;
+  p.ln2_w = o; o += H;
+  p.ln2_b = o; o += H;
+  p.w_fc = o; o += H * 4 * H;
+  p.b_fc = o; o += 4 * H;
+  p.w_mlpproj = o; o += 4 * H * H;
+  p.b_mlpproj = o; o += H;
+  p.total = o;
+  return p;
+}
+
+extern "C" int64_t ob_layer_param_count(const ob_layer_desc* d) {
+  const int64_t H = d->n_embd, V = d->vocab_size, P = d->n_positions;
+  switch (d->kind) {
+    case OB_KIND_EMBED: return V * H + P * H;
+    case OB_KIND_BLOCK: return block_params(H).total;
+    case OB_KIND_FINAL: return 2 * H + V * H;
+  }
+  return -1;
+}
+
+extern "C" int ob_layer_create(const ob_layer_desc* d, ob_layer_t* out) {
+  if (!d || !out) return ob_fail("create: null arg");
+  if (d->kind < 0 || d->kind > 2) return ob_fail("create: bad kind");
+  if (d->seq_len > 2048)
+    return ob_fail("create: seq_len > 2048 unsupported in round 1");
+  ob_layer* l = new ob_layer();
+  l->d = *d;
+  l->B = d->max_batch;
+  const int64_t Bm = d->max_batch, Sq = d->seq_len, H = d->n_embd,
+                nh = d->n_head, V = d->vocab_size;
+  const int64_t BS = Bm * Sq, BSH = BS * H;
+  int64_t o = 0;
+  switch (d->kind) {
+    case OB_KIND_EMBED:
+      l->slot_stride = 0;
+      break;
+    case OB_KIND_BLOCK:
+      l->o_x = o; o += BSH;
+      l->o_mean1 = o; o += BS;
+      l->o_rstd1 = o; o += BS;
+      l->o_ln1 = o; o += BSH;
+      l->o_qkv = o; o += BS * 3 * H;
+      l->o_p = o; o += Bm * nh * Sq * Sq;
+      l->o_attnm = o; o += BSH;
+      l->o_hmid = o; o += BSH;
+      l->o_mean2 = o; o += BS;
+      l->o_rstd2 = o; o += BS;
+      l->o_ln2 = o; o += BSH;
+      l->o_u = o; o += BS * 4 * H;
+      l->o_g = o; o += BS * 4 * H;
+      l->slot_stride = o;
+      break;
+    case OB_KIND_FINAL:
+      l->o_x = o; o += BSH;
+      l->o_mean1 = o; o += BS;
+      l->o_rstd1 = o; o += BS;
+      l->o_ln1 = o; o += BSH;
+      l->o_logits = o; o += BS * V;
+      l->o_lse = o; o += BS;
+      l->slot_stride = o;
+      break;
+  }
+  if (l->slot_stride > 0) {
+    if (hipMalloc(&l->stash, (size_t)l->slot_stride * d->n_slots *
+                                 sizeof(float)) != hipSuccess) {
+      delete l;
+      return ob_fail("create: stash hipMalloc of %lld floats failed",
+                     (long long)(l->slot_stride * d->n_slots));
+    }
+  }
+  if (d->kind == OB_KIND_EMBED || d->kind == OB_KIND_FINAL) {
+    if (hipMalloc(&l->ids, (size_t)BS * d->n_slots * sizeof(int64_t)) !=
+        hipSuccess) {
+      if (l->stash) hipFree(l->stash);
+      delete l;
+      return ob_fail("create: ids hipMalloc failed");
+    }
+  }
+  // grow the shared backward workspace
+  if (d->kind == OB_KIND_BLOCK) {
+    if (ws_ensure(&g_ws.dp, &g_ws.sz_dp, Bm * nh * Sq * Sq)) return 1;
+    if (ws_ensure(&g_ws.dqkv, &g_ws.sz_dqkv, BS * 3 * H)) return 1;
+    if (ws_ensure(&g_ws.b4h, &g_ws.sz_b4h, BS * 4 * H)) return 1;
+  }
+  if (d->kind != OB_KIND_EMBED) {
+    const int64_t need = BSH;
+    if (ws_ensure(&g_ws.bsh1, &g_ws.sz_bsh, need)) return 1;
+    if (ws_ensure(&g_ws.bsh2, &g_ws.sz_bsh2, need)) return 1;
+  }
+  *out = l;
+  return 0;
+}
+
+extern "C" int ob_layer_bind(ob_layer_t l, void* params, void* grads) {
+  if (!l) return ob_fail("bind: null layer");
+  l->params = (float*)params;
+  l->grads = (float*)grads;
+  return 0;
+}
+
+extern "C" int ob_layer_set_batch(ob_layer_t l, int32_t batch) {
+  if (!l) return ob_fail("set_batch: null layer");
+  if (batch < 1 || batch > l->d.max_batch)
+    return ob_fail("set_batch: %d out of range (max %d)", batch,
+                   l->d.max_batch);
+  l->B = batch;
+  return 0;
+}
+
+extern "C" int ob_layer_destroy(ob_layer_t l) {
+  if (!l) return 0;
+  if (l->stash) hipFree(l->stash);
+  if (l->ids) hipFree(l->ids);
+  delete l;
+  return 0;
+}
+
+// helper: plain strided-batch GEMM call (n1=batch, n2=1) or 2-level.
+static int gemm(int tA, int tB, int64_t M, int64_t N, int64_t K, float alpha,
+                const float* A, int64_t lda, int64_t sA1, int64_t sA2,
+                const float* B, int64_t ldb, int64_t sB1, int64_t sB2,
+                float beta, float* C, int64_t ldc, int64_t sC1, int64_t sC2,
+                int64_t n1, int64_t n2, const float* bias, const float* R,
+                int atomic, int splitk, void* stream) {
+  return ob_gemm_f32(tA, tB, M, N, K, alpha, A, lda, sA1, sA2, B, ldb, sB1,
+                     sB2, beta, C, ldc, sC1, sC2, n1, n2, bias, R, atomic,
+                     splitk, stream);
+}
+
+// pick a split-K factor for weight-grad GEMMs so the grid fills the chip
+// (256 CUs want >= ~512 blocks of 128x128).
+static int pick_splitk(int64_t M, int64_t N, int64_t K) {
+  const int64_t tiles = ((M + 127) / 128) * ((N + 127) / 128);
+  int sk = 1;
+  while (sk < 16 && tiles * sk < 512 && (K / (sk * 2)) >= 256) sk *= 2;
+  return sk;
+}
+
+// ---------------------------------------------------------------------------
+// forward
+// ---------------------------------------------------------------------------
+
+static int block_forward(ob_layer* l, int slot, const float* in, float* out,
+                         void* stream) {
+  const int64_t Sq = l->d.seq_len, H = l->d.n_embd, nh = l->d.n_head;
+  const int64_t B = l->B, BS = B * Sq;
+  const int64_t hd = H / nh;
+  float* st = l->stash + (int64_t)slot * l->slot_stride;
+  const float* p = l->params;
+  const BlockParams bp = block_params(H);
+
+  float* x = st + l->o_x;
+  OB_HIP(hipMemcpyAsync(x, in, BS * H * sizeof(float), hipMemcpyDeviceToDevice,
+                        S(stream)));
+  float* ln1 = st + l->o_ln1;
+  if (ob_layernorm_fwd_f32(x, p + bp.ln1_w, p + bp.ln1_b, ln1,
+                           st + l->o_mean1, st + l->o_rstd1, BS, H, 1e-5f,
+                           stream))
+    return 1;
+  float* qkv = st + l->o_qkv;
+  if (gemm(0, 0, BS, 3 * H, H, 1.f, ln1, H, 0, 0, p + bp.w_qkv, 3 * H, 0, 0,
+           0.f, qkv, 3 * H, 0, 0, 1, 1, p + bp.b_qkv, nullptr, 0, 1, stream))
+    return 1;
+  // scores: P[b,h] = Q @ K^T ; Q/K are strided slices of qkv
+  float* P = st + l->o_p;
+  if (gemm(0, 1, Sq, Sq, hd, 1.f,
+           qkv, 3 * H, Sq * 3 * H, hd,            // Q slice
+           qkv + H, 3 * H, Sq * 3 * H, hd,        // K slice (stored [S,hd])
+           0.f, P, Sq, nh * Sq * Sq, Sq * Sq, B, nh, nullptr, nullptr, 0, 1,
+           stream))
+    return 1;
+  if (ob_softmax_causal_fwd_f32(P, B * nh, Sq, 1.f / sqrtf((float)hd), stream))
+    return 1;
+  // attn_merged[b,:,h*hd:(h+1)*hd] = P[b,h] @ V[b,h]
+  float* am = st + l->o_attnm;
+  if (gemm(0, 0, Sq, hd, Sq, 1.f, P, Sq, nh * Sq * Sq, Sq * Sq, qkv + 2 * H,
+           3 * H, Sq * 3 * H, hd, 0.f, am, H, Sq * H, hd, B, nh, nullptr,
+           nullptr, 0, 1, stream))
+    return 1;
+  // h_mid = attn_merged @ w_attnproj + b + x (residual)
+  float* hmid = st + l->o_hmid;
+  if (gemm(0, 0, BS, H, H, 1.f, am, H, 0, 0, p + bp.w_attnproj, H, 0, 0, 0.f,
+           hmid, H, 0, 0, 1, 1, p + bp.b_attnproj, x, 0, 1, stream))
+    return 1;
+  float* ln2 = st + l->o_ln2;
+  if (ob_layernorm_fwd_f32(hmid, p + bp.ln2_w, p + bp.ln2_b, ln2,
+                           st + l->o_mean2, st + l->o_rstd2, BS, H, 1e-5f,
+                           stream))
+    return 1;
+  float* u = st + l->o_u;
+  if (gemm(0, 0, BS, 4 * H, H, 1.f, ln2, H, 0, 0, p + bp.w_fc, 4 * H, 0, 0,
+           0.f, u, 4 * H, 0, 0, 1, 1, p + bp.b_fc, nullptr, 0, 1, stream))
+    return 1;
+  float* gact = st + l->o_g;
+  if (ob_gelu_fwd_f32(u, gact, BS * 4 * H, stream)) return 1;
+  if (gemm(0, 0, BS, H, 4 * H, 1.f, gact, 4 * H, 0, 0, p + bp.w_mlpproj, H, 0,
+           0, 0.f, out, H, 0, 0, 1, 1, p + bp.b_mlpproj, hmid, 0, 1, stream))
+    return 1;
+  return 0;
+}
+
+static int final_forward(ob_layer* l, int slot, const float* in, float* out,
+                         const int64_t* labels, void* stream) {
+  const int64_t Sq = l->d.seq_len, H = l->d.n_embd, V = l->d.vocab_size;
+  const int64_t B = l->B, BS = B * Sq;
+  float* st = l->stash + (int64_t)slot * l->slot_stride;
+  const float* p = l->params;
+  if (!labels) return ob_fail("final_forward: labels required");
+
+  float* x = st + l->o_x;
+  OB_HIP(hipMemcpyAsync(x, in, BS * H * sizeof(float), hipMemcpyDeviceToDevice,
+                        S(stream)));
+  int64_t* labs = l->ids + (int64_t)slot * (int64_t)l->d.max_batch * Sq;
+  OB_HIP(hipMemcpyAsync(labs, labels, BS * sizeof(int64_t),
+                        hipMemcpyDeviceToDevice, S(stream)));
+  float* lnf = st + l->o_ln1;
+  if (ob_layernorm_fwd_f32(x, p + 0, p + H, lnf, st + l->o_mean1,
+                           st + l->o_rstd1, BS, H, 1e-5f, stream))
+    return 1;
+  float* logits = st + l->o_logits;
+  // logits = ln_out @ w_lm^T   (w_lm stored [V,H])
+  if (gemm(0, 1, BS, V, H, 1.f, lnf, H, 0, 0, p + 2 * H, H, 0, 0, 0.f, logits,
+           V, 0, 0, 1, 1, nullptr, nullptr, 0, 1, stream))
+    return 1;
+  OB_HIP(hipMemsetAsync(out, 0, sizeof(float), S(stream)));
+  if (ob_ce_fwd_f32(logits, labs, st + l->o_lse, out, B, Sq, V, stream))
+    return 1;
+  return 0;
+}
+
+extern "C" int ob_layer_forward(ob_layer_t l, int32_t slot, const void* in,
+                                void* out, const int64_t* labels,
+                                void* stream) {
+  if (!l) return ob_fail("forward: null layer");
+  if (!l->params) return ob_fail("forward: params not bound");
+  if (slot < 0 || slot >= l->d.n_slots) return ob_fail("forward: bad slot");
+  switch (l->d.kind) {
+    case OB_KIND_EMBED: {
+      const int64_t Sq = l->d.seq_len, H = l->d.n_embd, V = l->d.vocab_size;
+      const int64_t BS = (int64_t)l->B * Sq;
+      int64_t* ids = l->ids + (int64_t)slot * (int64_t)l->d.max_batch * Sq;
+      OB_HIP(hipMemcpyAsync(ids, in, BS * sizeof(int64_t),
+                            hipMemcpyDeviceToDevice, S(stream)));
+      return ob_embed_fwd_f32(ids, l->params, l->params + V * H, (float*)out,
+                              l->B, Sq, H, stream);
+    }
+    case OB_KIND_BLOCK:
+      return block_forward(l, slot, (const float*)in, (float*)out, stream);
+    case OB_KIND_FINAL:
+      return final_forward(l, slot, (const float*)in, (float*)out, labels,
+                           stream);
+  }
+  return ob_fail("forward: bad kind");
+}
+
+// ---------------------------------------------------------------------------
+// backward (accumulates into the bound grad buffer)
+// ---------------------------------------------------------------------------
+
+static int block_backward(ob_layer* l, int slot, const float* dout, float* din,
+                          void* stream) {
+  const int64_t Sq = l->d.seq_len, H = l->d.n_embd, nh = l->d.n_head;
+  const int64_t B = l->B, BS = B * Sq;
+  const int64_t hd = H / nh;
+  const float scale = 1.f / sqrtf((float)hd);
+  float* st = l->stash + (int64_t)slot * l->slot_stride;
+  const float* p = l->params;
+  float* g = l->grads;
+  const BlockParams bp = block_params(H);
+
+  float* x = st + l->o_x;
+  float* ln1 = st + l->o_ln1;
+  float* qkv = st + l->o_qkv;
+  float* P = st + l->o_p;
+  float* am = st + l->o_attnm;
+  float* hmid = st + l->o_hmid;
+  float* ln2 = st + l->o_ln2;
+  float* u = st + l->o_u;
+  float* gact = st + l->o_g;
+
+  float* DY4 = g_ws.b4h;    // [BS,4H]
+  float* DLN = g_ws.bsh1;   // [BS,H]
+  float* DATT = g_ws.bsh2;  // [BS,H]
+  float* DQKV = g_ws.dqkv;  // [BS,3H]
+  float* DP = g_ws.dp;      // [B*nh,S,S]
+
+  // din starts as d(h_mid) accumulator = dout (residual skip)
+  OB_HIP(hipMemcpyAsync(din, dout, BS * H * sizeof(float),
+                        hipMemcpyDeviceToDevice, S(stream)));
+  // ---- MLP backward ----
+  // dg = dout @ w_mlpproj^T
+  if (gemm(0, 1, BS, 4 * H, H, 1.f, dout, H, 0, 0, p + bp.w_mlpproj, H, 0, 0,
+           0.f, DY4, 4 * H, 0, 0, 1, 1, nullptr, nullptr, 0, 1, stream))
+    return 1;
+  // dW_mlpproj += g^T @ dout ; db += colsum(dout)
+  if (gemm(1, 0, 4 * H, H, BS, 1.f, gact, 4 * H, 0, 0, dout, H, 0, 0, 1.f,
+           g + bp.w_mlpproj, H, 0, 0, 1, 1, nullptr, nullptr, 1,
+           pick_splitk(4 * H, H, BS), stream))
+    return 1;
+  if (ob_colsum_f32(dout, g + bp.b_mlpproj, BS, H, stream)) return 1;
+  // du = dg * gelu'(u)   (in place on DY4)
+  if (ob_gelu_bwd_f32(u, DY4, DY4, BS * 4 * H, stream)) return 1;
+  // dW_fc += ln2^T @ du ; db_fc += colsum(du)
+  if (gemm(1, 0, H, 4 * H, BS, 1.f, ln2, H, 0, 0, DY4, 4 * H, 0, 0, 1.f,
+           g + bp.w_fc, 4 * H, 0, 0, 1, 1, nullptr, nullptr, 1,
+           pick_splitk(H, 4 * H, BS), stream))
+    return 1;
+  if (ob_colsum_f32(DY4, g + bp.b_fc, BS, 4 * H, stream)) return 1;
+  // d_ln2out = du @ w_fc^T
+  if (gemm(0, 1, BS, H, 4 * H, 1.f, DY4, 4 * H, 0, 0, p + bp.w_fc, 4 * H, 0,
+           0, 0.f, DLN, H, 0, 0, 1, 1, nullptr, nullptr, 0, 1, stream))
+    return 1;
+  // ln2 backward: din += dx ; dw/db accumulate
+  if (ob_layernorm_bwd_f32(hmid, p + bp.ln2_w, st + l->o_mean2,
+                           st + l->o_rstd2, DLN, din, g + bp.ln2_w,
+                           g + bp.ln2_b, BS, H, 1, stream))
+    return 1;
+  // ---- attention projection backward (din now = d_hmid) ----
+  if (gemm(0, 1, BS, H, H, 1.f, din, H, 0, 0, p + bp.w_attnproj, H, 0, 0, 0.f,
+           DATT, H, 0, 0, 1, 1, nullptr, nullptr, 0, 1, stream))
+    return 1;
+  if (gemm(1, 0, H, H, BS, 1.f, am, H, 0, 0, din, H, 0, 0, 1.f,
+           g + bp.w_attnproj, H, 0, 0, 1, 1, nullptr, nullptr, 1,
+           pick_splitk(H, H, BS), stream))
+    return 1;
+  if (ob_colsum_f32(din, g + bp.b_attnproj, BS, H, stream)) return 1;
+  // ---- attention core backward ----
+  // dP = dO @ V^T
+  if (gemm(0, 1, Sq, Sq, hd, 1.f, DATT, H, Sq * H, hd, qkv + 2 * H, 3 * H,
+           Sq * 3 * H, hd, 0.f, DP, Sq, nh * Sq * Sq, Sq * Sq, B, nh, nullptr,
+           nullptr, 0, 1, stream))
+    return 1;
+  // dS = softmax_bwd(P, dP)  in place on DP
+  if (ob_softmax_causal_bwd_f32(P, DP, B * nh, Sq, stream)) return 1;
+  // dQ = scale * dS @ K
+  if (gemm(0, 0, Sq, hd, Sq, scale, DP, Sq, nh * Sq * Sq, Sq * Sq, qkv + H,
+           3 * H, Sq * 3 * H, hd, 0.f, DQKV, 3 * H, Sq * 3 * H, hd, B, nh,
+           nullptr, nullptr, 0, 1, stream))
+    return 1;
+  // dK = scale * dS^T @ Q
+  if (gemm(1, 0, Sq, hd, Sq, scale, DP, Sq, nh * Sq * Sq, Sq * Sq, qkv, 3 * H,
+           Sq * 3 * H, hd, 0.f, DQKV + H, 3 * H, Sq * 3 * H, hd, B, nh,
+           nullptr, nullptr, 0, 1, stream))
+    return 1;
+  // dV = P^T @ dO
+  if (gemm(1, 0, Sq, hd, Sq, 1.f, P, Sq, nh * Sq * Sq, Sq * Sq, DATT, H,
+           Sq * H, hd, 0.f, DQKV + 2 * H, 3 * H, Sq * 3 * H, hd, B, nh,
+           nullptr, nullptr, 0, 1, stream))
+    return 1;
+  // ---- QKV projection backward ----
+  if (ob_colsum_f32(DQKV, g + bp.b_qkv, BS, 3 * H, stream)) return 1;
+  if (gemm(1, 0, H, 3 * H, BS, 1.f, ln1, H, 0, 0, DQKV, 3 * H, 0, 0, 1.f,
+           g + bp.w_qkv, 3 * H, 0, 0, 1, 1, nullptr, nullptr, 1,
+           pick_splitk(H, 3 * H, BS), stream))
+    return 1;
+  if (gemm(0, 1, BS, H, 3 * H, 1.f, DQKV, 3 * H, 0, 0, p + bp.w_qkv, 3 * H, 0,
+           0, 0.f, DLN, H, 0, 0, 1, 1, nullptr, nullptr, 0, 1, stream))
+    return 1;
+  // ln1 backward: din += dx
+  if (ob_layernorm_bwd_f32(x, p + bp.ln1_w, st + l->o_mean1, st + l->o_rstd1,
+                           DLN, din, g + bp.ln1_w, g + bp.ln1_b, BS, H, 1,
+                           stream))
+    return 1;
+  return 0;
+}
+
+static int final_backward(ob_layer* l, int slot, const float* dout, float* din,
+                          void* stream) {
+  const int64_t Sq = l->d.seq_len, H = l->d.n_embd, V = l->d.vocab_size;
+  const int64_t B = l->B, BS = B * Sq;
+  float* st = l->stash + (int64_t)slot * l->slot_stride;
+  const float* p = l->params;
+  float* g = l->grads;
+  float* x = st + l->o_x;
+  float* lnf = st + l->o_ln1;
+  float* logits = st + l->o_logits;  // becomes dlogits in place
+  int64_t* labs = l->ids + (int64_t)slot * (int64_t)l->d.max_batch * Sq;
+
+  if (ob_ce_bwd_f32(logits, labs, st + l->o_lse, (const float*)dout, B, Sq, V,
+                    stream))
+    return 1;
+  // dW_lm += dlogits^T @ ln_out
+  if (gemm(1, 0, V, H, BS, 1.f, logits, V, 0, 0, lnf, H, 0, 0, 1.f, g + 2 * H,
+           H, 0, 0, 1, 1, nullptr, nullptr, 1, 1, stream))
+    return 1;
+  // d_lnout = dlogits @ w_lm
+  if (gemm(0, 0, BS, H, V, 1.f, logits, V, 0, 0, p + 2 * H, H, 0, 0, 0.f,
+           g_ws.bsh1, H, 0, 0, 1, 1, nullptr, nullptr, 0, 1, stream))
+    return 1;
+  if (ob_layernorm_bwd_f32(x, p + 0, st + l->o_mean1, st + l->o_rstd1,
+                           g_ws.bsh1, din, g + 0, g + H, BS, H, 0, stream))
+    return 1;
+  return 0;
+}
+
+extern "C" int ob_layer_backward(ob_layer_t l, int32_t slot, const void* dout,
+                                 void* din, void* stream) {
+  if (!l) return ob_fail("backward: null layer");
+  if (!l->params || !l->grads) return ob_fail("backward: buffers not bound");
+  if (slot < 0 || slot >= l->d.n_slots) return ob_fail("backward: bad slot");
+  switch (l->d.kind) {
+    case OB_KIND_EMBED: {
+      const int64_t Sq = l->d.seq_len, H = l->d.n_embd, V = l->d.vocab_size;
+      int64_t* ids = l->ids + (int64_t)slot * (int64_t)l->d.max_batch * Sq;
+      return ob_embed_bwd_f32(ids, (const float*)dout, l->grads,
+                              l->grads + V * H, l->B, Sq, H, stream);
+    }
+    case OB_KIND_BLOCK:
+      if (!dout || !din) return ob_fail("block backward: dout/din required");
+      return block_backward(l, slot, (const float*)dout, (float*)din, stream);
+    case OB_KIND_FINAL:
+      if (!din) return ob_fail("final backward: din required");
+      return final_backward(l, slot, (const float*)dout, (float*)din, stream);
+  }
+  return ob_fail("backward: bad kind");
+}

