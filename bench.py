@@ -1,0 +1,235 @@
+#!/usr/bin/env python3
+# bench.py — the reference's headline benchmark (BASELINE.json: tokens/sec,
+# whole node, GPT-2 1F1B) on the MI355X-native hot path.
+#
+# One "step" = one full training step of the 1F1B pipeline over the global
+# microbatch (gpt2.yaml: global_microbatch 128 sequences of S=1024 = 131072
+# tokens/step), i.e. pipeline.train() + per-layer DP all-reduce + fused
+# AdamW — the exact three calls of the reference's _train_step
+# (engine.py:645-649).  Synthetic data (no network), random-init weights,
+# fp32 (the reference's compute dtype; bf16 is §8 f4).
+#
+#   python bench.py --gpus N --steps K --warmup W
+#
+# N>1 is launched by the driver via torch.distributed.run (one rank/GPU,
+# RCCL).  Topology: N=1 -> 1 stage; N=2 -> 2 stages; N=4 -> 4 stages;
+# N=8 -> 4 stages x 2 DP replicas (BASELINE config[2]).
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+TOPOLOGY = {1: (1, 1), 2: (2, 1), 4: (4, 1), 8: (4, 2)}  # gpus -> (stages, replicas)
+F32_MFMA_PEAK_TF = 157.3  # gfx950 f32 matrix peak (MI355X_MICROARCH.md)
+
+
+def build_world(args, device):
+    from oobleck_amd.config import GPT2_SMALL, TrainingConfig
+    from oobleck_amd.engine import (DataParallelEngine, even_stage_split,
+                                    make_rank_grid)
+    from oobleck_amd.layer import Layer
+    from oobleck_amd.optimizer import FusedAdamW, WarmupLR
+    from oobleck_amd.pipeline import OobleckPipeline, SyntheticDataLoader
+
+    mc = GPT2_SMALL
+    tc = TrainingConfig(seq_len=args.seq_len,
+                        microbatch_size=args.microbatch,
+                        global_microbatch_size=args.global_batch)
+    stages, replicas = TOPOLOGY[args.gpus]
+    assert stages * replicas == args.gpus
+    mb_total = tc.num_microbatches
+    assert mb_total % replicas == 0
+    mb_per_pipe = mb_total // replicas
+
+    stage_layers = even_stage_split(mc, stages)
+    pipelines, my_pipeline = [], None
+    for pid in range(replicas):
+        ranks = [pid * stages + s for s in range(stages)]
+        grid = make_rank_grid(mc.n_layers_total, stage_layers,
+                              [[r] for r in ranks])
+        loader = SyntheticDataLoader(mc, tc.microbatch_size, tc.seq_len,
+                                     seed=1234 + pid)
+        p = OobleckPipeline(pid, grid, mc, tc, loader, mb_per_pipe, device)
+        p.initialize_distributed_fsdp()
+        p.initialize_distributed_pipeline()
+        pipelines.append(p)
+    for p in pipelines:
+        if p.my_pipeline:
+            def layer_factory(lid, pg, n_slots):
+                return Layer(lid, mc, tc.microbatch_size, tc.seq_len,
+                             n_slots, device)
+
+            def optimizer_factory(layers):
+                opt = FusedAdamW(layers, lr=tc.lr,
+                                 betas=(tc.adam_beta1, tc.adam_beta2),
+                                 eps=tc.adam_eps, weight_decay=tc.weight_decay)
+                return opt, WarmupLR(opt, tc.warmup_steps)
+
+            p.initialize_execution(layer_factory, optimizer_factory)
+            my_pipeline = p
+    dp = DataParallelEngine(pipelines)
+    return mc, tc, pipelines, my_pipeline, dp
+
+
+def train_step(my_pipeline, dp):
+    # the reference's _train_step (engine.py:645-649): pipeline.train() ->
+    # dp all-reduce -> optimizer step
+    my_pipeline.train()
+    dp.do_allreduce(my_pipeline)
+    my_pipeline.execution.optimizer_step()
+    # grads accumulate across microbatches within a step; clear for the next
+    for layer in my_pipeline.execution._layers:
+        layer.zero_grads()
+
+
+def measure_roofline(args, device):
+    """Dominant-kernel roofline: the fp32 MFMA GEMM at its most-executed hot
+    shape (the MLP fc GEMM of config[0]: M=B*S=8192, N=4H=3072, K=H=768).
+    achieved = algorithmic FLOPs per launch / avg launch duration (HIP
+    events on the launch stream).  DESIGN.md derives the per-unit figures."""
+    from tests.gpu_helpers import gemm
+    M, N, K = args.microbatch * args.seq_len, 4 * 768, 768
+    A = torch.randn(M, K, device=device)
+    B = torch.randn(K, N, device=device)
+    C = torch.empty(M, N, device=device)
+    for _ in range(3):
+        gemm(A, B, C, M=M, N=N, K=K, lda=K, ldb=N, ldc=N)
+    torch.cuda.synchronize()
+    start, end = torch.cuda.Event(True), torch.cuda.Event(True)
+    reps = 50
+    start.record()
+    for _ in range(reps):
+        gemm(A, B, C, M=M, N=N, K=K, lda=K, ldb=N, ldc=N)
+    end.record()
+    torch.cuda.synchronize()
+    avg_s = start.elapsed_time(end) / 1000.0 / reps
+    flops = 2.0 * M * N * K
+    achieved_tf = flops / avg_s / 1e12
+    traffic = None
+    pmc_file = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                            "profiles", "pmc_gemm_fc.json")
+    if os.path.exists(pmc_file):
+        with open(pmc_file) as f:
+            traffic = json.load(f).get("hbm_bytes_per_launch")
+    return {
+        "bound": "mfma", "achieved": round(achieved_tf, 2),
+        "peak": F32_MFMA_PEAK_TF, "unit": "TFLOP/s",
+        "frac": round(achieved_tf / F32_MFMA_PEAK_TF, 4),
+        "traffic": traffic,
+        "kernel": "k_gemm_f32 (MLP fc, M=8192 N=3072 K=768, fp32 MFMA)",
+        "avg_launch_ms": round(avg_s * 1e3, 4),
+    }
+
+
+def measure_cpu_baseline():
+    """The oracle (CPU restatement of the reference's arithmetic) timed on
+    this box's host cores — the reported baseline (kind=port), bounded to a
+    ~10-30 s sample: one fwd+bwd microbatch at B=1, S=512."""
+    from oracle.gpt2_oracle import OracleConfig, stage_forward_backward
+    from oracle.gpt2_oracle import init_layer_params
+    cores = os.cpu_count() or 8
+    torch.set_num_threads(cores)
+    oc = OracleConfig()
+    flats = [init_layer_params(oc, oc.layer_kind(i), 42 + i)
+             for i in range(oc.n_layers_total)]
+    B, S = 1, 512
+    g = torch.Generator().manual_seed(0)
+    ids = torch.randint(0, oc.vocab_size, (B, S), generator=g)
+    stage_forward_backward(oc, flats, list(range(oc.n_layers_total)), ids,
+                           labels=ids.clone())  # warm-up
+    t0 = time.perf_counter()
+    stage_forward_backward(oc, flats, list(range(oc.n_layers_total)), ids,
+                           labels=ids.clone())
+    dt = time.perf_counter() - t0
+    return {
+        "value": round(B * S / dt, 2), "unit": "tokens/s", "cores": cores,
+        "kind": "port",
+        "sample": f"oracle GPT-2-small fwd+bwd, B={B} S={S} ({B*S} tokens), "
+                  f"1 timed iter after 1 warmup, torch {torch.__version__} CPU",
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--seq-len", type=int, default=1024)
+    ap.add_argument("--microbatch", type=int, default=8)
+    ap.add_argument("--global-batch", type=int, default=128)
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    ap.add_argument("--skip-roofline", action="store_true")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    assert world == args.gpus, f"WORLD_SIZE {world} != --gpus {args.gpus}"
+
+    torch.cuda.set_device(local_rank)
+    device = torch.device("cuda", local_rank)
+    if world > 1:
+        dist.init_process_group("nccl")
+    else:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29571")
+        dist.init_process_group("nccl", rank=0, world_size=1)
+
+    mc, tc, pipelines, my_pipeline, dp = build_world(args, device)
+
+    for _ in range(args.warmup):
+        train_step(my_pipeline, dp)
+    dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        train_step(my_pipeline, dp)
+    dist.barrier()
+    torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    # MAX over ranks
+    e = torch.tensor([elapsed], device=device)
+    dist.all_reduce(e, op=dist.ReduceOp.MAX)
+    elapsed = float(e.item())
+
+    if rank == 0:
+        tokens_per_step = args.global_batch * args.seq_len
+        value = tokens_per_step * args.steps / elapsed
+        stages, replicas = TOPOLOGY[args.gpus]
+        result = {
+            "metric": "tokens/sec (whole node) GPT-2 1F1B",
+            "value": round(value, 1),
+            "unit": "tokens/s",
+            "n_gpus": args.gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 1),
+            "higher_is_better": True,
+            "scaling": "strong",  # global batch fixed at 128 as N grows
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "f32",       # the reference's compute dtype (fp32)
+            "data": "synthetic",
+            "config": {
+                "workload": "gpt2.yaml GPT-2-small 1F1B, global microbatch "
+                            "128x1024 tokens, fp32",
+                "model": "gpt2",
+                "global_batch": args.global_batch,
+                "seq_len": args.seq_len,
+                "parallelism": f"pp{stages}dp{replicas}",
+            },
+        }
+        if not args.skip_roofline:
+            result["roofline"] = measure_roofline(args, device)
+        if not args.skip_cpu_baseline:
+            result["cpu_baseline"] = measure_cpu_baseline()
+        print(json.dumps(result))
+    dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
