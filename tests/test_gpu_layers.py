@@ -153,7 +153,10 @@ def test_full_model_single_gpu_parity(dims, B, S):
         torch.testing.assert_close(layer.flat_grad.cpu(), ref_grads[lid],
                                    rtol=1e-3, atol=1e-3)
 
-    # fused AdamW step parity
+    # fused AdamW step parity: oracle Adam on the GPU's OWN grads (grad
+    # parity is asserted above; at step 1 Adam ≈ lr*sign(g), so feeding the
+    # oracle a slightly different grad would amplify fp noise to 2*lr)
+    gpu_grads = [layer.flat_grad.cpu().clone() for layer in layers]
     from oobleck_amd.optimizer import FusedAdamW
     opt = FusedAdamW(layers, lr=1e-3, weight_decay=0.01)
     opt.step()
@@ -162,7 +165,7 @@ def test_full_model_single_gpu_parity(dims, B, S):
         p = flats[lid].clone()
         m = torch.zeros_like(p)
         v = torch.zeros_like(p)
-        adamw_step(p, ref_grads[lid], m, v, 1, 1e-3, weight_decay=0.01)
+        adamw_step(p, gpu_grads[lid], m, v, 1, 1e-3, weight_decay=0.01)
         torch.testing.assert_close(layer.flat_param.cpu(), p,
                                    rtol=1e-4, atol=1e-5)
 
