@@ -164,7 +164,16 @@ def main():
     ap.add_argument("--global-batch", type=int, default=128)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     ap.add_argument("--skip-roofline", action="store_true")
+    ap.add_argument("--probe", choices=["gemm"], default=None,
+                    help="run ONLY the dominant-kernel roofline probe "
+                         "(for clean rocprofv3 kernel-trace/PMC capture)")
     args = ap.parse_args()
+
+    if args.probe == "gemm":
+        torch.cuda.set_device(0)
+        print(json.dumps({"probe": "gemm",
+                          "roofline": measure_roofline(args, torch.device("cuda", 0))}))
+        return
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
