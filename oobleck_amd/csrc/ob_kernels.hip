@@ -89,75 +89,87 @@ __device__ __forceinline__ float block_max256(float v, float* lds4) {
 // ---------------------------------------------------------------------------
 
 #define GEMM_BM 128
-#define GEMM_BN 128
 #define GEMM_BK 32
-#define GEMM_LDT 132  // padded LDS row stride (floats)
+#define GEMM_LDT 132  // padded LDS row stride for the BM=128 A tile (floats)
 
 // stage dst[k][c] = src[k][c] (src already offset; ld = src row stride).
-// kmax/cmax = valid extents (zero-fill outside).  256 threads.
+// kmax/cmax = valid extents (zero-fill outside; EDGE=false skips every
+// guard for interior tiles).  256 threads; NC = tile width (64 or 128).
+template <int NC, bool EDGE>
 __device__ __forceinline__ void stage_direct(float* dst, const float* src,
                                              int64_t ld, int kmax, int cmax) {
-  const int c4 = (threadIdx.x & 31) * 4;  // 0..124
-  const int k0 = threadIdx.x >> 5;        // 0..7
+  constexpr int LDT = NC + 4;
+  const int c4 = (threadIdx.x % (NC / 4)) * 4;
+  const int k0 = threadIdx.x / (NC / 4);
+  constexpr int KSTEP = 1024 / NC;  // rows covered per pass
 #pragma unroll
-  for (int it = 0; it < 4; ++it) {
-    const int k = k0 + it * 8;
-    float4 v = {0.f, 0.f, 0.f, 0.f};
-    if (k < kmax) {
-      if (c4 + 3 < cmax) {
-        v = *reinterpret_cast<const float4*>(src + (int64_t)k * ld + c4);
-      } else {
-        if (c4 + 0 < cmax) v.x = src[(int64_t)k * ld + c4 + 0];
-        if (c4 + 1 < cmax) v.y = src[(int64_t)k * ld + c4 + 1];
-        if (c4 + 2 < cmax) v.z = src[(int64_t)k * ld + c4 + 2];
-        if (c4 + 3 < cmax) v.w = src[(int64_t)k * ld + c4 + 3];
+  for (int it = 0; it < NC / 32; ++it) {
+    const int k = k0 + it * KSTEP;
+    if (EDGE) {
+      float4 v = {0.f, 0.f, 0.f, 0.f};
+      if (k < kmax) {
+        if (c4 + 3 < cmax) {
+          v = *reinterpret_cast<const float4*>(src + (int64_t)k * ld + c4);
+        } else {
+          if (c4 + 0 < cmax) v.x = src[(int64_t)k * ld + c4 + 0];
+          if (c4 + 1 < cmax) v.y = src[(int64_t)k * ld + c4 + 1];
+          if (c4 + 2 < cmax) v.z = src[(int64_t)k * ld + c4 + 2];
+          if (c4 + 3 < cmax) v.w = src[(int64_t)k * ld + c4 + 3];
+        }
       }
+      *reinterpret_cast<float4*>(dst + k * LDT + c4) = v;
+    } else {
+      *reinterpret_cast<float4*>(dst + k * LDT + c4) =
+          *reinterpret_cast<const float4*>(src + (int64_t)k * ld + c4);
     }
-    *reinterpret_cast<float4*>(dst + k * GEMM_LDT + c4) = v;
   }
 }
 
 // stage dst[k][c] = src[c][k] (transpose; src offset to (c0,k0); ld = src
 // row stride).  Reads float4 along k (coalesced 128B per 8 lanes), writes
 // columns; the +4 row pad makes the column writes conflict-free.
+template <int NC, bool EDGE>
 __device__ __forceinline__ void stage_transpose(float* dst, const float* src,
                                                 int64_t ld, int kmax, int cmax) {
+  constexpr int LDT = NC + 4;
   const int k4 = (threadIdx.x & 7) * 4;  // 0..28
   const int c0 = threadIdx.x >> 3;       // 0..31
 #pragma unroll
-  for (int it = 0; it < 4; ++it) {
+  for (int it = 0; it < NC / 32; ++it) {
     const int c = c0 + it * 32;
     float4 v = {0.f, 0.f, 0.f, 0.f};
-    if (c < cmax) {
-      if (k4 + 3 < kmax) {
-        v = *reinterpret_cast<const float4*>(src + (int64_t)c * ld + k4);
-      } else {
-        if (k4 + 0 < kmax) v.x = src[(int64_t)c * ld + k4 + 0];
-        if (k4 + 1 < kmax) v.y = src[(int64_t)c * ld + k4 + 1];
-        if (k4 + 2 < kmax) v.z = src[(int64_t)c * ld + k4 + 2];
-        if (k4 + 3 < kmax) v.w = src[(int64_t)c * ld + k4 + 3];
-      }
+    if (!EDGE || (c < cmax && k4 + 3 < kmax)) {
+      v = *reinterpret_cast<const float4*>(src + (int64_t)c * ld + k4);
+    } else if (c < cmax) {
+      if (k4 + 0 < kmax) v.x = src[(int64_t)c * ld + k4 + 0];
+      if (k4 + 1 < kmax) v.y = src[(int64_t)c * ld + k4 + 1];
+      if (k4 + 2 < kmax) v.z = src[(int64_t)c * ld + k4 + 2];
+      if (k4 + 3 < kmax) v.w = src[(int64_t)c * ld + k4 + 3];
     }
-    dst[(k4 + 0) * GEMM_LDT + c] = v.x;
-    dst[(k4 + 1) * GEMM_LDT + c] = v.y;
-    dst[(k4 + 2) * GEMM_LDT + c] = v.z;
-    dst[(k4 + 3) * GEMM_LDT + c] = v.w;
+    dst[(k4 + 0) * LDT + c] = v.x;
+    dst[(k4 + 1) * LDT + c] = v.y;
+    dst[(k4 + 2) * LDT + c] = v.z;
+    dst[(k4 + 3) * LDT + c] = v.w;
   }
 }
 
-template <bool TA, bool TB, bool ATOMIC>
+// BM=128 always; BN ∈ {64,128}: 4 waves in 2x2, each owning 64 x (BN/2)
+// output = 2 x (BN/64) fragments of 32x32.
+template <bool TA, bool TB, bool ATOMIC, bool EDGE, int BN>
 __global__ __launch_bounds__(256, 4) void k_gemm_f32(
     const float* __restrict__ A, const float* __restrict__ B,
     float* __restrict__ C, const float* __restrict__ bias,
     const float* __restrict__ R, int M, int N, int K, int64_t lda, int64_t ldb,
     int64_t ldc, int64_t sA1, int64_t sA2, int64_t sB1, int64_t sB2,
     int64_t sC1, int64_t sC2, int n2, float alpha, float beta, int nbn) {
+  constexpr int FN = BN / 64;  // B fragments per wave (1 or 2)
+  constexpr int LDB_T = BN + 4;
   __shared__ float As[GEMM_BK * GEMM_LDT];
-  __shared__ float Bs[GEMM_BK * GEMM_LDT];
+  __shared__ float Bs[GEMM_BK * LDB_T];
 
   const int tile = blockIdx.x;
   const int bm = tile / nbn, bn = tile % nbn;
-  const int m0 = bm * GEMM_BM, n0 = bn * GEMM_BN;
+  const int m0 = bm * GEMM_BM, n0 = bn * BN;
 
   const int z = blockIdx.z;
   const int i1 = z / n2, i2 = z % n2;
@@ -183,42 +195,45 @@ __global__ __launch_bounds__(256, 4) void k_gemm_f32(
   for (int kt = kbeg; kt < kend; kt += GEMM_BK) {
     const int kmax = kend - kt;
     if (TA)
-      stage_direct(As, A + (int64_t)kt * lda + m0, lda, min(kmax, GEMM_BK),
-                   min(M - m0, GEMM_BM));
+      stage_direct<GEMM_BM, EDGE>(As, A + (int64_t)kt * lda + m0, lda,
+                                  min(kmax, GEMM_BK), min(M - m0, GEMM_BM));
     else
-      stage_transpose(As, A + (int64_t)m0 * lda + kt, lda, min(kmax, GEMM_BK),
-                      min(M - m0, GEMM_BM));
+      stage_transpose<GEMM_BM, EDGE>(As, A + (int64_t)m0 * lda + kt, lda,
+                                     min(kmax, GEMM_BK), min(M - m0, GEMM_BM));
     if (TB)
-      stage_transpose(Bs, B + (int64_t)n0 * ldb + kt, ldb, min(kmax, GEMM_BK),
-                      min(N - n0, GEMM_BN));
+      stage_transpose<BN, EDGE>(Bs, B + (int64_t)n0 * ldb + kt, ldb,
+                                min(kmax, GEMM_BK), min(N - n0, BN));
     else
-      stage_direct(Bs, B + (int64_t)kt * ldb + n0, ldb, min(kmax, GEMM_BK),
-                   min(N - n0, GEMM_BN));
+      stage_direct<BN, EDGE>(Bs, B + (int64_t)kt * ldb + n0, ldb,
+                             min(kmax, GEMM_BK), min(N - n0, BN));
     __syncthreads();
 
 #pragma unroll
     for (int kk = 0; kk < GEMM_BK / 2; ++kk) {
       const float* ar = As + (kk * 2 + kh) * GEMM_LDT + wr * 64;
-      const float* br = Bs + (kk * 2 + kh) * GEMM_LDT + wc * 64;
+      const float* br = Bs + (kk * 2 + kh) * LDB_T + wc * (BN / 2);
       const float a0 = ar[il], a1 = ar[32 + il];
-      const float b0 = br[il], b1 = br[32 + il];
+      const float b0 = br[il];
       acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
-      acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
       acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
-      acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
+      if (FN == 2) {
+        const float b1 = br[32 + il];
+        acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
+        acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
+      }
     }
     __syncthreads();
   }
 
   // epilogue
-  const int mw = m0 + wr * 64, nw = n0 + wc * 64;
+  const int mw = m0 + wr * 64, nw = n0 + wc * (BN / 2);
 #define OB_EPI(ACC, TI, TJ)                                                   \
   {                                                                           \
     const int nn = nw + (TJ)*32 + il;                                         \
-    if (nn < N) {                                                             \
+    if (!EDGE || nn < N) {                                                    \
       _Pragma("unroll") for (int r = 0; r < 16; ++r) {                        \
         const int mm = mw + (TI)*32 + (r & 3) + 8 * (r >> 2) + 4 * kh;        \
-        if (mm < M) {                                                         \
+        if (!EDGE || mm < M) {                                                \
           float v = alpha * ACC[r];                                           \
           if (ATOMIC) {                                                       \
             atomicAdd(&C[(int64_t)mm * ldc + nn], v);                         \
@@ -233,9 +248,11 @@ __global__ __launch_bounds__(256, 4) void k_gemm_f32(
     }                                                                         \
   }
   OB_EPI(acc00, 0, 0)
-  OB_EPI(acc01, 0, 1)
   OB_EPI(acc10, 1, 0)
-  OB_EPI(acc11, 1, 1)
+  if (FN == 2) {
+    OB_EPI(acc01, 0, 1)
+    OB_EPI(acc11, 1, 1)
+  }
 #undef OB_EPI
 }
 
@@ -253,28 +270,42 @@ extern "C" int ob_gemm_f32(int transA, int transB, int64_t M, int64_t N,
     return ob_fail("gemm: splitk>1 requires atomic stores");
   if (atomic && (bias || residual))
     return ob_fail("gemm: atomic epilogue excludes bias/residual");
+  const int BN = (N <= 64) ? 64 : 128;  // narrow tiles for head_dim GEMMs
   const int nbm = (int)((M + GEMM_BM - 1) / GEMM_BM);
-  const int nbn = (int)((N + GEMM_BN - 1) / GEMM_BN);
+  const int nbn = (int)((N + BN - 1) / BN);
+  // guard-free interior variant when no tile has a tail anywhere
+  const bool edge = (M % GEMM_BM) || (N % BN) || (K % GEMM_BK);
   dim3 grid(nbm * nbn, splitk, (unsigned)(n1 * n2));
   dim3 block(256);
-#define OB_GEMM_LAUNCH(TA_, TB_, AT_)                                       \
-  k_gemm_f32<TA_, TB_, AT_><<<grid, block, 0, S(stream)>>>(                 \
+#define OB_GEMM_LAUNCH4(TA_, TB_, AT_, ED_, BN_)                            \
+  k_gemm_f32<TA_, TB_, AT_, ED_, BN_><<<grid, block, 0, S(stream)>>>(       \
       (const float*)A, (const float*)B, (float*)C, (const float*)bias,      \
       (const float*)residual, (int)M, (int)N, (int)K, lda, ldb, ldc,        \
       strideA1, strideA2, strideB1, strideB2, strideC1, strideC2, (int)n2,  \
       alpha, beta, nbn)
+#define OB_GEMM_LAUNCH2(TA_, TB_, AT_)                                      \
+  do {                                                                      \
+    if (BN == 64) {                                                         \
+      if (edge) OB_GEMM_LAUNCH4(TA_, TB_, AT_, true, 64);                   \
+      else OB_GEMM_LAUNCH4(TA_, TB_, AT_, false, 64);                       \
+    } else {                                                                \
+      if (edge) OB_GEMM_LAUNCH4(TA_, TB_, AT_, true, 128);                  \
+      else OB_GEMM_LAUNCH4(TA_, TB_, AT_, false, 128);                      \
+    }                                                                       \
+  } while (0)
   const int sel = (transA ? 4 : 0) | (transB ? 2 : 0) | (atomic ? 1 : 0);
   switch (sel) {
-    case 0: OB_GEMM_LAUNCH(false, false, false); break;
-    case 1: OB_GEMM_LAUNCH(false, false, true); break;
-    case 2: OB_GEMM_LAUNCH(false, true, false); break;
-    case 3: OB_GEMM_LAUNCH(false, true, true); break;
-    case 4: OB_GEMM_LAUNCH(true, false, false); break;
-    case 5: OB_GEMM_LAUNCH(true, false, true); break;
-    case 6: OB_GEMM_LAUNCH(true, true, false); break;
-    case 7: OB_GEMM_LAUNCH(true, true, true); break;
+    case 0: OB_GEMM_LAUNCH2(false, false, false); break;
+    case 1: OB_GEMM_LAUNCH2(false, false, true); break;
+    case 2: OB_GEMM_LAUNCH2(false, true, false); break;
+    case 3: OB_GEMM_LAUNCH2(false, true, true); break;
+    case 4: OB_GEMM_LAUNCH2(true, false, false); break;
+    case 5: OB_GEMM_LAUNCH2(true, false, true); break;
+    case 6: OB_GEMM_LAUNCH2(true, true, false); break;
+    case 7: OB_GEMM_LAUNCH2(true, true, true); break;
   }
-#undef OB_GEMM_LAUNCH
+#undef OB_GEMM_LAUNCH2
+#undef OB_GEMM_LAUNCH4
   OB_LAUNCH_CHECK();
   return 0;
 }
