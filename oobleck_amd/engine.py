@@ -28,27 +28,25 @@ def even_stage_split(cfg: ModelConfig, n_stages: int) -> list[list[int]]:
     """Contiguous, parameter-balanced split of the model's layers into
     n_stages (stand-in for the C++ planner's stage assignment; the planner
     itself is out of scope this round — SURVEY.md §8 f2)."""
+    import bisect
+
     from .params import layer_param_numel
     L = cfg.n_layers_total
-    assert n_stages <= L
+    assert 1 <= n_stages <= L
     weights = [layer_param_numel(cfg, cfg.layer_kind(i)) for i in range(L)]
-    total = sum(weights)
-    target = total / n_stages
-    stages, cur, acc = [], [], 0.0
-    remaining_stages = n_stages
-    for lid in range(L):
-        cur.append(lid)
-        acc += weights[lid]
-        remaining = L - lid - 1
-        if (acc >= target and remaining_stages > 1 and remaining >= remaining_stages - 1):
-            stages.append(cur)
-            cur, acc = [], 0.0
-            remaining_stages -= 1
-    if cur:
-        stages.append(cur)
-    while len(stages) < n_stages:  # degenerate tiny models
-        stages.append([stages[-1].pop()])
-    return stages
+    prefix = [0]
+    for w in weights:
+        prefix.append(prefix[-1] + w)
+    bounds = [0]
+    for i in range(1, n_stages):
+        target = prefix[-1] * i / n_stages
+        pos = bisect.bisect_left(prefix, target)
+        if pos > 0 and abs(prefix[pos - 1] - target) < abs(prefix[pos] - target):
+            pos -= 1
+        pos = max(bounds[-1] + 1, min(pos, L - (n_stages - i)))
+        bounds.append(pos)
+    bounds.append(L)
+    return [list(range(bounds[i], bounds[i + 1])) for i in range(n_stages)]
 
 
 class DataParallelEngine:

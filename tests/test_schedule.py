@@ -136,3 +136,15 @@ def test_two_stage_interleave_known_good():
         ["BackwardPass"],                                      # B1
         ["SendGrad"],
     ]
+
+
+def test_even_stage_split():
+    from oobleck_amd.config import GPT2_SMALL, GPT2_XL
+    from oobleck_amd.engine import even_stage_split
+    for cfg in (GPT2_SMALL, GPT2_XL):
+        for n in (1, 2, 4, 7):
+            st = even_stage_split(cfg, n)
+            assert len(st) == n
+            flat = [l for s in st for l in s]
+            assert flat == list(range(cfg.n_layers_total))
+            assert all(len(s) >= 1 for s in st)
