@@ -30,10 +30,22 @@ def even_stage_split(cfg: ModelConfig, n_stages: int) -> list[list[int]]:
     itself is out of scope this round — SURVEY.md §8 f2)."""
     import bisect
 
-    from .params import layer_param_numel
+    from .params import KIND_BLOCK, KIND_FINAL
     L = cfg.n_layers_total
     assert 1 <= n_stages <= L
-    weights = [layer_param_numel(cfg, cfg.layer_kind(i)) for i in range(L)]
+    # balance by compute (matmul FLOPs/token ∝ matmul params), not by
+    # parameter count: the embedding layer is a gather (≈free) while the
+    # final layer's lm_head GEMM costs ≈ V*H/(12*H^2) blocks of work.
+    H, V = cfg.n_embd, cfg.vocab_size
+
+    def w(lid):
+        kind = cfg.layer_kind(lid)
+        if kind == KIND_BLOCK:
+            return 12 * H * H
+        if kind == KIND_FINAL:
+            return V * H
+        return H  # embedding: negligible compute
+    weights = [w(i) for i in range(L)]
     prefix = [0]
     for w in weights:
         prefix.append(prefix[-1] + w)
