@@ -153,6 +153,95 @@ __device__ __forceinline__ void stage_transpose(float* dst, const float* src,
   }
 }
 
+// split load/write staging for the 2-phase (double-buffered) pipeline:
+// loads issue at the top of iteration t for tile t+1 and land under the
+// MFMAs; the write into the other LDS buffer happens after the MFMAs, one
+// barrier per K-tile (T14 async-STAGE split, cdna_hip_programming.md G15).
+template <int NC, bool EDGE>
+struct StageRegs {
+  float4 v[NC / 32];
+};
+
+template <int NC, bool EDGE>
+__device__ __forceinline__ void stage_load_direct(StageRegs<NC, EDGE>& r,
+                                                  const float* src, int64_t ld,
+                                                  int kmax, int cmax) {
+  const int c4 = (threadIdx.x % (NC / 4)) * 4;
+  const int k0 = threadIdx.x / (NC / 4);
+  constexpr int KSTEP = 1024 / NC;
+#pragma unroll
+  for (int it = 0; it < NC / 32; ++it) {
+    const int k = k0 + it * KSTEP;
+    if (EDGE) {
+      float4 v = {0.f, 0.f, 0.f, 0.f};
+      if (k < kmax) {
+        if (c4 + 3 < cmax) {
+          v = *reinterpret_cast<const float4*>(src + (int64_t)k * ld + c4);
+        } else {
+          if (c4 + 0 < cmax) v.x = src[(int64_t)k * ld + c4 + 0];
+          if (c4 + 1 < cmax) v.y = src[(int64_t)k * ld + c4 + 1];
+          if (c4 + 2 < cmax) v.z = src[(int64_t)k * ld + c4 + 2];
+          if (c4 + 3 < cmax) v.w = src[(int64_t)k * ld + c4 + 3];
+        }
+      }
+      r.v[it] = v;
+    } else {
+      r.v[it] = *reinterpret_cast<const float4*>(src + (int64_t)k * ld + c4);
+    }
+  }
+}
+
+template <int NC, bool EDGE>
+__device__ __forceinline__ void stage_write_direct(const StageRegs<NC, EDGE>& r,
+                                                   float* dst) {
+  constexpr int LDT = NC + 4;
+  const int c4 = (threadIdx.x % (NC / 4)) * 4;
+  const int k0 = threadIdx.x / (NC / 4);
+  constexpr int KSTEP = 1024 / NC;
+#pragma unroll
+  for (int it = 0; it < NC / 32; ++it)
+    *reinterpret_cast<float4*>(dst + (k0 + it * KSTEP) * LDT + c4) = r.v[it];
+}
+
+template <int NC, bool EDGE>
+__device__ __forceinline__ void stage_load_transpose(StageRegs<NC, EDGE>& r,
+                                                     const float* src,
+                                                     int64_t ld, int kmax,
+                                                     int cmax) {
+  const int k4 = (threadIdx.x & 7) * 4;
+  const int c0 = threadIdx.x >> 3;
+#pragma unroll
+  for (int it = 0; it < NC / 32; ++it) {
+    const int c = c0 + it * 32;
+    float4 v = {0.f, 0.f, 0.f, 0.f};
+    if (!EDGE || (c < cmax && k4 + 3 < kmax)) {
+      v = *reinterpret_cast<const float4*>(src + (int64_t)c * ld + k4);
+    } else if (c < cmax) {
+      if (k4 + 0 < kmax) v.x = src[(int64_t)c * ld + k4 + 0];
+      if (k4 + 1 < kmax) v.y = src[(int64_t)c * ld + k4 + 1];
+      if (k4 + 2 < kmax) v.z = src[(int64_t)c * ld + k4 + 2];
+      if (k4 + 3 < kmax) v.w = src[(int64_t)c * ld + k4 + 3];
+    }
+    r.v[it] = v;
+  }
+}
+
+template <int NC, bool EDGE>
+__device__ __forceinline__ void stage_write_transpose(
+    const StageRegs<NC, EDGE>& r, float* dst) {
+  constexpr int LDT = NC + 4;
+  const int k4 = (threadIdx.x & 7) * 4;
+  const int c0 = threadIdx.x >> 3;
+#pragma unroll
+  for (int it = 0; it < NC / 32; ++it) {
+    const int c = c0 + it * 32;
+    dst[(k4 + 0) * LDT + c] = r.v[it].x;
+    dst[(k4 + 1) * LDT + c] = r.v[it].y;
+    dst[(k4 + 2) * LDT + c] = r.v[it].z;
+    dst[(k4 + 3) * LDT + c] = r.v[it].w;
+  }
+}
+
 // BM=128 always; BN ∈ {64,128}: 4 waves in 2x2, each owning 64 x (BN/2)
 // output = 2 x (BN/64) fragments of 32x32.
 template <bool TA, bool TB, bool ATOMIC, bool EDGE, int BN>
@@ -256,6 +345,141 @@ __global__ __launch_bounds__(256, 4) void k_gemm_f32(
 #undef OB_EPI
 }
 
+// 2-phase double-buffered variant: loads for tile t+1 issue before the
+// MFMAs of tile t and land under them; LDS write + one barrier per tile.
+template <bool TA, bool TB, bool ATOMIC, bool EDGE, int BN>
+__global__ __launch_bounds__(256, 2) void k_gemm_f32_p2(
+    const float* __restrict__ A, const float* __restrict__ B,
+    float* __restrict__ C, const float* __restrict__ bias,
+    const float* __restrict__ R, int M, int N, int K, int64_t lda, int64_t ldb,
+    int64_t ldc, int64_t sA1, int64_t sA2, int64_t sB1, int64_t sB2,
+    int64_t sC1, int64_t sC2, int n2, float alpha, float beta, int nbn) {
+  constexpr int FN = BN / 64;
+  constexpr int LDB_T = BN + 4;
+  __shared__ float As[2][GEMM_BK * GEMM_LDT];
+  __shared__ float Bs[2][GEMM_BK * LDB_T];
+
+  const int tile = blockIdx.x;
+  const int bm = tile / nbn, bn = tile % nbn;
+  const int m0 = bm * GEMM_BM, n0 = bn * BN;
+
+  const int z = blockIdx.z;
+  const int i1 = z / n2, i2 = z % n2;
+  A += (int64_t)i1 * sA1 + (int64_t)i2 * sA2;
+  B += (int64_t)i1 * sB1 + (int64_t)i2 * sB2;
+  C += (int64_t)i1 * sC1 + (int64_t)i2 * sC2;
+  if (R) R += (int64_t)i1 * sC1 + (int64_t)i2 * sC2;
+
+  const int splitk = gridDim.y;
+  const int kchunk = ((K + splitk * GEMM_BK - 1) / (splitk * GEMM_BK)) * GEMM_BK;
+  const int kbeg = blockIdx.y * kchunk;
+  const int kend = min(K, kbeg + kchunk);
+  if (kbeg >= kend) return;
+
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int wr = w >> 1, wc = w & 1;
+  const int il = lane & 31, kh = lane >> 5;
+
+#define OB_P2_LOAD(KT)                                                        \
+  {                                                                           \
+    const int kmax_ = min(kend - (KT), GEMM_BK);                              \
+    if (TA)                                                                   \
+      stage_load_direct<GEMM_BM, EDGE>(ra, A + (int64_t)(KT)*lda + m0, lda,   \
+                                       kmax_, min(M - m0, GEMM_BM));          \
+    else                                                                      \
+      stage_load_transpose<GEMM_BM, EDGE>(ra, A + (int64_t)m0 * lda + (KT),   \
+                                          lda, kmax_, min(M - m0, GEMM_BM));  \
+    if (TB)                                                                   \
+      stage_load_transpose<BN, EDGE>(rb, B + (int64_t)n0 * ldb + (KT), ldb,   \
+                                     kmax_, min(N - n0, BN));                 \
+    else                                                                      \
+      stage_load_direct<BN, EDGE>(rb, B + (int64_t)(KT)*ldb + n0, ldb, kmax_, \
+                                  min(N - n0, BN));                           \
+  }
+#define OB_P2_WRITE(BUF)                                                      \
+  {                                                                           \
+    if (TA)                                                                   \
+      stage_write_direct<GEMM_BM, EDGE>(ra, As[BUF]);                         \
+    else                                                                      \
+      stage_write_transpose<GEMM_BM, EDGE>(ra, As[BUF]);                      \
+    if (TB)                                                                   \
+      stage_write_transpose<BN, EDGE>(rb, Bs[BUF]);                           \
+    else                                                                      \
+      stage_write_direct<BN, EDGE>(rb, Bs[BUF]);                              \
+  }
+
+  f32x16 acc00 = {}, acc01 = {}, acc10 = {}, acc11 = {};
+
+  {
+    StageRegs<GEMM_BM, EDGE> ra;
+    StageRegs<BN, EDGE> rb;
+    OB_P2_LOAD(kbeg)
+    OB_P2_WRITE(0)
+  }
+  __syncthreads();
+
+#define OB_P2_MFMA(BUF)                                                       \
+  _Pragma("unroll") for (int kk = 0; kk < GEMM_BK / 2; ++kk) {                \
+    const float* ar = As[BUF] + (kk * 2 + kh) * GEMM_LDT + wr * 64;           \
+    const float* br = Bs[BUF] + (kk * 2 + kh) * LDB_T + wc * (BN / 2);        \
+    const float a0 = ar[il], a1 = ar[32 + il];                                \
+    const float b0 = br[il];                                                  \
+    acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);     \
+    acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);     \
+    if (FN == 2) {                                                            \
+      const float b1 = br[32 + il];                                           \
+      acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);   \
+      acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);   \
+    }                                                                         \
+  }
+
+  // steady state is branch-free: the last tile is peeled out of the loop.
+  int cur = 0;
+  for (int kt = kbeg; kt + GEMM_BK < kend; kt += GEMM_BK) {
+    StageRegs<GEMM_BM, EDGE> ra;
+    StageRegs<BN, EDGE> rb;
+    OB_P2_LOAD(kt + GEMM_BK)
+    OB_P2_MFMA(cur)
+    OB_P2_WRITE(cur ^ 1)
+    __syncthreads();
+    cur ^= 1;
+  }
+  OB_P2_MFMA(cur)
+#undef OB_P2_LOAD
+#undef OB_P2_WRITE
+#undef OB_P2_MFMA
+
+  const int mw = m0 + wr * 64, nw = n0 + wc * (BN / 2);
+#define OB_EPI(ACC, TI, TJ)                                                   \
+  {                                                                           \
+    const int nn = nw + (TJ)*32 + il;                                         \
+    if (!EDGE || nn < N) {                                                    \
+      _Pragma("unroll") for (int r = 0; r < 16; ++r) {                        \
+        const int mm = mw + (TI)*32 + (r & 3) + 8 * (r >> 2) + 4 * kh;        \
+        if (!EDGE || mm < M) {                                                \
+          float v = alpha * ACC[r];                                           \
+          if (ATOMIC) {                                                       \
+            atomicAdd(&C[(int64_t)mm * ldc + nn], v);                         \
+          } else {                                                            \
+            if (bias) v += bias[nn];                                          \
+            if (R) v += R[(int64_t)mm * ldc + nn];                            \
+            if (beta != 0.f) v += beta * C[(int64_t)mm * ldc + nn];           \
+            C[(int64_t)mm * ldc + nn] = v;                                    \
+          }                                                                   \
+        }                                                                     \
+      }                                                                       \
+    }                                                                         \
+  }
+  OB_EPI(acc00, 0, 0)
+  OB_EPI(acc10, 1, 0)
+  if (FN == 2) {
+    OB_EPI(acc01, 0, 1)
+    OB_EPI(acc11, 1, 1)
+  }
+#undef OB_EPI
+}
+
 extern "C" int ob_gemm_f32(int transA, int transB, int64_t M, int64_t N,
                            int64_t K, float alpha, const void* A, int64_t lda,
                            int64_t strideA1, int64_t strideA2, const void* B,
@@ -275,14 +499,30 @@ extern "C" int ob_gemm_f32(int transA, int transB, int64_t M, int64_t N,
   const int nbn = (int)((N + BN - 1) / BN);
   // guard-free interior variant when no tile has a tail anywhere
   const bool edge = (M % GEMM_BM) || (N % BN) || (K % GEMM_BK);
+  // 2-phase double-buffered pipeline when the K loop is deep enough to
+  // overlap (ob_GEMM_V1=1 in the environment falls back, for A/B runs).
+  static const bool force_v1 = [] {
+    const char* e = getenv("OB_GEMM_V1");
+    return e && e[0] == '1';
+  }();
+  const bool p2 = !force_v1 && K >= 2 * GEMM_BK;
   dim3 grid(nbm * nbn, splitk, (unsigned)(n1 * n2));
   dim3 block(256);
 #define OB_GEMM_LAUNCH4(TA_, TB_, AT_, ED_, BN_)                            \
-  k_gemm_f32<TA_, TB_, AT_, ED_, BN_><<<grid, block, 0, S(stream)>>>(       \
-      (const float*)A, (const float*)B, (float*)C, (const float*)bias,      \
-      (const float*)residual, (int)M, (int)N, (int)K, lda, ldb, ldc,        \
-      strideA1, strideA2, strideB1, strideB2, strideC1, strideC2, (int)n2,  \
-      alpha, beta, nbn)
+  do {                                                                      \
+    if (p2)                                                                 \
+      k_gemm_f32_p2<TA_, TB_, AT_, ED_, BN_><<<grid, block, 0, S(stream)>>>(\
+          (const float*)A, (const float*)B, (float*)C, (const float*)bias,  \
+          (const float*)residual, (int)M, (int)N, (int)K, lda, ldb, ldc,    \
+          strideA1, strideA2, strideB1, strideB2, strideC1, strideC2,       \
+          (int)n2, alpha, beta, nbn);                                       \
+    else                                                                    \
+      k_gemm_f32<TA_, TB_, AT_, ED_, BN_><<<grid, block, 0, S(stream)>>>(   \
+          (const float*)A, (const float*)B, (float*)C, (const float*)bias,  \
+          (const float*)residual, (int)M, (int)N, (int)K, lda, ldb, ldc,    \
+          strideA1, strideA2, strideB1, strideB2, strideC1, strideC2,       \
+          (int)n2, alpha, beta, nbn);                                       \
+  } while (0)
 #define OB_GEMM_LAUNCH2(TA_, TB_, AT_)                                      \
   do {                                                                      \
     if (BN == 64) {                                                         \
