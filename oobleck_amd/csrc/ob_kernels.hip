@@ -505,7 +505,9 @@ extern "C" int ob_gemm_f32(int transA, int transB, int64_t M, int64_t N,
     const char* e = getenv("OB_GEMM_V1");
     return e && e[0] == '1';
   }();
-  const bool p2 = !force_v1 && K >= 2 * GEMM_BK;
+  // P2 wins on interior shapes (+3..27%) but loses on edge shapes
+  // (lm_head N=50257: 113.7->103.1 TF measured) - gate on !edge.
+  const bool p2 = !force_v1 && !edge && K >= 2 * GEMM_BK;
   dim3 grid(nbm * nbn, splitk, (unsigned)(n1 * n2));
   dim3 block(256);
 #define OB_GEMM_LAUNCH4(TA_, TB_, AT_, ED_, BN_)                            \

@@ -32,6 +32,10 @@ SHAPES = [
     ("attn_scores NT b96", dict(M=S, N=S, K=hd, transB=1, batch=B * nh)),
     ("attn_PV NN b96", dict(M=S, N=hd, K=S, batch=B * nh)),
     ("attn_dV TN b96", dict(M=S, N=hd, K=S, transA=1, batch=B * nh)),
+    # diagnostics: the guide's fp32 reference point (122 TF untuned @4096^3)
+    ("diag_4096cubed NN", dict(M=4096, N=4096, K=4096)),
+    ("diag_sqK3072 NN", dict(M=8192, N=3072, K=3072)),
+    ("diag_xl_fc NN", dict(M=2048, N=6400, K=1600)),
 ]
 
 
