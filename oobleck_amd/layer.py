@@ -39,9 +39,10 @@ def _ptr(t: torch.Tensor | None) -> ctypes.c_void_p:
 
 
 class Layer:
-    """One fx-shard (embedding / block / ln_f+lm_head+loss) on one GPU,
-    NO_SHARD strategy (1 GPU per stage replica — the graded configs;
-    FULL_SHARD intra-stage sharding is §8 f1, next)."""
+    """One fx-shard (embedding / block / ln_f+lm_head+loss).  NO_SHARD when
+    the stage's per-layer process group has 1 rank; FULL_SHARD (per-rank
+    parameter shards, all-gather on unshard, per-microbatch reduce-scatter
+    of grads — reference layer.py:96-111 chooses by group size) otherwise."""
 
     def __init__(self, layer_id: int, cfg: ModelConfig, max_batch: int,
                  seq_len: int, n_slots: int, device: torch.device,
@@ -58,14 +59,31 @@ class Layer:
             max_batch=max_batch, seq_len=seq_len, n_slots=n_slots)
         n = ext.ob_layer_param_count(ctypes.byref(self._desc))
         assert n > 0
-        flat = init_layer_params(cfg, layer_id, seed, init_style).to(device)
-        grad = torch.zeros(n, dtype=torch.float32, device=device)
+        init = init_layer_params(cfg, layer_id, seed, init_style).to(device)
+        group_size = 1
+        if process_group is not None and torch.distributed.is_initialized():
+            group_size = torch.distributed.get_world_size(process_group)
         handle = ctypes.c_void_p()
         check(ext.ob_layer_create(ctypes.byref(self._desc),
                                   ctypes.byref(handle)), "layer_create")
         self._h = handle
-        check(ext.ob_layer_bind(self._h, _ptr(flat), _ptr(grad)), "layer_bind")
-        self._param_handle = _FlatParamHandle(flat, grad)
+        self._sharded = None
+        if group_size > 1:
+            from .sharding import ShardedFlatParam
+            self._sharded = ShardedFlatParam(n, process_group, device,
+                                             init_full=init)
+            del init
+            check(ext.ob_layer_bind(self._h, _ptr(self._sharded.full),
+                                    _ptr(self._sharded.full_grad)),
+                  "layer_bind")
+            self._param_handle = _FlatParamHandle(self._sharded.shard,
+                                                  self._sharded.shard.grad)
+        else:
+            flat = init
+            grad = torch.zeros(n, dtype=torch.float32, device=device)
+            check(ext.ob_layer_bind(self._h, _ptr(flat), _ptr(grad)),
+                  "layer_bind")
+            self._param_handle = _FlatParamHandle(flat, grad)
         self._batch = max_batch
 
     # -- compute ------------------------------------------------------------
@@ -84,18 +102,27 @@ class Layer:
 
     def forward_slot(self, slot: int, x: torch.Tensor, out: torch.Tensor,
                      labels: torch.Tensor | None = None) -> None:
+        if self._sharded is not None:
+            self._sharded.unshard()  # pre_forward_hook (layer.py:147-153)
         check(get_ext().ob_layer_forward(self._h, slot, _ptr(x), _ptr(out),
                                          _ptr(labels), _stream_ptr()),
               f"forward layer {self.layer_id}")
 
     def backward_slot(self, slot: int, dout: torch.Tensor | None,
                       din: torch.Tensor | None) -> None:
+        if self._sharded is not None:
+            self._sharded.unshard()  # pre_backward_hook (layer.py:160-166)
         check(get_ext().ob_layer_backward(self._h, slot, _ptr(dout), _ptr(din),
                                           _stream_ptr()),
               f"backward layer {self.layer_id}")
+        if self._sharded is not None:
+            # post_backward_hook: reduce-scatter + accumulate (layer.py:167-225)
+            self._sharded.reduce_scatter_grad()
 
     def zero_grads(self) -> None:
         self.flat_grad.zero_()
+        if self._sharded is not None:
+            self._sharded.full_grad.zero_()
 
     # -- distributed surface (reference layer.py:272-291) --------------------
     def _shard_param(self, tensor: torch.Tensor, number: int) -> list[torch.Tensor]:

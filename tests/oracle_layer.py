@@ -13,12 +13,24 @@ from oracle.gpt2_oracle import (KIND_FINAL, OracleConfig, layer_forward)
 
 class OracleLayer:
     def __init__(self, layer_id: int, cfg: OracleConfig, flat: torch.Tensor,
-                 n_slots: int = 4):
+                 n_slots: int = 4, process_group=None):
         self.layer_id = layer_id
         self.cfg = cfg
         self.kind = cfg.layer_kind(layer_id)
-        self.flat_param = flat.clone()
-        self.flat_param.grad = torch.zeros_like(flat)
+        self._n = flat.numel()
+        self._sharded = None
+        group_size = 1
+        if process_group is not None and torch.distributed.is_initialized():
+            group_size = torch.distributed.get_world_size(process_group)
+        if group_size > 1:
+            from oobleck_amd.sharding import ShardedFlatParam
+            self._sharded = ShardedFlatParam(self._n, process_group,
+                                             torch.device("cpu"),
+                                             init_full=flat)
+            self.flat_param = self._sharded.shard
+        else:
+            self.flat_param = flat.clone()
+            self.flat_param.grad = torch.zeros_like(flat)
         self._batch = 0
         self._saved: dict[int, tuple] = {}
 
@@ -31,10 +43,18 @@ class OracleLayer:
 
     def zero_grads(self) -> None:
         self.flat_grad.zero_()
+        if self._sharded is not None:
+            self._sharded.full_grad.zero_()
+
+    def _full_params(self) -> torch.Tensor:
+        if self._sharded is not None:
+            self._sharded.unshard()
+            return self._sharded.full[:self._n]
+        return self.flat_param
 
     def forward_slot(self, slot: int, x: torch.Tensor, out: torch.Tensor,
                      labels: torch.Tensor | None = None) -> None:
-        flat = self.flat_param.detach().clone().requires_grad_(True)
+        flat = self._full_params().detach().clone().requires_grad_(True)
         xi = x.detach().clone()
         if xi.is_floating_point():
             xi.requires_grad_(True)
@@ -50,7 +70,11 @@ class OracleLayer:
             y.backward()
         else:
             torch.autograd.backward(y, dout.reshape(y.shape))
-        self.flat_param.grad += flat.grad
+        if self._sharded is not None:
+            self._sharded.full_grad[:self._n] += flat.grad
+            self._sharded.reduce_scatter_grad()
+        else:
+            self.flat_param.grad += flat.grad
         if din is not None and xi.is_floating_point():
             din.copy_(xi.grad)
 

@@ -201,8 +201,59 @@ def _run_copy(rank: int, world: int, tmp: str):
 
 
 # ---------------------------------------------------------------------------
+# FULL_SHARD intra-stage sharding (§8 f1): one stage on 2 fsdp ranks — both
+# ranks hold parameter shards, all-gather on unshard, per-microbatch
+# reduce-scatter of grads.  Parity vs the single-process oracle.
+# ---------------------------------------------------------------------------
 
-@pytest.mark.parametrize("target", [_run_2stage, _run_dp, _run_copy],
-                         ids=["2stage_parity", "dp_allreduce", "layer_copy"])
+def _run_fsdp(rank: int, world: int, tmp: str):
+    _setup(rank, world, tmp)
+    from oobleck_amd.engine import DataParallelEngine, make_rank_grid
+    from oobleck_amd.pipeline import OobleckPipeline
+    from tests.oracle_layer import NoOpOptimizer, OracleLayer
+
+    mc, oc, tc = _configs()
+    flats = _flats(oc)
+    L = oc.n_layers_total
+    grid = make_rank_grid(L, [list(range(L))], [[0, 1]])
+
+    class Loader:
+        def __iter__(self):
+            return iter({"input_ids": i, "labels": l}
+                        for i, l in _batches(oc, MB, seed=11))
+
+    pipe = OobleckPipeline(0, grid, mc, tc, Loader(), MB, torch.device("cpu"))
+    pipe.initialize_distributed_fsdp()
+    pipe.initialize_distributed_pipeline()
+    pipe.initialize_execution(
+        layer_factory=lambda lid, pg, n_slots: OracleLayer(
+            lid, oc, flats[lid], process_group=pg),
+        optimizer_factory=lambda layers: (NoOpOptimizer(layers), None))
+    dp = DataParallelEngine([pipe])
+    pipe.train()
+    dp.do_allreduce(pipe)
+
+    losses_ref, grads_ref = _reference_grads(oc, flats, _batches(oc, MB, seed=11))
+    total_ref = sum(l.item() for l in losses_ref)
+    assert abs(pipe.execution.total_loss.item() - total_ref) < 1e-4 * abs(total_ref)
+    for layer in pipe.execution._layers:
+        sh = layer._sharded
+        full_ref = torch.zeros(sh.padded)
+        # reference-faithful semantics: every fsdp rank of a stage consumes
+        # the SAME microbatches (the sampler keys on pipeline_index,
+        # dataloader.py:43-100), so the post-backward reduce-scatter SUM
+        # yields world_size x the single-rank gradient — the reference
+        # never rescales (layer.py:199-223).
+        full_ref[:sh.n_params] = world * grads_ref[layer.layer_id]
+        expect = full_ref[rank * sh.shard_size:(rank + 1) * sh.shard_size]
+        torch.testing.assert_close(layer.flat_grad, expect,
+                                   rtol=1e-4, atol=1e-5)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("target", [_run_2stage, _run_dp, _run_copy, _run_fsdp],
+                         ids=["2stage_parity", "dp_allreduce", "layer_copy",
+                              "fsdp_full_shard"])
 def test_multiprocess(target, tmp_path):
     mp.spawn(target, args=(2, str(tmp_path)), nprocs=2, join=True)
