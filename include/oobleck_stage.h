@@ -155,6 +155,28 @@ int ob_gelu_bwd_f32(const void* u, const void* dg, void* du, int64_t n,
 /* db[n] += sum_m X[m,n] (bias gradient). */
 int ob_colsum_f32(const void* X, void* db, int64_t M, int64_t N, void* stream);
 
+/* ---- bf16 mixed-precision path ------------------------
+ * bf16 MFMA GEMM (v_mfma_f32_32x32x16_bf16, fp32 accumulate).  Same
+ * call shape as ob_gemm_f32; A/B are bf16; out_kind: 0 = bf16 C,
+ * 1 = fp32 C, 2 = fp32 atomicAdd (weight grads / split-K).  lda/ldb must
+ * be multiples of 8 (16-byte staging). */
+int ob_gemm_bf16(int transA, int transB, int64_t M, int64_t N, int64_t K,
+                 float alpha, const void* A, int64_t lda,
+                 int64_t strideA1, int64_t strideA2,
+                 const void* B, int64_t ldb,
+                 int64_t strideB1, int64_t strideB2,
+                 float beta, void* C, int64_t ldc,
+                 int64_t strideC1, int64_t strideC2,
+                 int64_t n1, int64_t n2,
+                 const void* bias, const void* residual, int out_kind,
+                 int splitk, void* stream);
+
+/* dtype casts (weight shadows / activation conversion). */
+int ob_f32_to_bf16(const void* x, void* y, int64_t n, void* stream);
+int ob_f32_to_bf16_t(const void* x, void* y, int64_t rows, int64_t cols,
+                     void* stream);  /* y[c][r] = x[r][c] */
+int ob_bf16_to_f32(const void* x, void* y, int64_t n, void* stream);
+
 const char* ob_last_error(void);
 
 /* Build stamp: returns the gfx arch this library was compiled for. */

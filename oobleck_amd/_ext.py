@@ -53,6 +53,12 @@ def _configure(lib: ctypes.CDLL) -> ctypes.CDLL:
     lib.ob_gelu_fwd_f32.argtypes = [vp, vp, i64, vp]
     lib.ob_gelu_bwd_f32.argtypes = [vp, vp, vp, i64, vp]
     lib.ob_colsum_f32.argtypes = [vp, vp, i64, i64, vp]
+    lib.ob_gemm_bf16.argtypes = [i32, i32, i64, i64, i64, f32, vp, i64, i64,
+                                 i64, vp, i64, i64, i64, f32, vp, i64, i64,
+                                 i64, i64, i64, vp, vp, i32, i32, vp]
+    lib.ob_f32_to_bf16.argtypes = [vp, vp, i64, vp]
+    lib.ob_f32_to_bf16_t.argtypes = [vp, vp, i64, i64, vp]
+    lib.ob_bf16_to_f32.argtypes = [vp, vp, i64, vp]
     return lib
 
 

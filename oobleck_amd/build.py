@@ -7,7 +7,9 @@ import subprocess
 
 PKG_DIR = pathlib.Path(__file__).resolve().parent
 SO_PATH = PKG_DIR / "libob_stage.so"
-SOURCES = [PKG_DIR / "csrc" / "ob_kernels.hip", PKG_DIR / "csrc" / "ob_layer.hip"]
+SOURCES = [PKG_DIR / "csrc" / "ob_kernels.hip",
+           PKG_DIR / "csrc" / "ob_kernels_bf16.hip",
+           PKG_DIR / "csrc" / "ob_layer.hip"]
 HEADERS = [PKG_DIR.parent / "include" / "oobleck_stage.h",
            PKG_DIR / "csrc" / "ob_internal.h"]
 
