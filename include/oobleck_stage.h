@@ -65,6 +65,8 @@ typedef struct {
   int32_t seq_len;      /* S */
   int32_t n_slots;      /* in-flight microbatches (pipeline buffer slots,
                            deepspeed-compatible num_pipe_buffers) */
+  int32_t dtype;        /* 0 = fp32; 1 = bf16 activations/GEMMs with fp32
+                           master weights + fp32 grads (§8 f4) */
 } ob_layer_desc;
 
 /* Number of fp32 elements in the layer's flat parameter buffer.  Layout is
@@ -103,6 +105,11 @@ int ob_layer_backward(ob_layer_t l, int32_t slot, const void* dout, void* din,
                       void* stream);
 
 int ob_layer_destroy(ob_layer_t l);
+
+/* bf16 mode only: refresh the extension's bf16 weight shadows (plain +
+ * transposed layouts) from the fp32 master params — call after ob_layer_bind
+ * and after every optimizer step. */
+int ob_layer_refresh_weights(ob_layer_t l, void* stream);
 
 /* Fused AdamW over a flat buffer (replaces torch AdamW(fused=True) of
  * pipeline.py:117-127; decoupled weight decay + bias correction, eps added

@@ -23,6 +23,7 @@ class ObLayerDesc(ctypes.Structure):
         ("max_batch", ctypes.c_int32),
         ("seq_len", ctypes.c_int32),
         ("n_slots", ctypes.c_int32),
+        ("dtype", ctypes.c_int32),
     ]
 
 
@@ -39,6 +40,7 @@ def _configure(lib: ctypes.CDLL) -> ctypes.CDLL:
     lib.ob_layer_forward.argtypes = [vp, i32, vp, vp, vp, vp]
     lib.ob_layer_backward.argtypes = [vp, i32, vp, vp, vp]
     lib.ob_layer_destroy.argtypes = [vp]
+    lib.ob_layer_refresh_weights.argtypes = [vp, vp]
     lib.ob_adamw_step.argtypes = [vp, vp, vp, vp, i64, i32, f32, f32, f32,
                                   f32, f32, vp]
     lib.ob_gemm_f32.argtypes = [i32, i32, i64, i64, i64, f32, vp, i64, i64,

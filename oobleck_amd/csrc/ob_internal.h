@@ -39,3 +39,37 @@ int ob_ce_fwd_f32(const float* logits, const int64_t* labels, float* lse,
 int ob_ce_bwd_f32(float* logits_to_dlogits, const int64_t* labels,
                   const float* lse, const float* dloss_or_null, int64_t B,
                   int64_t S, int64_t V, void* stream);
+
+// bf16-path launchers (defined in ob_kernels_bf16.hip)
+extern "C" {
+int ob_f32_to_bf16_t_ld(const void* x, void* y, int64_t rows, int64_t cols,
+                        int64_t out_ld, void* stream);
+int ob_layernorm_fwd_bf16(const void* x, const void* w, const void* b,
+                          void* y, void* mean, void* rstd, int64_t rows,
+                          int64_t H, float eps, void* stream);
+int ob_layernorm_bwd_bf16(const void* x, const void* w, const void* mean,
+                          const void* rstd, const void* dy, void* dx,
+                          void* dw, void* db, int64_t rows, int64_t H,
+                          int dx_accum, void* stream);
+int ob_softmax_causal_fwd_bf16(void* scores, int64_t batch, int64_t Sq,
+                               float scale, void* stream);
+int ob_softmax_causal_bwd_bf16(const void* P, void* dP, int64_t batch,
+                               int64_t Sq, void* stream);
+int ob_gelu_fwd_bf16(const void* u, void* g, int64_t n, void* stream);
+int ob_gelu_bwd_bf16(const void* u, const void* dg, void* du, int64_t n,
+                     void* stream);
+int ob_colsum_bf16(const void* X, void* db, int64_t M, int64_t N,
+                   void* stream);
+int ob_embed_fwd_bf16(const void* ids, const void* wte, const void* wpe,
+                      void* out, int64_t B, int64_t Sq, int64_t H,
+                      void* stream);
+int ob_embed_bwd_bf16(const void* ids, const void* dout, void* dwte,
+                      void* dwpe, int64_t B, int64_t Sq, int64_t H,
+                      void* stream);
+int ob_ce_fwd_bf16(const void* logits, const void* labels, void* lse,
+                   void* loss, int64_t B, int64_t Sq, int64_t V, int64_t ld,
+                   void* stream);
+int ob_ce_bwd_bf16(void* logits, const void* labels, const void* lse,
+                   const void* dloss, int64_t B, int64_t Sq, int64_t V,
+                   int64_t ld, void* stream);
+}

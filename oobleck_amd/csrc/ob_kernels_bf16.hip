@@ -298,15 +298,18 @@ __global__ __launch_bounds__(256) void k_f32_to_bf16(const float* __restrict__ x
     y[i] = (__bf16)x[i];
 }
 
-// transposed cast: y[o][i] = x[i][o] for x [rows=in, cols=out]
+// transposed cast: y[c][r] = x[r][c] for x [rows, cols]; y row stride
+// out_ld >= rows (pad entries beyond rows are left untouched — zero the
+// buffer once before the first call when out_ld > rows).
 __global__ __launch_bounds__(256) void k_f32_to_bf16_t(const float* __restrict__ x,
                                                        __bf16* __restrict__ y,
                                                        int64_t rows,
-                                                       int64_t cols) {
+                                                       int64_t cols,
+                                                       int64_t out_ld) {
   for (int64_t idx = (int64_t)blockIdx.x * 256 + threadIdx.x; idx < rows * cols;
        idx += (int64_t)gridDim.x * 256) {
     const int64_t r = idx / cols, c = idx % cols;
-    y[c * rows + r] = (__bf16)x[r * cols + c];
+    y[c * out_ld + r] = (__bf16)x[r * cols + c];
   }
 }
 
@@ -318,7 +321,7 @@ __global__ __launch_bounds__(256) void k_bf16_to_f32(const __bf16* __restrict__ 
     y[i] = (float)x[i];
 }
 
-static inline int64_t bmin64(int64_t a, int64_t b) { return a < b ? a : b; }
+__host__ __device__ static inline int64_t bmin64(int64_t a, int64_t b) { return a < b ? a : b; }
 
 extern "C" int ob_f32_to_bf16(const void* x, void* y, int64_t n, void* stream) {
   k_f32_to_bf16<<<(int)bmin64((n + 255) / 256, 4096), 256, 0, S(stream)>>>(
@@ -329,13 +332,450 @@ extern "C" int ob_f32_to_bf16(const void* x, void* y, int64_t n, void* stream) {
 extern "C" int ob_f32_to_bf16_t(const void* x, void* y, int64_t rows,
                                 int64_t cols, void* stream) {
   k_f32_to_bf16_t<<<(int)bmin64((rows * cols + 255) / 256, 4096), 256, 0,
-                    S(stream)>>>((const float*)x, (__bf16*)y, rows, cols);
+                    S(stream)>>>((const float*)x, (__bf16*)y, rows, cols,
+                                 rows);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+extern "C" int ob_f32_to_bf16_t_ld(const void* x, void* y, int64_t rows,
+                                   int64_t cols, int64_t out_ld,
+                                   void* stream) {
+  k_f32_to_bf16_t<<<(int)bmin64((rows * cols + 255) / 256, 4096), 256, 0,
+                    S(stream)>>>((const float*)x, (__bf16*)y, rows, cols,
+                                 out_ld);
   OB_LAUNCH_CHECK();
   return 0;
 }
 extern "C" int ob_bf16_to_f32(const void* x, void* y, int64_t n, void* stream) {
   k_bf16_to_f32<<<(int)bmin64((n + 255) / 256, 4096), 256, 0, S(stream)>>>(
       (const __bf16*)x, (float*)y, n);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+// ---------------------------------------------------------------------------
+// bf16-I/O elementwise / reduction kernels (fp32 math, fp32 params & grads).
+// Mirrors the fp32 kernels in ob_kernels.hip; separate instantiations keep
+// the proven fp32 path untouched.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ float bwave_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+  return v;
+}
+__device__ __forceinline__ float bblock_sum256(float v, float* lds4) {
+  v = bwave_sum(v);
+  const int wid = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) lds4[wid] = v;
+  __syncthreads();
+  const float r = lds4[0] + lds4[1] + lds4[2] + lds4[3];
+  __syncthreads();
+  return r;
+}
+__device__ __forceinline__ float bblock_max256(float v, float* lds4) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_down(v, off, 64));
+  const int wid = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) lds4[wid] = v;
+  __syncthreads();
+  const float r = fmaxf(fmaxf(lds4[0], lds4[1]), fmaxf(lds4[2], lds4[3]));
+  __syncthreads();
+  return r;
+}
+
+__global__ __launch_bounds__(256) void k_ln_fwd_bf16(
+    const __bf16* __restrict__ x, const float* __restrict__ w,
+    const float* __restrict__ b, __bf16* __restrict__ y,
+    float* __restrict__ mean, float* __restrict__ rstd, int64_t rows, int H,
+    float eps) {
+  __shared__ float lds4[4];
+  for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
+    const __bf16* xr = x + row * H;
+    float s = 0.f, sq = 0.f;
+    for (int c = threadIdx.x; c < H; c += 256) {
+      const float v = bf2f(xr[c]);
+      s += v;
+      sq += v * v;
+    }
+    const float mu = bblock_sum256(s, lds4) / H;
+    const float var = bblock_sum256(sq, lds4) / H - mu * mu;
+    const float rs = rsqrtf(var + eps);
+    if (threadIdx.x == 0) {
+      mean[row] = mu;
+      rstd[row] = rs;
+    }
+    __bf16* yr = y + row * H;
+    for (int c = threadIdx.x; c < H; c += 256)
+      yr[c] = (__bf16)((bf2f(xr[c]) - mu) * rs * w[c] + b[c]);
+    __syncthreads();
+  }
+}
+
+extern "C" int ob_layernorm_fwd_bf16(const void* x, const void* w,
+                                     const void* b, void* y, void* mean,
+                                     void* rstd, int64_t rows, int64_t H,
+                                     float eps, void* stream) {
+  const int grid = (int)bmin64(rows, 16384);
+  k_ln_fwd_bf16<<<grid, 256, 0, S(stream)>>>(
+      (const __bf16*)x, (const float*)w, (const float*)b, (__bf16*)y,
+      (float*)mean, (float*)rstd, rows, (int)H, eps);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+#define BLN_CHUNK 16
+template <bool DX_ACCUM>
+__global__ __launch_bounds__(256) void k_ln_bwd_bf16(
+    const __bf16* __restrict__ x, const float* __restrict__ w,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const __bf16* __restrict__ dy, __bf16* __restrict__ dx,
+    float* __restrict__ dw, float* __restrict__ db, int64_t rows, int H) {
+  __shared__ float lds4[4];
+  const int64_t r0 = (int64_t)blockIdx.x * BLN_CHUNK;
+  float accw[8] = {0}, accb[8] = {0};
+  const int64_t rend = bmin64(rows, r0 + BLN_CHUNK);
+  for (int64_t row = r0; row < rend; ++row) {
+    const __bf16* xr = x + row * H;
+    const __bf16* dyr = dy + row * H;
+    const float mu = mean[row], rs = rstd[row];
+    float s1 = 0.f, s2 = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = threadIdx.x + j * 256;
+      if (c < H) {
+        const float xhat = (bf2f(xr[c]) - mu) * rs;
+        const float dyv = bf2f(dyr[c]);
+        const float dyw = dyv * w[c];
+        s1 += dyw * xhat;
+        s2 += dyw;
+        accw[j] += dyv * xhat;
+        accb[j] += dyv;
+      }
+    }
+    const float m1 = bblock_sum256(s1, lds4) / H;
+    const float m2 = bblock_sum256(s2, lds4) / H;
+    __bf16* dxr = dx + row * H;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = threadIdx.x + j * 256;
+      if (c < H) {
+        const float xhat = (bf2f(xr[c]) - mu) * rs;
+        const float v = rs * (bf2f(dyr[c]) * w[c] - m2 - xhat * m1);
+        dxr[c] = (__bf16)(DX_ACCUM ? bf2f(dxr[c]) + v : v);
+      }
+    }
+    __syncthreads();
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int c = threadIdx.x + j * 256;
+    if (c < H) {
+      atomicAdd(&dw[c], accw[j]);
+      atomicAdd(&db[c], accb[j]);
+    }
+  }
+}
+
+extern "C" int ob_layernorm_bwd_bf16(const void* x, const void* w,
+                                     const void* mean, const void* rstd,
+                                     const void* dy, void* dx, void* dw,
+                                     void* db, int64_t rows, int64_t H,
+                                     int dx_accum, void* stream) {
+  if (H > 2048) return ob_fail("ln_bwd_bf16: H > 2048 unsupported");
+  const int grid = (int)((rows + BLN_CHUNK - 1) / BLN_CHUNK);
+  if (dx_accum)
+    k_ln_bwd_bf16<true><<<grid, 256, 0, S(stream)>>>(
+        (const __bf16*)x, (const float*)w, (const float*)mean,
+        (const float*)rstd, (const __bf16*)dy, (__bf16*)dx, (float*)dw,
+        (float*)db, rows, (int)H);
+  else
+    k_ln_bwd_bf16<false><<<grid, 256, 0, S(stream)>>>(
+        (const __bf16*)x, (const float*)w, (const float*)mean,
+        (const float*)rstd, (const __bf16*)dy, (__bf16*)dx, (float*)dw,
+        (float*)db, rows, (int)H);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+__global__ __launch_bounds__(256) void k_softmax_fwd_bf16(
+    __bf16* __restrict__ scores, int Sq, float scale) {
+  __shared__ float lds4[4];
+  const int64_t rid = blockIdx.x;
+  const int row = (int)(rid % Sq);
+  __bf16* p = scores + rid * Sq;
+  const int valid = row + 1;
+  float v[8];
+  float mx = -INFINITY;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int c = threadIdx.x + j * 256;
+    v[j] = (c < valid) ? bf2f(p[c]) * scale : -INFINITY;
+    mx = fmaxf(mx, v[j]);
+  }
+  mx = bblock_max256(mx, lds4);
+  float sum = 0.f;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    v[j] = (v[j] == -INFINITY) ? 0.f : __expf(v[j] - mx);
+    sum += v[j];
+  }
+  sum = bblock_sum256(sum, lds4);
+  const float inv = 1.f / sum;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int c = threadIdx.x + j * 256;
+    if (c < Sq) p[c] = (__bf16)(v[j] * inv);
+  }
+}
+
+extern "C" int ob_softmax_causal_fwd_bf16(void* scores, int64_t batch,
+                                          int64_t Sq, float scale,
+                                          void* stream) {
+  if (Sq > 2048) return ob_fail("softmax_bf16: S > 2048 unsupported");
+  k_softmax_fwd_bf16<<<(unsigned)(batch * Sq), 256, 0, S(stream)>>>(
+      (__bf16*)scores, (int)Sq, scale);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+__global__ __launch_bounds__(256) void k_softmax_bwd_bf16(
+    const __bf16* __restrict__ P, __bf16* __restrict__ dP, int Sq) {
+  __shared__ float lds4[4];
+  const int64_t rid = blockIdx.x;
+  const int row = (int)(rid % Sq);
+  const __bf16* pr = P + rid * Sq;
+  __bf16* dr = dP + rid * Sq;
+  const int valid = row + 1;
+  float pv[8], dv[8];
+  float t = 0.f;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int c = threadIdx.x + j * 256;
+    pv[j] = (c < valid) ? bf2f(pr[c]) : 0.f;
+    dv[j] = (c < valid) ? bf2f(dr[c]) : 0.f;
+    t += pv[j] * dv[j];
+  }
+  t = bblock_sum256(t, lds4);
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int c = threadIdx.x + j * 256;
+    if (c < Sq) dr[c] = (__bf16)(pv[j] * (dv[j] - t));
+  }
+}
+
+extern "C" int ob_softmax_causal_bwd_bf16(const void* P, void* dP,
+                                          int64_t batch, int64_t Sq,
+                                          void* stream) {
+  if (Sq > 2048) return ob_fail("softmax_bwd_bf16: S > 2048 unsupported");
+  k_softmax_bwd_bf16<<<(unsigned)(batch * Sq), 256, 0, S(stream)>>>(
+      (const __bf16*)P, (__bf16*)dP, (int)Sq);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+#define BGELU_K 0.7978845608028654f
+#define BGELU_C 0.044715f
+
+__global__ __launch_bounds__(256) void k_gelu_fwd_bf16(
+    const __bf16* __restrict__ u, __bf16* __restrict__ g, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * 256) {
+    const float x = bf2f(u[i]);
+    const float t = tanhf(BGELU_K * (x + BGELU_C * x * x * x));
+    g[i] = (__bf16)(0.5f * x * (1.f + t));
+  }
+}
+__global__ __launch_bounds__(256) void k_gelu_bwd_bf16(
+    const __bf16* __restrict__ u, const __bf16* __restrict__ dg,
+    __bf16* __restrict__ du, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * 256) {
+    const float x = bf2f(u[i]);
+    const float t = tanhf(BGELU_K * (x + BGELU_C * x * x * x));
+    const float d = 0.5f * (1.f + t) +
+                    0.5f * x * (1.f - t * t) * BGELU_K * (1.f + 3.f * BGELU_C * x * x);
+    du[i] = (__bf16)(bf2f(dg[i]) * d);
+  }
+}
+extern "C" int ob_gelu_fwd_bf16(const void* u, void* g, int64_t n,
+                                void* stream) {
+  k_gelu_fwd_bf16<<<(int)bmin64((n + 255) / 256, 2048), 256, 0, S(stream)>>>(
+      (const __bf16*)u, (__bf16*)g, n);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+extern "C" int ob_gelu_bwd_bf16(const void* u, const void* dg, void* du,
+                                int64_t n, void* stream) {
+  k_gelu_bwd_bf16<<<(int)bmin64((n + 255) / 256, 2048), 256, 0, S(stream)>>>(
+      (const __bf16*)u, (const __bf16*)dg, (__bf16*)du, n);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+__global__ __launch_bounds__(256) void k_colsum_bf16(
+    const __bf16* __restrict__ X, float* __restrict__ db, int64_t M,
+    int64_t N) {
+  const int64_t c = (int64_t)blockIdx.x * 256 + threadIdx.x;
+  if (c >= N) return;
+  const int64_t r0 = (int64_t)blockIdx.y * 256;
+  const int64_t r1 = bmin64(M, r0 + 256);
+  float acc = 0.f;
+  for (int64_t r = r0; r < r1; ++r) acc += bf2f(X[r * N + c]);
+  atomicAdd(&db[c], acc);
+}
+extern "C" int ob_colsum_bf16(const void* X, void* db, int64_t M, int64_t N,
+                              void* stream) {
+  dim3 grid((unsigned)((N + 255) / 256), (unsigned)((M + 255) / 256));
+  k_colsum_bf16<<<grid, 256, 0, S(stream)>>>((const __bf16*)X, (float*)db, M,
+                                             N);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+__global__ __launch_bounds__(256) void k_embed_fwd_bf16(
+    const int64_t* __restrict__ ids, const float* __restrict__ wte,
+    const float* __restrict__ wpe, __bf16* __restrict__ out, int64_t BS,
+    int Sq, int H) {
+  for (int64_t row = blockIdx.x; row < BS; row += gridDim.x) {
+    const int64_t id = ids[row];
+    const int s = (int)(row % Sq);
+    const float* te = wte + id * H;
+    const float* pe = wpe + (int64_t)s * H;
+    __bf16* o = out + row * H;
+    for (int c = threadIdx.x; c < H; c += 256) o[c] = (__bf16)(te[c] + pe[c]);
+  }
+}
+extern "C" int ob_embed_fwd_bf16(const void* ids, const void* wte,
+                                 const void* wpe, void* out, int64_t B,
+                                 int64_t Sq, int64_t H, void* stream) {
+  k_embed_fwd_bf16<<<(int)bmin64(B * Sq, 16384), 256, 0, S(stream)>>>(
+      (const int64_t*)ids, (const float*)wte, (const float*)wpe, (__bf16*)out,
+      B * Sq, (int)Sq, (int)H);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+__global__ __launch_bounds__(256) void k_embed_bwd_bf16(
+    const int64_t* __restrict__ ids, const __bf16* __restrict__ dout,
+    float* __restrict__ dwte, float* __restrict__ dwpe, int64_t BS, int Sq,
+    int H) {
+  for (int64_t row = blockIdx.x; row < BS; row += gridDim.x) {
+    const int64_t id = ids[row];
+    const int s = (int)(row % Sq);
+    const __bf16* d = dout + row * H;
+    for (int c = threadIdx.x; c < H; c += 256) {
+      const float v = bf2f(d[c]);
+      atomicAdd(&dwte[id * H + c], v);
+      atomicAdd(&dwpe[(int64_t)s * H + c], v);
+    }
+  }
+}
+extern "C" int ob_embed_bwd_bf16(const void* ids, const void* dout, void* dwte,
+                                 void* dwpe, int64_t B, int64_t Sq, int64_t H,
+                                 void* stream) {
+  k_embed_bwd_bf16<<<(int)bmin64(B * Sq, 16384), 256, 0, S(stream)>>>(
+      (const int64_t*)ids, (const __bf16*)dout, (float*)dwte, (float*)dwpe,
+      B * Sq, (int)Sq, (int)H);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+// CE with bf16 logits at row stride ld (padded to 8-half alignment); the
+// shift/mean semantics match the fp32 kernels.
+__global__ __launch_bounds__(256) void k_ce_fwd_bf16(
+    const __bf16* __restrict__ logits, const int64_t* __restrict__ labels,
+    float* __restrict__ lse, float* __restrict__ loss, int64_t BS, int Sq,
+    int V, int64_t ld) {
+  __shared__ float lmax[4], lsum[4];
+  for (int64_t row = blockIdx.x; row < BS; row += gridDim.x) {
+    const __bf16* lr = logits + row * ld;
+    float m = -INFINITY, s = 0.f;
+    for (int c = threadIdx.x; c < V; c += 256) {
+      const float x = bf2f(lr[c]);
+      if (x > m) {
+        s = s * __expf(m - x) + 1.f;
+        m = x;
+      } else {
+        s += __expf(x - m);
+      }
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      const float om = __shfl_down(m, off, 64);
+      const float os = __shfl_down(s, off, 64);
+      const float nm = fmaxf(m, om);
+      const float t1 = (m > -INFINITY) ? s * __expf(m - nm) : 0.f;
+      const float t2 = (om > -INFINITY) ? os * __expf(om - nm) : 0.f;
+      s = t1 + t2;
+      m = nm;
+    }
+    const int wid = threadIdx.x >> 6;
+    if ((threadIdx.x & 63) == 0) {
+      lmax[wid] = m;
+      lsum[wid] = s;
+    }
+    __syncthreads();
+    const float bm = fmaxf(fmaxf(lmax[0], lmax[1]), fmaxf(lmax[2], lmax[3]));
+    float bs = 0.f;
+#pragma unroll
+    for (int wi = 0; wi < 4; ++wi)
+      bs += (lmax[wi] > -INFINITY) ? lsum[wi] * __expf(lmax[wi] - bm) : 0.f;
+    const float l = bm + __logf(bs);
+    const int s_pos = (int)(row % Sq);
+    if (threadIdx.x == 0) {
+      lse[row] = l;
+      if (s_pos < Sq - 1) {
+        const int64_t lab = labels[row + 1];
+        const int64_t B = BS / Sq;
+        const float inv = 1.f / (float)(B * (Sq - 1));
+        atomicAdd(loss, (l - bf2f(lr[lab])) * inv);
+      }
+    }
+    __syncthreads();
+  }
+}
+extern "C" int ob_ce_fwd_bf16(const void* logits, const void* labels,
+                              void* lse, void* loss, int64_t B, int64_t Sq,
+                              int64_t V, int64_t ld, void* stream) {
+  k_ce_fwd_bf16<<<(int)bmin64(B * Sq, 16384), 256, 0, S(stream)>>>(
+      (const __bf16*)logits, (const int64_t*)labels, (float*)lse,
+      (float*)loss, B * Sq, (int)Sq, (int)V, ld);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+__global__ __launch_bounds__(256) void k_ce_bwd_bf16(
+    __bf16* __restrict__ logits, const int64_t* __restrict__ labels,
+    const float* __restrict__ lse, const float* __restrict__ dloss,
+    int64_t BS, int Sq, int V, int64_t ld) {
+  const float dl = dloss ? *dloss : 1.f;
+  const int64_t B = BS / Sq;
+  const float scale = dl / (float)(B * (Sq - 1));
+  for (int64_t row = blockIdx.y; row < BS; row += gridDim.y) {
+    __bf16* lr = logits + row * ld;
+    const int s_pos = (int)(row % Sq);
+    const float l = lse[row];
+    const bool valid = s_pos < Sq - 1;
+    const int64_t lab = valid ? labels[row + 1] : -1;
+    for (int64_t c = (int64_t)blockIdx.x * 256 + threadIdx.x; c < ld;
+         c += (int64_t)gridDim.x * 256) {
+      float g = 0.f;
+      if (valid && c < V)
+        g = scale * (__expf(bf2f(lr[c]) - l) - (c == lab ? 1.f : 0.f));
+      lr[c] = (__bf16)g;
+    }
+  }
+}
+extern "C" int ob_ce_bwd_bf16(void* logits, const void* labels,
+                              const void* lse, const void* dloss, int64_t B,
+                              int64_t Sq, int64_t V, int64_t ld,
+                              void* stream) {
+  dim3 grid((unsigned)bmin64((ld + 255) / 256, 256),
+            (unsigned)bmin64(B * Sq, 16384));
+  k_ce_bwd_bf16<<<grid, 256, 0, S(stream)>>>(
+      (__bf16*)logits, (const int64_t*)labels, (const float*)lse,
+      (const float*)dloss, B * Sq, (int)Sq, (int)V, ld);
   OB_LAUNCH_CHECK();
   return 0;
 }

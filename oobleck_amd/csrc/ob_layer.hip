@@ -60,6 +60,16 @@ struct ob_layer {
   int64_t o_x = 0, o_mean1 = 0, o_rstd1 = 0, o_ln1 = 0, o_qkv = 0, o_p = 0,
           o_attnm = 0, o_hmid = 0, o_mean2 = 0, o_rstd2 = 0, o_ln2 = 0,
           o_u = 0, o_g = 0, o_logits = 0, o_lse = 0;
+  // bf16 mode (d.dtype == 1): weight shadows in plain + transposed layouts
+  // so every GEMM operand has k contiguous in memory (see
+  // ob_kernels_bf16.hip header).  The stash keeps its fp32-sized offsets
+  // and bf16 data occupies the first half of each region (2x overcommit,
+  // traded for zero layout churn; memory is plentiful at 288 GB).
+  __bf16* shadows = nullptr;
+  int64_t v_pad = 0;  // vocab rounded up to 8 (16-byte bf16 alignment)
+  int64_t sh_qkv = 0, sh_qkv_t = 0, sh_ap = 0, sh_ap_t = 0, sh_fc = 0,
+          sh_fc_t = 0, sh_mp = 0, sh_mp_t = 0, sh_lm = 0, sh_lm_t = 0;
+  int64_t shadow_count = 0;
 };
 
 // parameter offsets (canonical layout, oracle/gpt2_oracle.py::layer_param_spec)
@@ -154,6 +164,29 @@ extern "C" int ob_layer_create(const ob_layer_desc* d, ob_layer_t* out) {
       return ob_fail("create: ids hipMalloc failed");
     }
   }
+  l->v_pad = (V + 7) / 8 * 8;
+  if (d->dtype == 1) {
+    int64_t o2 = 0;
+    if (d->kind == OB_KIND_BLOCK) {
+      l->sh_qkv = o2; o2 += 3 * H * H;
+      l->sh_qkv_t = o2; o2 += 3 * H * H;
+      l->sh_ap = o2; o2 += H * H;
+      l->sh_ap_t = o2; o2 += H * H;
+      l->sh_fc = o2; o2 += 4 * H * H;
+      l->sh_fc_t = o2; o2 += 4 * H * H;
+      l->sh_mp = o2; o2 += 4 * H * H;
+      l->sh_mp_t = o2; o2 += 4 * H * H;
+    } else if (d->kind == OB_KIND_FINAL) {
+      l->sh_lm = o2; o2 += V * H;
+      l->sh_lm_t = o2; o2 += H * l->v_pad;
+    }
+    l->shadow_count = o2;
+    if (o2 > 0) {
+      if (hipMalloc(&l->shadows, (size_t)o2 * sizeof(__bf16)) != hipSuccess)
+        return ob_fail("create: shadow hipMalloc failed");
+      OB_HIP(hipMemset(l->shadows, 0, (size_t)o2 * sizeof(__bf16)));
+    }
+  }
   // grow the shared backward workspace
   if (d->kind == OB_KIND_BLOCK) {
     if (ws_ensure(&g_ws.dp, &g_ws.sz_dp, Bm * nh * Sq * Sq)) return 1;
@@ -189,8 +222,46 @@ extern "C" int ob_layer_destroy(ob_layer_t l) {
   if (!l) return 0;
   if (l->stash) hipFree(l->stash);
   if (l->ids) hipFree(l->ids);
+  if (l->shadows) hipFree(l->shadows);
   delete l;
   return 0;
+}
+
+extern "C" int ob_layer_refresh_weights(ob_layer_t l, void* stream) {
+  if (!l) return ob_fail("refresh: null layer");
+  if (l->d.dtype != 1) return 0;  // fp32 mode: nothing to do
+  if (!l->params) return ob_fail("refresh: params not bound");
+  const int64_t H = l->d.n_embd, V = l->d.vocab_size;
+  const float* p = l->params;
+  __bf16* sh = l->shadows;
+  switch (l->d.kind) {
+    case OB_KIND_EMBED:
+      return 0;  // embedding reads the fp32 master directly
+    case OB_KIND_BLOCK: {
+      const BlockParams bp = block_params(H);
+      if (ob_f32_to_bf16(p + bp.w_qkv, sh + l->sh_qkv, 3 * H * H, stream) ||
+          ob_f32_to_bf16_t_ld(p + bp.w_qkv, sh + l->sh_qkv_t, H, 3 * H, H,
+                              stream) ||
+          ob_f32_to_bf16(p + bp.w_attnproj, sh + l->sh_ap, H * H, stream) ||
+          ob_f32_to_bf16_t_ld(p + bp.w_attnproj, sh + l->sh_ap_t, H, H, H,
+                              stream) ||
+          ob_f32_to_bf16(p + bp.w_fc, sh + l->sh_fc, 4 * H * H, stream) ||
+          ob_f32_to_bf16_t_ld(p + bp.w_fc, sh + l->sh_fc_t, H, 4 * H, H,
+                              stream) ||
+          ob_f32_to_bf16(p + bp.w_mlpproj, sh + l->sh_mp, 4 * H * H, stream) ||
+          ob_f32_to_bf16_t_ld(p + bp.w_mlpproj, sh + l->sh_mp_t, 4 * H, H,
+                              4 * H, stream))
+        return 1;
+      return 0;
+    }
+    case OB_KIND_FINAL:
+      if (ob_f32_to_bf16(p + 2 * H, sh + l->sh_lm, V * H, stream) ||
+          ob_f32_to_bf16_t_ld(p + 2 * H, sh + l->sh_lm_t, V, H, l->v_pad,
+                              stream))
+        return 1;
+      return 0;
+  }
+  return ob_fail("refresh: bad kind");
 }
 
 // helper: plain strided-batch GEMM call (n1=batch, n2=1) or 2-level.
@@ -326,6 +397,238 @@ static int final_forward(ob_layer* l, int slot, const float* in, float* out,
   return 0;
 }
 
+// ---------------------------------------------------------------------------
+// bf16 mixed-precision path (d.dtype == 1): bf16 activations + bf16 MFMA
+// GEMMs with fp32 accumulate, fp32 master weights/biases/LN params, fp32
+// grads.  Weight operands come from the plain/transposed shadows so every
+// GEMM stages its operands with k contiguous (see ob_kernels_bf16.hip).
+// ---------------------------------------------------------------------------
+
+static int gemm_bf(int tA, int tB, int64_t M, int64_t N, int64_t K,
+                   float alpha, const void* A, int64_t lda, int64_t sA1,
+                   int64_t sA2, const void* B, int64_t ldb, int64_t sB1,
+                   int64_t sB2, void* C, int64_t ldc, int64_t sC1,
+                   int64_t sC2, int64_t n1, int64_t n2, const float* bias,
+                   const void* R, int out_kind, int splitk, void* stream) {
+  return ob_gemm_bf16(tA, tB, M, N, K, alpha, A, lda, sA1, sA2, B, ldb, sB1,
+                      sB2, 0.f, C, ldc, sC1, sC2, n1, n2, bias, R, out_kind,
+                      splitk, stream);
+}
+
+static int block_forward_bf16(ob_layer* l, int slot, const __bf16* in,
+                              __bf16* out, void* stream) {
+  const int64_t Sq = l->d.seq_len, H = l->d.n_embd, nh = l->d.n_head;
+  const int64_t B = l->B, BS = B * Sq;
+  const int64_t hd = H / nh;
+  float* st = l->stash + (int64_t)slot * l->slot_stride;
+  const float* p = l->params;
+  const __bf16* sh = l->shadows;
+  const BlockParams bp = block_params(H);
+
+  __bf16* x = (__bf16*)(st + l->o_x);
+  OB_HIP(hipMemcpyAsync(x, in, BS * H * sizeof(__bf16),
+                        hipMemcpyDeviceToDevice, S(stream)));
+  __bf16* ln1 = (__bf16*)(st + l->o_ln1);
+  if (ob_layernorm_fwd_bf16(x, p + bp.ln1_w, p + bp.ln1_b, ln1,
+                            st + l->o_mean1, st + l->o_rstd1, BS, H, 1e-5f,
+                            stream))
+    return 1;
+  __bf16* qkv = (__bf16*)(st + l->o_qkv);
+  if (gemm_bf(0, 1, BS, 3 * H, H, 1.f, ln1, H, 0, 0, sh + l->sh_qkv_t, H, 0,
+              0, qkv, 3 * H, 0, 0, 1, 1, p + bp.b_qkv, nullptr, 0, 1, stream))
+    return 1;
+  __bf16* P = (__bf16*)(st + l->o_p);
+  if (gemm_bf(0, 1, Sq, Sq, hd, 1.f, qkv, 3 * H, Sq * 3 * H, hd, qkv + H,
+              3 * H, Sq * 3 * H, hd, P, Sq, nh * Sq * Sq, Sq * Sq, B, nh,
+              nullptr, nullptr, 0, 1, stream))
+    return 1;
+  if (ob_softmax_causal_fwd_bf16(P, B * nh, Sq, 1.f / sqrtf((float)hd),
+                                 stream))
+    return 1;
+  __bf16* am = (__bf16*)(st + l->o_attnm);
+  if (gemm_bf(0, 0, Sq, hd, Sq, 1.f, P, Sq, nh * Sq * Sq, Sq * Sq,
+              qkv + 2 * H, 3 * H, Sq * 3 * H, hd, am, H, Sq * H, hd, B, nh,
+              nullptr, nullptr, 0, 1, stream))
+    return 1;
+  __bf16* hmid = (__bf16*)(st + l->o_hmid);
+  if (gemm_bf(0, 1, BS, H, H, 1.f, am, H, 0, 0, sh + l->sh_ap_t, H, 0, 0,
+              hmid, H, 0, 0, 1, 1, p + bp.b_attnproj, x, 0, 1, stream))
+    return 1;
+  __bf16* ln2 = (__bf16*)(st + l->o_ln2);
+  if (ob_layernorm_fwd_bf16(hmid, p + bp.ln2_w, p + bp.ln2_b, ln2,
+                            st + l->o_mean2, st + l->o_rstd2, BS, H, 1e-5f,
+                            stream))
+    return 1;
+  __bf16* u = (__bf16*)(st + l->o_u);
+  if (gemm_bf(0, 1, BS, 4 * H, H, 1.f, ln2, H, 0, 0, sh + l->sh_fc_t, H, 0,
+              0, u, 4 * H, 0, 0, 1, 1, p + bp.b_fc, nullptr, 0, 1, stream))
+    return 1;
+  __bf16* gact = (__bf16*)(st + l->o_g);
+  if (ob_gelu_fwd_bf16(u, gact, BS * 4 * H, stream)) return 1;
+  if (gemm_bf(0, 1, BS, H, 4 * H, 1.f, gact, 4 * H, 0, 0, sh + l->sh_mp_t,
+              4 * H, 0, 0, out, H, 0, 0, 1, 1, p + bp.b_mlpproj, hmid, 0, 1,
+              stream))
+    return 1;
+  return 0;
+}
+
+static int final_forward_bf16(ob_layer* l, int slot, const __bf16* in,
+                              float* out, const int64_t* labels,
+                              void* stream) {
+  const int64_t Sq = l->d.seq_len, H = l->d.n_embd, V = l->d.vocab_size;
+  const int64_t B = l->B, BS = B * Sq;
+  float* st = l->stash + (int64_t)slot * l->slot_stride;
+  const float* p = l->params;
+  if (!labels) return ob_fail("final_forward_bf16: labels required");
+
+  __bf16* x = (__bf16*)(st + l->o_x);
+  OB_HIP(hipMemcpyAsync(x, in, BS * H * sizeof(__bf16),
+                        hipMemcpyDeviceToDevice, S(stream)));
+  int64_t* labs = l->ids + (int64_t)slot * (int64_t)l->d.max_batch * Sq;
+  OB_HIP(hipMemcpyAsync(labs, labels, BS * sizeof(int64_t),
+                        hipMemcpyDeviceToDevice, S(stream)));
+  __bf16* lnf = (__bf16*)(st + l->o_ln1);
+  if (ob_layernorm_fwd_bf16(x, p + 0, p + H, lnf, st + l->o_mean1,
+                            st + l->o_rstd1, BS, H, 1e-5f, stream))
+    return 1;
+  __bf16* logits = (__bf16*)(st + l->o_logits);
+  if (gemm_bf(0, 1, BS, V, H, 1.f, lnf, H, 0, 0, l->shadows + l->sh_lm, H, 0,
+              0, logits, l->v_pad, 0, 0, 1, 1, nullptr, nullptr, 0, 1,
+              stream))
+    return 1;
+  OB_HIP(hipMemsetAsync(out, 0, sizeof(float), S(stream)));
+  if (ob_ce_fwd_bf16(logits, labs, st + l->o_lse, out, B, Sq, V, l->v_pad,
+                     stream))
+    return 1;
+  return 0;
+}
+
+static int block_backward_bf16(ob_layer* l, int slot, const __bf16* dout,
+                               __bf16* din, void* stream) {
+  const int64_t Sq = l->d.seq_len, H = l->d.n_embd, nh = l->d.n_head;
+  const int64_t B = l->B, BS = B * Sq;
+  const int64_t hd = H / nh;
+  const float scale = 1.f / sqrtf((float)hd);
+  float* st = l->stash + (int64_t)slot * l->slot_stride;
+  const float* p = l->params;
+  float* g = l->grads;
+  const __bf16* sh = l->shadows;
+  const BlockParams bp = block_params(H);
+
+  __bf16* x = (__bf16*)(st + l->o_x);
+  __bf16* ln1 = (__bf16*)(st + l->o_ln1);
+  __bf16* qkv = (__bf16*)(st + l->o_qkv);
+  __bf16* P = (__bf16*)(st + l->o_p);
+  __bf16* am = (__bf16*)(st + l->o_attnm);
+  __bf16* hmid = (__bf16*)(st + l->o_hmid);
+  __bf16* ln2 = (__bf16*)(st + l->o_ln2);
+  __bf16* u = (__bf16*)(st + l->o_u);
+  __bf16* gact = (__bf16*)(st + l->o_g);
+
+  __bf16* DY4 = (__bf16*)g_ws.b4h;
+  __bf16* DLN = (__bf16*)g_ws.bsh1;
+  __bf16* DATT = (__bf16*)g_ws.bsh2;
+  __bf16* DQKV = (__bf16*)g_ws.dqkv;
+  __bf16* DP = (__bf16*)g_ws.dp;
+
+  OB_HIP(hipMemcpyAsync(din, dout, BS * H * sizeof(__bf16),
+                        hipMemcpyDeviceToDevice, S(stream)));
+  // ---- MLP ----
+  if (gemm_bf(0, 1, BS, 4 * H, H, 1.f, dout, H, 0, 0, sh + l->sh_mp, H, 0, 0,
+              DY4, 4 * H, 0, 0, 1, 1, nullptr, nullptr, 0, 1, stream))
+    return 1;
+  if (gemm_bf(1, 0, 4 * H, H, BS, 1.f, gact, 4 * H, 0, 0, dout, H, 0, 0,
+              g + bp.w_mlpproj, H, 0, 0, 1, 1, nullptr, nullptr, 2,
+              pick_splitk(4 * H, H, BS), stream))
+    return 1;
+  if (ob_colsum_bf16(dout, g + bp.b_mlpproj, BS, H, stream)) return 1;
+  if (ob_gelu_bwd_bf16(u, DY4, DY4, BS * 4 * H, stream)) return 1;
+  if (gemm_bf(1, 0, H, 4 * H, BS, 1.f, ln2, H, 0, 0, DY4, 4 * H, 0, 0,
+              g + bp.w_fc, 4 * H, 0, 0, 1, 1, nullptr, nullptr, 2,
+              pick_splitk(H, 4 * H, BS), stream))
+    return 1;
+  if (ob_colsum_bf16(DY4, g + bp.b_fc, BS, 4 * H, stream)) return 1;
+  if (gemm_bf(0, 1, BS, H, 4 * H, 1.f, DY4, 4 * H, 0, 0, sh + l->sh_fc,
+              4 * H, 0, 0, DLN, H, 0, 0, 1, 1, nullptr, nullptr, 0, 1,
+              stream))
+    return 1;
+  if (ob_layernorm_bwd_bf16(hmid, p + bp.ln2_w, st + l->o_mean2,
+                            st + l->o_rstd2, DLN, din, g + bp.ln2_w,
+                            g + bp.ln2_b, BS, H, 1, stream))
+    return 1;
+  // ---- attention projection ----
+  if (gemm_bf(0, 1, BS, H, H, 1.f, din, H, 0, 0, sh + l->sh_ap, H, 0, 0,
+              DATT, H, 0, 0, 1, 1, nullptr, nullptr, 0, 1, stream))
+    return 1;
+  if (gemm_bf(1, 0, H, H, BS, 1.f, am, H, 0, 0, din, H, 0, 0,
+              g + bp.w_attnproj, H, 0, 0, 1, 1, nullptr, nullptr, 2,
+              pick_splitk(H, H, BS), stream))
+    return 1;
+  if (ob_colsum_bf16(din, g + bp.b_attnproj, BS, H, stream)) return 1;
+  // ---- attention core ----
+  if (gemm_bf(0, 1, Sq, Sq, hd, 1.f, DATT, H, Sq * H, hd, qkv + 2 * H, 3 * H,
+              Sq * 3 * H, hd, DP, Sq, nh * Sq * Sq, Sq * Sq, B, nh, nullptr,
+              nullptr, 0, 1, stream))
+    return 1;
+  if (ob_softmax_causal_bwd_bf16(P, DP, B * nh, Sq, stream)) return 1;
+  if (gemm_bf(0, 0, Sq, hd, Sq, scale, DP, Sq, nh * Sq * Sq, Sq * Sq,
+              qkv + H, 3 * H, Sq * 3 * H, hd, DQKV, 3 * H, Sq * 3 * H, hd, B,
+              nh, nullptr, nullptr, 0, 1, stream))
+    return 1;
+  if (gemm_bf(1, 0, Sq, hd, Sq, scale, DP, Sq, nh * Sq * Sq, Sq * Sq, qkv,
+              3 * H, Sq * 3 * H, hd, DQKV + H, 3 * H, Sq * 3 * H, hd, B, nh,
+              nullptr, nullptr, 0, 1, stream))
+    return 1;
+  if (gemm_bf(1, 0, Sq, hd, Sq, 1.f, P, Sq, nh * Sq * Sq, Sq * Sq, DATT, H,
+              Sq * H, hd, DQKV + 2 * H, 3 * H, Sq * 3 * H, hd, B, nh,
+              nullptr, nullptr, 0, 1, stream))
+    return 1;
+  // ---- QKV projection ----
+  if (ob_colsum_bf16(DQKV, g + bp.b_qkv, BS, 3 * H, stream)) return 1;
+  if (gemm_bf(1, 0, H, 3 * H, BS, 1.f, ln1, H, 0, 0, DQKV, 3 * H, 0, 0,
+              g + bp.w_qkv, 3 * H, 0, 0, 1, 1, nullptr, nullptr, 2,
+              pick_splitk(H, 3 * H, BS), stream))
+    return 1;
+  if (gemm_bf(0, 1, BS, H, 3 * H, 1.f, DQKV, 3 * H, 0, 0, sh + l->sh_qkv,
+              3 * H, 0, 0, DLN, H, 0, 0, 1, 1, nullptr, nullptr, 0, 1,
+              stream))
+    return 1;
+  if (ob_layernorm_bwd_bf16(x, p + bp.ln1_w, st + l->o_mean1,
+                            st + l->o_rstd1, DLN, din, g + bp.ln1_w,
+                            g + bp.ln1_b, BS, H, 1, stream))
+    return 1;
+  return 0;
+}
+
+static int final_backward_bf16(ob_layer* l, int slot, const float* dout,
+                               __bf16* din, void* stream) {
+  const int64_t Sq = l->d.seq_len, H = l->d.n_embd, V = l->d.vocab_size;
+  const int64_t B = l->B, BS = B * Sq;
+  float* st = l->stash + (int64_t)slot * l->slot_stride;
+  const float* p = l->params;
+  float* g = l->grads;
+  __bf16* x = (__bf16*)(st + l->o_x);
+  __bf16* lnf = (__bf16*)(st + l->o_ln1);
+  __bf16* logits = (__bf16*)(st + l->o_logits);
+  int64_t* labs = l->ids + (int64_t)slot * (int64_t)l->d.max_batch * Sq;
+  __bf16* DLN = (__bf16*)g_ws.bsh1;
+
+  if (ob_ce_bwd_bf16(logits, labs, st + l->o_lse, dout, B, Sq, V, l->v_pad,
+                     stream))
+    return 1;
+  if (gemm_bf(1, 0, V, H, BS, 1.f, logits, l->v_pad, 0, 0, lnf, H, 0, 0,
+              g + 2 * H, H, 0, 0, 1, 1, nullptr, nullptr, 2, 1, stream))
+    return 1;
+  if (gemm_bf(0, 1, BS, H, V, 1.f, logits, l->v_pad, 0, 0,
+              l->shadows + l->sh_lm_t, l->v_pad, 0, 0, DLN, H, 0, 0, 1, 1,
+              nullptr, nullptr, 0, 1, stream))
+    return 1;
+  if (ob_layernorm_bwd_bf16(x, p + 0, st + l->o_mean1, st + l->o_rstd1, DLN,
+                            din, g + 0, g + H, BS, H, 0, stream))
+    return 1;
+  return 0;
+}
+
 extern "C" int ob_layer_forward(ob_layer_t l, int32_t slot, const void* in,
                                 void* out, const int64_t* labels,
                                 void* stream) {
@@ -339,12 +642,21 @@ extern "C" int ob_layer_forward(ob_layer_t l, int32_t slot, const void* in,
       int64_t* ids = l->ids + (int64_t)slot * (int64_t)l->d.max_batch * Sq;
       OB_HIP(hipMemcpyAsync(ids, in, BS * sizeof(int64_t),
                             hipMemcpyDeviceToDevice, S(stream)));
+      if (l->d.dtype == 1)
+        return ob_embed_fwd_bf16(ids, l->params, l->params + V * H, out, l->B,
+                                 Sq, H, stream);
       return ob_embed_fwd_f32(ids, l->params, l->params + V * H, (float*)out,
                               l->B, Sq, H, stream);
     }
     case OB_KIND_BLOCK:
+      if (l->d.dtype == 1)
+        return block_forward_bf16(l, slot, (const __bf16*)in, (__bf16*)out,
+                                  stream);
       return block_forward(l, slot, (const float*)in, (float*)out, stream);
     case OB_KIND_FINAL:
+      if (l->d.dtype == 1)
+        return final_forward_bf16(l, slot, (const __bf16*)in, (float*)out,
+                                  labels, stream);
       return final_forward(l, slot, (const float*)in, (float*)out, labels,
                            stream);
   }
@@ -500,14 +812,23 @@ extern "C" int ob_layer_backward(ob_layer_t l, int32_t slot, const void* dout,
     case OB_KIND_EMBED: {
       const int64_t Sq = l->d.seq_len, H = l->d.n_embd, V = l->d.vocab_size;
       int64_t* ids = l->ids + (int64_t)slot * (int64_t)l->d.max_batch * Sq;
+      if (l->d.dtype == 1)
+        return ob_embed_bwd_bf16(ids, dout, l->grads, l->grads + V * H, l->B,
+                                 Sq, H, stream);
       return ob_embed_bwd_f32(ids, (const float*)dout, l->grads,
                               l->grads + V * H, l->B, Sq, H, stream);
     }
     case OB_KIND_BLOCK:
       if (!dout || !din) return ob_fail("block backward: dout/din required");
+      if (l->d.dtype == 1)
+        return block_backward_bf16(l, slot, (const __bf16*)dout, (__bf16*)din,
+                                   stream);
       return block_backward(l, slot, (const float*)dout, (float*)din, stream);
     case OB_KIND_FINAL:
       if (!din) return ob_fail("final backward: din required");
+      if (l->d.dtype == 1)
+        return final_backward_bf16(l, slot, (const float*)dout, (__bf16*)din,
+                                   stream);
       return final_backward(l, slot, (const float*)dout, (float*)din, stream);
   }
   return ob_fail("backward: bad kind");

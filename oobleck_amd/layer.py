@@ -47,16 +47,21 @@ class Layer:
     def __init__(self, layer_id: int, cfg: ModelConfig, max_batch: int,
                  seq_len: int, n_slots: int, device: torch.device,
                  process_group: torch.distributed.ProcessGroup | None = None,
-                 init_style: str = "gpt2", seed: int = 42):
+                 init_style: str = "gpt2", seed: int = 42,
+                 dtype: str = "f32"):
         self.layer_id = layer_id
         self.cfg = cfg
         self.kind = cfg.layer_kind(layer_id)
         self.device = device
+        assert dtype in ("f32", "bf16")
+        self.dtype = dtype
+        self.act_dtype = torch.float32 if dtype == "f32" else torch.bfloat16
         ext = get_ext()
         self._desc = ObLayerDesc(
             kind=self.kind, n_embd=cfg.n_embd, n_head=cfg.n_head,
             n_positions=cfg.n_positions, vocab_size=cfg.vocab_size,
-            max_batch=max_batch, seq_len=seq_len, n_slots=n_slots)
+            max_batch=max_batch, seq_len=seq_len, n_slots=n_slots,
+            dtype=0 if dtype == "f32" else 1)
         n = ext.ob_layer_param_count(ctypes.byref(self._desc))
         assert n > 0
         init = init_layer_params(cfg, layer_id, seed, init_style).to(device)
@@ -85,6 +90,7 @@ class Layer:
                   "layer_bind")
             self._param_handle = _FlatParamHandle(flat, grad)
         self._batch = max_batch
+        self.refresh_weights()
 
     # -- compute ------------------------------------------------------------
     @property
@@ -123,6 +129,13 @@ class Layer:
         self.flat_grad.zero_()
         if self._sharded is not None:
             self._sharded.full_grad.zero_()
+
+    def refresh_weights(self) -> None:
+        """bf16 mode: re-cast the extension's bf16 weight shadows from the
+        fp32 master params (after init and after every optimizer step)."""
+        if self.dtype == "bf16":
+            check(get_ext().ob_layer_refresh_weights(self._h, _stream_ptr()),
+                  f"refresh_weights layer {self.layer_id}")
 
     # -- distributed surface (reference layer.py:272-291) --------------------
     def _shard_param(self, tensor: torch.Tensor, number: int) -> list[torch.Tensor]:

@@ -57,6 +57,9 @@ class FusedAdamW:
                 p.numel(), self.step_count, self.lr, self.betas[0],
                 self.betas[1], self.eps, self.weight_decay, stream),
                 f"adamw layer {l.layer_id}")
+            refresh = getattr(l, "refresh_weights", None)
+            if refresh is not None:
+                refresh()  # bf16 shadows follow the fp32 master update
 
     def zero_grad(self) -> None:
         for l in self.layers:

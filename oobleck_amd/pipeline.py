@@ -66,6 +66,9 @@ class PipelineExecution:
         self.total_loss: torch.Tensor | None = None
         self._optimizer = optimizer
         self._lr_scheduler = lr_scheduler
+        import torch as _torch
+        self._act_dtype = getattr(layers[0], "act_dtype", _torch.float32) \
+            if layers else _torch.float32
         # lazily-allocated working buffers
         self._tmp: list[torch.Tensor | None] = [None, None]
         self._out_bufs: dict[int, torch.Tensor] = {}
@@ -85,7 +88,7 @@ class PipelineExecution:
         shape = self._hidden_shape(batch)
         t = self._tmp[i]
         if t is None or t.shape[0] < batch:
-            t = torch.empty(shape, dtype=torch.float32,
+            t = torch.empty(shape, dtype=self._act_dtype,
                             device=self.pipeline.device)
             self._tmp[i] = t
         return t[:batch]
@@ -121,7 +124,7 @@ class PipelineExecution:
                     out = self._out_bufs.get(buffer_id)
                     if out is None or out.shape[0] < batch:
                         out = torch.empty(self._hidden_shape(batch),
-                                          dtype=torch.float32,
+                                          dtype=self._act_dtype,
                                           device=self.pipeline.device)
                         self._out_bufs[buffer_id] = out
                     out = out[:batch]
@@ -154,7 +157,7 @@ class PipelineExecution:
                     din = self._din_bufs.get(buffer_id)
                     if din is None or din.shape[0] < batch:
                         din = torch.empty(self._hidden_shape(batch),
-                                          dtype=torch.float32,
+                                          dtype=self._act_dtype,
                                           device=self.pipeline.device)
                         self._din_bufs[buffer_id] = din
                     din = din[:batch]
