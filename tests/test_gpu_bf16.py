@@ -173,7 +173,8 @@ def test_bf16_full_model_parity():
     mc, oc = ModelConfig(**dims), OracleConfig(**dims)
     B, S = 2, 96
     L = oc.n_layers_total
-    flats = [init_layer_params(oc, i, 300 + i) for i in range(L)]
+    flats = [init_layer_params(oc, oc.layer_kind(i), 300 + i)
+             for i in range(L)]
     layers = []
     for lid in range(L):
         layer = Layer(lid, mc, B, S, 1, torch.device(DEV), dtype="bf16")
