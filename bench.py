@@ -25,7 +25,8 @@ import torch
 import torch.distributed as dist
 
 TOPOLOGY = {1: (1, 1), 2: (2, 1), 4: (4, 1), 8: (4, 2)}  # gpus -> (stages, replicas)
-F32_MFMA_PEAK_TF = 157.3  # gfx950 f32 matrix peak (MI355X_MICROARCH.md)
+F32_MFMA_PEAK_TF = 157.3   # gfx950 f32 matrix peak (MI355X_MICROARCH.md)
+BF16_MFMA_PEAK_TF = 2500.0  # gfx950 bf16 dense MFMA peak (NOT the 2:1-sparse 5 PF)
 
 
 def build_world(args, device):
@@ -62,7 +63,7 @@ def build_world(args, device):
         if p.my_pipeline:
             def layer_factory(lid, pg, n_slots):
                 return Layer(lid, mc, tc.microbatch_size, tc.seq_len,
-                             n_slots, device)
+                             n_slots, device, dtype=args.dtype)
 
             def optimizer_factory(layers):
                 opt = FusedAdamW(layers, lr=tc.lr,
@@ -88,40 +89,56 @@ def train_step(my_pipeline, dp):
 
 
 def measure_roofline(args, device):
-    """Dominant-kernel roofline: the fp32 MFMA GEMM at its most-executed hot
-    shape (the MLP fc GEMM of config[0]: M=B*S=8192, N=4H=3072, K=H=768).
+    """Dominant-kernel roofline: the MFMA GEMM at its most-executed hot
+    shape (the MLP fc GEMM of config[0]: M=B*S=8192, N=4H=3072, K=H=768;
+    bf16 mode uses the transposed weight shadow -> TB operand).
     achieved = algorithmic FLOPs per launch / avg launch duration (HIP
     events on the launch stream).  DESIGN.md derives the per-unit figures."""
-    from tests.gpu_helpers import gemm
     M, N, K = args.microbatch * args.seq_len, 4 * 768, 768
-    A = torch.randn(M, K, device=device)
-    B = torch.randn(K, N, device=device)
-    C = torch.empty(M, N, device=device)
+    bf16 = getattr(args, "dtype", "f32") == "bf16"
+    if bf16:
+        import tests.test_gpu_bf16 as tb
+        A = torch.randn(M, K, device=device).bfloat16()
+        B = torch.randn(N, K, device=device).bfloat16()  # transposed shadow
+        C = torch.empty(M, N, device=device, dtype=torch.bfloat16)
+        def run():
+            tb.gemm_bf16(A, B, C, transB=1, M=M, N=N, K=K, lda=K, ldb=K,
+                         ldc=N)
+    else:
+        from tests.gpu_helpers import gemm
+        A = torch.randn(M, K, device=device)
+        B = torch.randn(K, N, device=device)
+        C = torch.empty(M, N, device=device)
+        def run():
+            gemm(A, B, C, M=M, N=N, K=K, lda=K, ldb=N, ldc=N)
     for _ in range(3):
-        gemm(A, B, C, M=M, N=N, K=K, lda=K, ldb=N, ldc=N)
+        run()
     torch.cuda.synchronize()
     start, end = torch.cuda.Event(True), torch.cuda.Event(True)
     reps = 50
     start.record()
     for _ in range(reps):
-        gemm(A, B, C, M=M, N=N, K=K, lda=K, ldb=N, ldc=N)
+        run()
     end.record()
     torch.cuda.synchronize()
     avg_s = start.elapsed_time(end) / 1000.0 / reps
     flops = 2.0 * M * N * K
     achieved_tf = flops / avg_s / 1e12
+    peak = BF16_MFMA_PEAK_TF if bf16 else F32_MFMA_PEAK_TF
     traffic = None
     pmc_file = os.path.join(os.path.dirname(os.path.abspath(__file__)),
                             "profiles", "pmc_gemm_fc.json")
-    if os.path.exists(pmc_file):
+    if not bf16 and os.path.exists(pmc_file):
         with open(pmc_file) as f:
             traffic = json.load(f).get("hbm_bytes_per_launch")
     return {
         "bound": "mfma", "achieved": round(achieved_tf, 2),
-        "peak": F32_MFMA_PEAK_TF, "unit": "TFLOP/s",
-        "frac": round(achieved_tf / F32_MFMA_PEAK_TF, 4),
+        "peak": peak, "unit": "TFLOP/s",
+        "frac": round(achieved_tf / peak, 4),
         "traffic": traffic,
-        "kernel": "k_gemm_f32 (MLP fc, M=8192 N=3072 K=768, fp32 MFMA)",
+        "kernel": ("k_gemm_bf16 (MLP fc, M=8192 N=3072 K=768, bf16 MFMA)"
+                   if bf16 else
+                   "k_gemm_f32 (MLP fc, M=8192 N=3072 K=768, fp32 MFMA)"),
         "avg_launch_ms": round(avg_s * 1e3, 4),
     }
 
@@ -162,6 +179,9 @@ def main():
     ap.add_argument("--seq-len", type=int, default=1024)
     ap.add_argument("--microbatch", type=int, default=8)
     ap.add_argument("--global-batch", type=int, default=128)
+    ap.add_argument("--dtype", choices=["f32", "bf16"], default="f32",
+                    help="compute dtype (f32 = the reference dtype; "
+                         "bf16 = mixed precision, SURVEY §8 f4)")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     ap.add_argument("--skip-roofline", action="store_true")
     ap.add_argument("--probe", choices=["gemm"], default=None,
@@ -221,11 +241,11 @@ def main():
             "higher_is_better": True,
             "scaling": "strong",  # global batch fixed at 128 as N grows
             "vs_baseline": None,  # reference publishes no numbers
-            "dtype": "f32",       # the reference's compute dtype (fp32)
+            "dtype": args.dtype,  # f32 = the reference compute dtype
             "data": "synthetic",
             "config": {
                 "workload": "gpt2.yaml GPT-2-small 1F1B, global microbatch "
-                            "128x1024 tokens, fp32",
+                            f"128x1024 tokens, {args.dtype}",
                 "model": "gpt2",
                 "global_batch": args.global_batch,
                 "seq_len": args.seq_len,
