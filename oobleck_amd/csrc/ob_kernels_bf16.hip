@@ -67,15 +67,18 @@ __device__ __forceinline__ __bf16 bf_extract(const uint4& v, int j) {
 }
 
 // direct staging: img[row][k] = src[row][k], k contiguous in memory.
-// 256 threads; thread t covers row = t/2, k-halfs (t%2)*8 and +16.
-template <bool EDGE>
-__device__ __forceinline__ void bf_stage_direct(__bf16* img,
-                                                const __bf16* src, int64_t ld,
-                                                int rmax, int kmax) {
-  const int r = threadIdx.x >> 1;  // 0..127
+// 256 threads; NR = tile rows (64 or 128).
+template <bool EDGE, int NR>
+__device__ __forceinline__ void bf_stage_direct_n(__bf16* img,
+                                                  const __bf16* src,
+                                                  int64_t ld, int rmax,
+                                                  int kmax) {
+  constexpr int TPR = 1024 / (NR * 4);  // k-chunks per thread pass
+  const int r = threadIdx.x / (256 / NR);
+  const int kb0 = (threadIdx.x % (256 / NR)) * 8;
 #pragma unroll
-  for (int it = 0; it < 2; ++it) {
-    const int kb = (threadIdx.x & 1) * 8 + it * 16;
+  for (int it = 0; it < (NR * 4) / 256; ++it) {
+    const int kb = kb0 + it * (256 / NR) * 8;
     uint4 v = {0, 0, 0, 0};
     if (!EDGE || (r < rmax && kb + 7 < kmax)) {
       v = *reinterpret_cast<const uint4*>(src + (int64_t)r * ld + kb);
@@ -84,21 +87,30 @@ __device__ __forceinline__ void bf_stage_direct(__bf16* img,
     }
     *reinterpret_cast<uint4*>(img + r * BF_LDS_K + kb) = v;
   }
+  (void)sizeof(char[TPR >= 0 ? 1 : -1]);
+}
+
+template <bool EDGE>
+__device__ __forceinline__ void bf_stage_direct(__bf16* img,
+                                                const __bf16* src, int64_t ld,
+                                                int rmax, int kmax) {
+  bf_stage_direct_n<EDGE, 128>(img, src, ld, rmax, kmax);
 }
 
 // transpose staging: img[row][k] = src[k][row] (src k-major, ld = src row
 // stride).  Reads 16 B along the row dim (coalesced), writes 8 scattered
-// 2-byte LDS stores.
-template <bool EDGE>
-__device__ __forceinline__ void bf_stage_transpose(__bf16* img,
-                                                   const __bf16* src,
-                                                   int64_t ld, int rmax,
-                                                   int kmax) {
-  const int rb = (threadIdx.x & 15) * 8;
-  const int k0 = threadIdx.x >> 4;  // 0..15
+// 2-byte LDS stores.  NR = tile rows.
+template <bool EDGE, int NR>
+__device__ __forceinline__ void bf_stage_transpose_n(__bf16* img,
+                                                     const __bf16* src,
+                                                     int64_t ld, int rmax,
+                                                     int kmax) {
+  constexpr int RGROUPS = NR / 8;  // 8-row chunks
+  const int rb = (threadIdx.x % RGROUPS) * 8;
+  const int k0 = threadIdx.x / RGROUPS;
 #pragma unroll
-  for (int it = 0; it < 2; ++it) {
-    const int k = k0 + it * 16;
+  for (int it = 0; it < (RGROUPS * 32) / 256; ++it) {
+    const int k = k0 + it * (256 / RGROUPS);
     uint4 v = {0, 0, 0, 0};
     bool any = true;
     if (!EDGE || (k < kmax && rb + 7 < rmax)) {
@@ -116,7 +128,15 @@ __device__ __forceinline__ void bf_stage_transpose(__bf16* img,
   }
 }
 
-template <bool TA, bool TB, int OUT, bool EDGE>
+template <bool EDGE>
+__device__ __forceinline__ void bf_stage_transpose(__bf16* img,
+                                                   const __bf16* src,
+                                                   int64_t ld, int rmax,
+                                                   int kmax) {
+  bf_stage_transpose_n<EDGE, 128>(img, src, ld, rmax, kmax);
+}
+
+template <bool TA, bool TB, int OUT, bool EDGE, int BN>
 __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     void* __restrict__ Cv, const float* __restrict__ bias,
@@ -124,12 +144,13 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
     int64_t ldb, int64_t ldc, int64_t sA1, int64_t sA2, int64_t sB1,
     int64_t sB2, int64_t sC1, int64_t sC2, int n2, float alpha, float beta,
     int nbn) {
+  constexpr int FN = BN / 64;  // B fragments per wave (1 or 2)
   __shared__ __bf16 As[2][BF_BM * BF_LDS_K];
-  __shared__ __bf16 Bs[2][BF_BN * BF_LDS_K];
+  __shared__ __bf16 Bs[2][BN * BF_LDS_K];
 
   const int tile = blockIdx.x;
   const int bm = tile / nbn, bn = tile % nbn;
-  const int m0 = bm * BF_BM, n0 = bn * BF_BN;
+  const int m0 = bm * BF_BM, n0 = bn * BN;
 
   const int z = blockIdx.z;
   const int i1 = z / n2, i2 = z % n2;
@@ -163,11 +184,11 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
       bf_stage_direct<EDGE>(As[BUF], A + (int64_t)m0 * lda + (KT), lda,       \
                             min(M - m0, BF_BM), kmax_);                       \
     if (TB)                                                                   \
-      bf_stage_direct<EDGE>(Bs[BUF], B + (int64_t)n0 * ldb + (KT), ldb,       \
-                            min(N - n0, BF_BN), kmax_);                       \
+      bf_stage_direct_n<EDGE, BN>(Bs[BUF], B + (int64_t)n0 * ldb + (KT), ldb, \
+                                  min(N - n0, BN), kmax_);                    \
     else                                                                      \
-      bf_stage_transpose<EDGE>(Bs[BUF], B + (int64_t)(KT)*ldb + n0, ldb,      \
-                               min(N - n0, BF_BN), kmax_);                    \
+      bf_stage_transpose_n<EDGE, BN>(Bs[BUF], B + (int64_t)(KT)*ldb + n0,     \
+                                     ldb, min(N - n0, BN), kmax_);            \
   }
 
   f32x16 acc00 = {}, acc01 = {}, acc10 = {}, acc11 = {};
@@ -180,13 +201,15 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
     const bf16x8 a1 = *reinterpret_cast<const bf16x8*>(                       \
         As[BUF] + (wr * 64 + 32 + il) * BF_LDS_K + kb);                       \
     const bf16x8 b0 = *reinterpret_cast<const bf16x8*>(                       \
-        Bs[BUF] + (wc * 64 + il) * BF_LDS_K + kb);                            \
-    const bf16x8 b1 = *reinterpret_cast<const bf16x8*>(                       \
-        Bs[BUF] + (wc * 64 + 32 + il) * BF_LDS_K + kb);                       \
+        Bs[BUF] + (wc * (BN / 2) + il) * BF_LDS_K + kb);                      \
     acc00 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b0, acc00, 0, 0, 0);  \
-    acc01 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b1, acc01, 0, 0, 0);  \
     acc10 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b0, acc10, 0, 0, 0);  \
-    acc11 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b1, acc11, 0, 0, 0);  \
+    if (FN == 2) {                                                            \
+      const bf16x8 b1 = *reinterpret_cast<const bf16x8*>(                     \
+          Bs[BUF] + (wc * (BN / 2) + 32 + il) * BF_LDS_K + kb);               \
+      acc01 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b1, acc01, 0, 0, 0);\
+      acc11 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b1, acc11, 0, 0, 0);\
+    }                                                                         \
   }
 
   OB_BF_STAGE(0, kbeg)
@@ -202,7 +225,7 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
 #undef OB_BF_STAGE
 #undef OB_BF_MFMA
 
-  const int mw = m0 + wr * 64, nw = n0 + wc * 64;
+  const int mw = m0 + wr * 64, nw = n0 + wc * (BN / 2);
 #define OB_BF_EPI(ACC, TI, TJ)                                                \
   {                                                                           \
     const int nn = nw + (TJ)*32 + il;                                         \
@@ -229,9 +252,11 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
     }                                                                         \
   }
   OB_BF_EPI(acc00, 0, 0)
-  OB_BF_EPI(acc01, 0, 1)
   OB_BF_EPI(acc10, 1, 0)
-  OB_BF_EPI(acc11, 1, 1)
+  if (FN == 2) {
+    OB_BF_EPI(acc01, 0, 1)
+    OB_BF_EPI(acc11, 1, 1)
+  }
 #undef OB_BF_EPI
 }
 
@@ -249,17 +274,23 @@ extern "C" int ob_gemm_bf16(int transA, int transB, int64_t M, int64_t N,
     return ob_fail("gemm_bf16: splitk needs atomic f32 out");
   // 16-byte staging requires 8-half-aligned leading dims and bases
   if ((lda | ldb) & 7) return ob_fail("gemm_bf16: lda/ldb must be 8-aligned");
+  const int BN = (N <= 64) ? 64 : 128;  // narrow tiles for head_dim GEMMs
   const int nbm = (int)((M + BF_BM - 1) / BF_BM);
-  const int nbn = (int)((N + BF_BN - 1) / BF_BN);
-  const bool edge = (M % BF_BM) || (N % BF_BN) || (K % BF_BK);
+  const int nbn = (int)((N + BN - 1) / BN);
+  const bool edge = (M % BF_BM) || (N % BN) || (K % BF_BK);
   dim3 grid(nbm * nbn, splitk, (unsigned)(n1 * n2));
   dim3 block(256);
-#define OB_BFG_L(TA_, TB_, OUT_, ED_)                                        \
-  k_gemm_bf16<TA_, TB_, OUT_, ED_><<<grid, block, 0, S(stream)>>>(           \
+#define OB_BFG_L2(TA_, TB_, OUT_, ED_, BN_)                                  \
+  k_gemm_bf16<TA_, TB_, OUT_, ED_, BN_><<<grid, block, 0, S(stream)>>>(      \
       (const __bf16*)A, (const __bf16*)B, C, (const float*)bias,             \
       (const __bf16*)residual, (int)M, (int)N, (int)K, lda, ldb, ldc,        \
       strideA1, strideA2, strideB1, strideB2, strideC1, strideC2, (int)n2,   \
       alpha, beta, nbn)
+#define OB_BFG_L(TA_, TB_, OUT_, ED_)                                        \
+  do {                                                                       \
+    if (BN == 64) OB_BFG_L2(TA_, TB_, OUT_, ED_, 64);                        \
+    else OB_BFG_L2(TA_, TB_, OUT_, ED_, 128);                                \
+  } while (0)
 #define OB_BFG_OUT(TA_, TB_)                                                 \
   do {                                                                       \
     if (out_kind == BF_OUT_BF16) {                                           \
@@ -282,6 +313,7 @@ extern "C" int ob_gemm_bf16(int transA, int transB, int64_t M, int64_t N,
   }
 #undef OB_BFG_OUT
 #undef OB_BFG_L
+#undef OB_BFG_L2
   OB_LAUNCH_CHECK();
   return 0;
 }
@@ -529,12 +561,56 @@ __global__ __launch_bounds__(256) void k_softmax_fwd_bf16(
   }
 }
 
+__device__ __forceinline__ uint4 bf_pack8(const float* v) {
+  uint4 o;
+  o.x = bf_bits((__bf16)v[0]) | (bf_bits((__bf16)v[1]) << 16);
+  o.y = bf_bits((__bf16)v[2]) | (bf_bits((__bf16)v[3]) << 16);
+  o.z = bf_bits((__bf16)v[4]) | (bf_bits((__bf16)v[5]) << 16);
+  o.w = bf_bits((__bf16)v[6]) | (bf_bits((__bf16)v[7]) << 16);
+  return o;
+}
+
+__global__ __launch_bounds__(256) void k_softmax_fwd_bf16_v8(
+    __bf16* __restrict__ scores, int Sq, float scale) {
+  __shared__ float lds4[4];
+  const int64_t rid = blockIdx.x;
+  const int row = (int)(rid % Sq);
+  __bf16* p = scores + rid * Sq;
+  const int valid = row + 1;
+  const int c8 = threadIdx.x * 8;
+  float v[8];
+  uint4 in = {0, 0, 0, 0};
+  if (c8 < Sq) in = *reinterpret_cast<const uint4*>(p + c8);
+  float mx = -INFINITY;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    v[j] = (c8 + j < valid) ? bf2f(bf_extract(in, j)) * scale : -INFINITY;
+    mx = fmaxf(mx, v[j]);
+  }
+  mx = bblock_max256(mx, lds4);
+  float sum = 0.f;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    v[j] = (v[j] == -INFINITY) ? 0.f : __expf(v[j] - mx);
+    sum += v[j];
+  }
+  sum = bblock_sum256(sum, lds4);
+  const float inv = 1.f / sum;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) v[j] *= inv;
+  if (c8 < Sq) *reinterpret_cast<uint4*>(p + c8) = bf_pack8(v);
+}
+
 extern "C" int ob_softmax_causal_fwd_bf16(void* scores, int64_t batch,
                                           int64_t Sq, float scale,
                                           void* stream) {
   if (Sq > 2048) return ob_fail("softmax_bf16: S > 2048 unsupported");
-  k_softmax_fwd_bf16<<<(unsigned)(batch * Sq), 256, 0, S(stream)>>>(
-      (__bf16*)scores, (int)Sq, scale);
+  if (Sq % 8 == 0)
+    k_softmax_fwd_bf16_v8<<<(unsigned)(batch * Sq), 256, 0, S(stream)>>>(
+        (__bf16*)scores, (int)Sq, scale);
+  else
+    k_softmax_fwd_bf16<<<(unsigned)(batch * Sq), 256, 0, S(stream)>>>(
+        (__bf16*)scores, (int)Sq, scale);
   OB_LAUNCH_CHECK();
   return 0;
 }
@@ -564,12 +640,45 @@ __global__ __launch_bounds__(256) void k_softmax_bwd_bf16(
   }
 }
 
+__global__ __launch_bounds__(256) void k_softmax_bwd_bf16_v8(
+    const __bf16* __restrict__ P, __bf16* __restrict__ dP, int Sq) {
+  __shared__ float lds4[4];
+  const int64_t rid = blockIdx.x;
+  const int row = (int)(rid % Sq);
+  const __bf16* pr = P + rid * Sq;
+  __bf16* dr = dP + rid * Sq;
+  const int valid = row + 1;
+  const int c8 = threadIdx.x * 8;
+  uint4 pin = {0, 0, 0, 0}, din = {0, 0, 0, 0};
+  if (c8 < Sq) {
+    pin = *reinterpret_cast<const uint4*>(pr + c8);
+    din = *reinterpret_cast<const uint4*>(dr + c8);
+  }
+  float pv[8], dv[8];
+  float t = 0.f;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    pv[j] = (c8 + j < valid) ? bf2f(bf_extract(pin, j)) : 0.f;
+    dv[j] = (c8 + j < valid) ? bf2f(bf_extract(din, j)) : 0.f;
+    t += pv[j] * dv[j];
+  }
+  t = bblock_sum256(t, lds4);
+  float out[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) out[j] = pv[j] * (dv[j] - t);
+  if (c8 < Sq) *reinterpret_cast<uint4*>(dr + c8) = bf_pack8(out);
+}
+
 extern "C" int ob_softmax_causal_bwd_bf16(const void* P, void* dP,
                                           int64_t batch, int64_t Sq,
                                           void* stream) {
   if (Sq > 2048) return ob_fail("softmax_bwd_bf16: S > 2048 unsupported");
-  k_softmax_bwd_bf16<<<(unsigned)(batch * Sq), 256, 0, S(stream)>>>(
-      (const __bf16*)P, (__bf16*)dP, (int)Sq);
+  if (Sq % 8 == 0)
+    k_softmax_bwd_bf16_v8<<<(unsigned)(batch * Sq), 256, 0, S(stream)>>>(
+        (const __bf16*)P, (__bf16*)dP, (int)Sq);
+  else
+    k_softmax_bwd_bf16<<<(unsigned)(batch * Sq), 256, 0, S(stream)>>>(
+        (const __bf16*)P, (__bf16*)dP, (int)Sq);
   OB_LAUNCH_CHECK();
   return 0;
 }
@@ -624,11 +733,34 @@ __global__ __launch_bounds__(256) void k_colsum_bf16(
   for (int64_t r = r0; r < r1; ++r) acc += bf2f(X[r * N + c]);
   atomicAdd(&db[c], acc);
 }
+// vectorized: each thread owns 8 contiguous columns (one uint4 per row)
+__global__ __launch_bounds__(256) void k_colsum_bf16_v8(
+    const __bf16* __restrict__ X, float* __restrict__ db, int64_t M,
+    int64_t N) {
+  const int64_t c8 = ((int64_t)blockIdx.x * 256 + threadIdx.x) * 8;
+  if (c8 >= N) return;
+  const int64_t r0 = (int64_t)blockIdx.y * 512;
+  const int64_t r1 = bmin64(M, r0 + 512);
+  float acc[8] = {0};
+  for (int64_t r = r0; r < r1; ++r) {
+    const uint4 v = *reinterpret_cast<const uint4*>(X + r * N + c8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[j] += bf2f(bf_extract(v, j));
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) atomicAdd(&db[c8 + j], acc[j]);
+}
 extern "C" int ob_colsum_bf16(const void* X, void* db, int64_t M, int64_t N,
                               void* stream) {
-  dim3 grid((unsigned)((N + 255) / 256), (unsigned)((M + 255) / 256));
-  k_colsum_bf16<<<grid, 256, 0, S(stream)>>>((const __bf16*)X, (float*)db, M,
-                                             N);
+  if (N % 8 == 0) {
+    dim3 grid((unsigned)((N / 8 + 255) / 256), (unsigned)((M + 511) / 512));
+    k_colsum_bf16_v8<<<grid, 256, 0, S(stream)>>>((const __bf16*)X,
+                                                  (float*)db, M, N);
+  } else {
+    dim3 grid((unsigned)((N + 255) / 256), (unsigned)((M + 255) / 256));
+    k_colsum_bf16<<<grid, 256, 0, S(stream)>>>((const __bf16*)X, (float*)db,
+                                               M, N);
+  }
   OB_LAUNCH_CHECK();
   return 0;
 }
