@@ -304,6 +304,28 @@ extern "C" int ob_gemm_bf16(int transA, int transB, int64_t M, int64_t N,
       else OB_BFG_L(TA_, TB_, BF_OUT_F32_ATOMIC, false);                     \
     }                                                                        \
   } while (0)
+  // fast NT path: async glds staging (declared below; interior 128-aligned
+  // shapes with 16-byte-aligned bases/strides only)
+  extern int ob_gemm_bf16_nt_dispatch(
+      const void* A, const void* B, void* C, const void* bias,
+      const void* residual, int64_t M, int64_t N, int64_t K, int64_t lda,
+      int64_t ldb, int64_t ldc, int64_t sA1, int64_t sA2, int64_t sB1,
+      int64_t sB2, int64_t sC1, int64_t sC2, int64_t n1, int64_t n2,
+      float alpha, float beta, int out_kind, int splitk, void* stream);
+  static const bool no_glds = [] {
+    const char* e = getenv("OB_BF16_NOGLDS");
+    return e && e[0] == '1';
+  }();
+  const bool aligned16 =
+      ((reinterpret_cast<uintptr_t>(A) | reinterpret_cast<uintptr_t>(B)) %
+           16 == 0) &&
+      ((strideA1 | strideA2 | strideB1 | strideB2) % 8 == 0);
+  if (!no_glds && transA == 0 && transB == 1 && !edge && aligned16 &&
+      K >= 2 * BF_BK)
+    return ob_gemm_bf16_nt_dispatch(A, B, C, bias, residual, M, N, K, lda,
+                                    ldb, ldc, strideA1, strideA2, strideB1,
+                                    strideB2, strideC1, strideC2, n1, n2,
+                                    alpha, beta, out_kind, splitk, stream);
   const int sel = (transA ? 2 : 0) | (transB ? 1 : 0);
   switch (sel) {
     case 0: OB_BFG_OUT(false, false); break;
@@ -909,6 +931,178 @@ extern "C" int ob_ce_bwd_bf16(void* logits, const void* labels,
   k_ce_bwd_bf16<<<grid, 256, 0, S(stream)>>>(
       (__bf16*)logits, (const int64_t*)labels, (const float*)lse,
       (const float*)dloss, B * Sq, (int)Sq, (int)V, ld);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+// ---------------------------------------------------------------------------
+// Fast NT path (A [M,K] direct, B [N,K] direct — the fwd/dX hot shapes fed
+// by the transposed weight shadows): async global_load_lds width-16 staging
+// into lane-linear [row][32-half] images with a SOURCE-side slot swizzle
+// (rule 21: glds writes base+lane*16, so the bank swizzle moves to the
+// per-lane global address and the matching XOR on the read).  Swizzle key
+// (row&3)^((row>>2)&3) makes the 16-lane ds_read_b128 fragment groups hit
+// 16 distinct bank quads.  2-phase: issue next tile's glds, MFMA current,
+// one __syncthreads per tile (its implicit vmcnt(0) drains the DMA —
+// exactly the guide's minimum 2-phase glds recipe).  No staging registers
+// -> 4 blocks/CU (the reg-staged kernel measured 62% SQ_WAIT_ANY: parked
+// on HBM latency that 256-cycle bf16 MFMA segments cannot hide).
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ int bf_swz_key(int row) {
+  return (row & 3) ^ ((row >> 2) & 3);
+}
+
+template <int OUT>
+__global__ __launch_bounds__(256, 3) void k_gemm_bf16_nt_glds(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    void* __restrict__ Cv, const float* __restrict__ bias,
+    const __bf16* __restrict__ R, int M, int N, int K, int64_t lda,
+    int64_t ldb, int64_t ldc, int64_t sA1, int64_t sA2, int64_t sB1,
+    int64_t sB2, int64_t sC1, int64_t sC2, int n2, float alpha, float beta,
+    int nbn) {
+  // 8 KB per operand per buffer, lane-linear
+  __shared__ __bf16 As[2][128 * 32];
+  __shared__ __bf16 Bs[2][128 * 32];
+
+  const int tile = blockIdx.x;
+  const int bm = tile / nbn, bn = tile % nbn;
+  const int m0 = bm * 128, n0 = bn * 128;
+
+  const int z = blockIdx.z;
+  const int i1 = z / n2, i2 = z % n2;
+  A += (int64_t)i1 * sA1 + (int64_t)i2 * sA2;
+  B += (int64_t)i1 * sB1 + (int64_t)i2 * sB2;
+  float* Cf = reinterpret_cast<float*>(Cv);
+  __bf16* Cb = reinterpret_cast<__bf16*>(Cv);
+  const int64_t coff = (int64_t)i1 * sC1 + (int64_t)i2 * sC2;
+  Cf += coff;
+  Cb += coff;
+  if (R) R += coff;
+
+  const int splitk = gridDim.y;
+  const int kchunk = ((K + splitk * BF_BK - 1) / (splitk * BF_BK)) * BF_BK;
+  const int kbeg = blockIdx.y * kchunk;
+  const int kend = min(K, kbeg + kchunk);
+  if (kbeg >= kend) return;
+
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int wr = w >> 1, wc = w & 1;
+  const int il = lane & 31, kh = lane >> 5;
+
+  // glds geometry: one 1-KiB wave instruction covers 16 rows x 64 B.
+  // wave w stages rows [w*32, w*32+32) of each operand = 2 instructions.
+  // lane l within an instruction: row16 = l/4, slot = l%4; the element
+  // block written at [row][slot] must come from source half-range
+  // (slot ^ key(row)) * 8.
+  const int g_row16 = lane >> 2;        // 0..15
+  const int g_slot = lane & 3;          // 16-B slot within the 64-B row
+#define OB_NT_GLDS(BUF, KT)                                                   \
+  {                                                                           \
+    _Pragma("unroll") for (int half = 0; half < 2; ++half) {                  \
+      const int row = w * 32 + half * 16 + g_row16;                           \
+      const int srchalf = (g_slot ^ bf_swz_key(row)) * 8;                     \
+      const __bf16* asrc = A + (int64_t)(m0 + row) * lda + (KT) + srchalf;    \
+      const __bf16* bsrc = B + (int64_t)(n0 + row) * ldb + (KT) + srchalf;    \
+      auto albase = (__attribute__((address_space(3))) void*)                 \
+          (&As[BUF][(w * 32 + half * 16) * 32]);                              \
+      auto blbase = (__attribute__((address_space(3))) void*)                 \
+          (&Bs[BUF][(w * 32 + half * 16) * 32]);                              \
+      __builtin_amdgcn_global_load_lds(                                       \
+          (const __attribute__((address_space(1))) void*)asrc, albase, 16, 0, \
+          0);                                                                 \
+      __builtin_amdgcn_global_load_lds(                                       \
+          (const __attribute__((address_space(1))) void*)bsrc, blbase, 16, 0, \
+          0);                                                                 \
+    }                                                                         \
+  }
+
+  f32x16 acc00 = {}, acc01 = {}, acc10 = {}, acc11 = {};
+
+  // fragment read with the matching slot XOR
+#define OB_NT_FRAG(IMG, ROW, SLOT)                                            \
+  (*reinterpret_cast<const bf16x8*>(                                          \
+      IMG + (ROW)*32 + (((SLOT) ^ bf_swz_key(ROW)) * 8)))
+
+#define OB_NT_MFMA(BUF)                                                       \
+  _Pragma("unroll") for (int ks = 0; ks < 2; ++ks) {                          \
+    const int slot = ks * 2 + kh;                                             \
+    const int ra0 = wr * 64 + il, ra1 = wr * 64 + 32 + il;                    \
+    const int rb0 = wc * 64 + il, rb1 = wc * 64 + 32 + il;                    \
+    const bf16x8 a0 = OB_NT_FRAG(As[BUF], ra0, slot);                         \
+    const bf16x8 a1 = OB_NT_FRAG(As[BUF], ra1, slot);                         \
+    const bf16x8 b0 = OB_NT_FRAG(Bs[BUF], rb0, slot);                         \
+    const bf16x8 b1 = OB_NT_FRAG(Bs[BUF], rb1, slot);                         \
+    acc00 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b0, acc00, 0, 0, 0);  \
+    acc01 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b1, acc01, 0, 0, 0);  \
+    acc10 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b0, acc10, 0, 0, 0);  \
+    acc11 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b1, acc11, 0, 0, 0);  \
+  }
+
+  OB_NT_GLDS(0, kbeg)
+  __syncthreads();
+  int cur = 0;
+  for (int kt = kbeg; kt + BF_BK < kend; kt += BF_BK) {
+    OB_NT_GLDS(cur ^ 1, kt + BF_BK)
+    OB_NT_MFMA(cur)
+    __syncthreads();  // implicit vmcnt(0): next buffer's DMA has landed
+    cur ^= 1;
+  }
+  OB_NT_MFMA(cur)
+#undef OB_NT_GLDS
+#undef OB_NT_FRAG
+#undef OB_NT_MFMA
+
+  const int mw = m0 + wr * 64, nw = n0 + wc * 64;
+#define OB_NT_EPI(ACC, TI, TJ)                                                \
+  {                                                                           \
+    const int nn = nw + (TJ)*32 + il;                                         \
+    _Pragma("unroll") for (int r = 0; r < 16; ++r) {                          \
+      const int mm = mw + (TI)*32 + (r & 3) + 8 * (r >> 2) + 4 * kh;          \
+      float v = alpha * ACC[r];                                               \
+      if (OUT == BF_OUT_F32_ATOMIC) {                                         \
+        atomicAdd(&Cf[(int64_t)mm * ldc + nn], v);                            \
+      } else {                                                                \
+        if (bias) v += bias[nn];                                              \
+        if (R) v += bf2f(R[(int64_t)mm * ldc + nn]);                          \
+        if (OUT == BF_OUT_F32) {                                              \
+          if (beta != 0.f) v += beta * Cf[(int64_t)mm * ldc + nn];            \
+          Cf[(int64_t)mm * ldc + nn] = v;                                     \
+        } else {                                                              \
+          if (beta != 0.f) v += beta * bf2f(Cb[(int64_t)mm * ldc + nn]);      \
+          Cb[(int64_t)mm * ldc + nn] = (__bf16)v;                             \
+        }                                                                     \
+      }                                                                       \
+    }                                                                         \
+  }
+  OB_NT_EPI(acc00, 0, 0)
+  OB_NT_EPI(acc01, 0, 1)
+  OB_NT_EPI(acc10, 1, 0)
+  OB_NT_EPI(acc11, 1, 1)
+#undef OB_NT_EPI
+}
+
+int ob_gemm_bf16_nt_dispatch(const void* A, const void* B, void* C,
+                             const void* bias, const void* residual,
+                             int64_t M, int64_t N, int64_t K, int64_t lda,
+                             int64_t ldb, int64_t ldc, int64_t sA1,
+                             int64_t sA2, int64_t sB1, int64_t sB2,
+                             int64_t sC1, int64_t sC2, int64_t n1, int64_t n2,
+                             float alpha, float beta, int out_kind, int splitk,
+                             void* stream) {
+  const int nbm = (int)(M / 128), nbn = (int)(N / 128);
+  dim3 grid(nbm * nbn, splitk, (unsigned)(n1 * n2));
+  dim3 block(256);
+#define OB_NTG(OUT_)                                                         \
+  k_gemm_bf16_nt_glds<OUT_><<<grid, block, 0, S(stream)>>>(                  \
+      (const __bf16*)A, (const __bf16*)B, C, (const float*)bias,             \
+      (const __bf16*)residual, (int)M, (int)N, (int)K, lda, ldb, ldc, sA1,   \
+      sA2, sB1, sB2, sC1, sC2, (int)n2, alpha, beta, nbn)
+  if (out_kind == BF_OUT_BF16) OB_NTG(BF_OUT_BF16);
+  else if (out_kind == BF_OUT_F32) OB_NTG(BF_OUT_F32);
+  else OB_NTG(BF_OUT_F32_ATOMIC);
+#undef OB_NTG
   OB_LAUNCH_CHECK();
   return 0;
 }
