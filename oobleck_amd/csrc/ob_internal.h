@@ -72,4 +72,6 @@ int ob_ce_fwd_bf16(const void* logits, const void* labels, void* lse,
 int ob_ce_bwd_bf16(void* logits, const void* labels, const void* lse,
                    const void* dloss, int64_t B, int64_t Sq, int64_t V,
                    int64_t ld, void* stream);
+int ob_transpose_bf16(const void* in, void* out, int64_t R, int64_t C,
+                      void* stream);
 }

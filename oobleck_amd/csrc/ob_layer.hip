@@ -29,7 +29,10 @@ struct Workspace {
   float* bsh2 = nullptr;   // [B, S, H]
   float* dqkv = nullptr;   // [B, S, 3H]
   float* b4h = nullptr;    // [B, S, 4H]
-  int64_t sz_dp = 0, sz_bsh = 0, sz_bsh2 = 0, sz_dqkv = 0, sz_b4h = 0;
+  float* t1 = nullptr;     // bf16 [dim, BS] transposed operand (as floats)
+  float* t2 = nullptr;
+  int64_t sz_dp = 0, sz_bsh = 0, sz_bsh2 = 0, sz_dqkv = 0, sz_b4h = 0,
+          sz_t1 = 0, sz_t2 = 0;
 };
 static Workspace g_ws;
 
@@ -192,6 +195,10 @@ extern "C" int ob_layer_create(const ob_layer_desc* d, ob_layer_t* out) {
     if (ws_ensure(&g_ws.dp, &g_ws.sz_dp, Bm * nh * Sq * Sq)) return 1;
     if (ws_ensure(&g_ws.dqkv, &g_ws.sz_dqkv, BS * 3 * H)) return 1;
     if (ws_ensure(&g_ws.b4h, &g_ws.sz_b4h, BS * 4 * H)) return 1;
+    if (d->dtype == 1) {
+      if (ws_ensure(&g_ws.t1, &g_ws.sz_t1, (4 * H * BS + 1) / 2)) return 1;
+      if (ws_ensure(&g_ws.t2, &g_ws.sz_t2, (4 * H * BS + 1) / 2)) return 1;
+    }
   }
   if (d->kind != OB_KIND_EMBED) {
     const int64_t need = BSH;
@@ -415,6 +422,25 @@ static int gemm_bf(int tA, int tB, int64_t M, int64_t N, int64_t K,
                       splitk, stream);
 }
 
+// weight-grad GEMM for bf16: materialize X^T / dY^T (cheap HBM transpose)
+// so the GEMM runs on the fast NT glds path instead of the
+// transpose-staged TN case (~150 TF measured there).
+static int dw_bf16(const __bf16* act, int64_t actw, const __bf16* dY,
+                   int64_t dyw, int64_t BS, float* gout, int64_t ldc,
+                   void* stream) {
+  if ((actw % 128) || (dyw % 128) || (BS % 128))
+    return gemm_bf(1, 0, actw, dyw, BS, 1.f, act, actw, 0, 0, dY, dyw, 0, 0,
+                   gout, ldc, 0, 0, 1, 1, nullptr, nullptr, 2,
+                   pick_splitk(actw, dyw, BS), stream);
+  __bf16* XT = (__bf16*)g_ws.t1;
+  __bf16* DYT = (__bf16*)g_ws.t2;
+  if (ob_transpose_bf16(act, XT, BS, actw, stream)) return 1;
+  if (ob_transpose_bf16(dY, DYT, BS, dyw, stream)) return 1;
+  return gemm_bf(0, 1, actw, dyw, BS, 1.f, XT, BS, 0, 0, DYT, BS, 0, 0, gout,
+                 ldc, 0, 0, 1, 1, nullptr, nullptr, 2,
+                 pick_splitk(actw, dyw, BS), stream);
+}
+
 static int block_forward_bf16(ob_layer* l, int slot, const __bf16* in,
                               __bf16* out, void* stream) {
   const int64_t Sq = l->d.seq_len, H = l->d.n_embd, nh = l->d.n_head;
@@ -537,15 +563,11 @@ static int block_backward_bf16(ob_layer* l, int slot, const __bf16* dout,
   if (gemm_bf(0, 1, BS, 4 * H, H, 1.f, dout, H, 0, 0, sh + l->sh_mp, H, 0, 0,
               DY4, 4 * H, 0, 0, 1, 1, nullptr, nullptr, 0, 1, stream))
     return 1;
-  if (gemm_bf(1, 0, 4 * H, H, BS, 1.f, gact, 4 * H, 0, 0, dout, H, 0, 0,
-              g + bp.w_mlpproj, H, 0, 0, 1, 1, nullptr, nullptr, 2,
-              pick_splitk(4 * H, H, BS), stream))
+  if (dw_bf16(gact, 4 * H, dout, H, BS, g + bp.w_mlpproj, H, stream))
     return 1;
   if (ob_colsum_bf16(dout, g + bp.b_mlpproj, BS, H, stream)) return 1;
   if (ob_gelu_bwd_bf16(u, DY4, DY4, BS * 4 * H, stream)) return 1;
-  if (gemm_bf(1, 0, H, 4 * H, BS, 1.f, ln2, H, 0, 0, DY4, 4 * H, 0, 0,
-              g + bp.w_fc, 4 * H, 0, 0, 1, 1, nullptr, nullptr, 2,
-              pick_splitk(H, 4 * H, BS), stream))
+  if (dw_bf16(ln2, H, DY4, 4 * H, BS, g + bp.w_fc, 4 * H, stream))
     return 1;
   if (ob_colsum_bf16(DY4, g + bp.b_fc, BS, 4 * H, stream)) return 1;
   if (gemm_bf(0, 1, BS, H, 4 * H, 1.f, DY4, 4 * H, 0, 0, sh + l->sh_fc,
@@ -560,9 +582,7 @@ static int block_backward_bf16(ob_layer* l, int slot, const __bf16* dout,
   if (gemm_bf(0, 1, BS, H, H, 1.f, din, H, 0, 0, sh + l->sh_ap, H, 0, 0,
               DATT, H, 0, 0, 1, 1, nullptr, nullptr, 0, 1, stream))
     return 1;
-  if (gemm_bf(1, 0, H, H, BS, 1.f, am, H, 0, 0, din, H, 0, 0,
-              g + bp.w_attnproj, H, 0, 0, 1, 1, nullptr, nullptr, 2,
-              pick_splitk(H, H, BS), stream))
+  if (dw_bf16(am, H, din, H, BS, g + bp.w_attnproj, H, stream))
     return 1;
   if (ob_colsum_bf16(din, g + bp.b_attnproj, BS, H, stream)) return 1;
   // ---- attention core ----
@@ -585,9 +605,7 @@ static int block_backward_bf16(ob_layer* l, int slot, const __bf16* dout,
     return 1;
   // ---- QKV projection ----
   if (ob_colsum_bf16(DQKV, g + bp.b_qkv, BS, 3 * H, stream)) return 1;
-  if (gemm_bf(1, 0, H, 3 * H, BS, 1.f, ln1, H, 0, 0, DQKV, 3 * H, 0, 0,
-              g + bp.w_qkv, 3 * H, 0, 0, 1, 1, nullptr, nullptr, 2,
-              pick_splitk(H, 3 * H, BS), stream))
+  if (dw_bf16(ln1, H, DQKV, 3 * H, BS, g + bp.w_qkv, 3 * H, stream))
     return 1;
   if (gemm_bf(0, 1, BS, H, 3 * H, 1.f, DQKV, 3 * H, 0, 0, sh + l->sh_qkv,
               3 * H, 0, 0, DLN, H, 0, 0, 1, 1, nullptr, nullptr, 0, 1,
