@@ -304,14 +304,9 @@ extern "C" int ob_gemm_bf16(int transA, int transB, int64_t M, int64_t N,
       else OB_BFG_L(TA_, TB_, BF_OUT_F32_ATOMIC, false);                     \
     }                                                                        \
   } while (0)
-  // fast NT path: async glds staging (declared below; interior 128-aligned
-  // shapes with 16-byte-aligned bases/strides only)
-  extern int ob_gemm_bf16_nt_dispatch(
-      const void* A, const void* B, void* C, const void* bias,
-      const void* residual, int64_t M, int64_t N, int64_t K, int64_t lda,
-      int64_t ldb, int64_t ldc, int64_t sA1, int64_t sA2, int64_t sB1,
-      int64_t sB2, int64_t sC1, int64_t sC2, int64_t n1, int64_t n2,
-      float alpha, float beta, int out_kind, int splitk, void* stream);
+  // fast NT path: async glds staging (ob_internal.h declares the
+  // dispatch; interior 128-aligned shapes with 16-byte-aligned
+  // bases/strides only)
   static const bool no_glds = [] {
     const char* e = getenv("OB_BF16_NOGLDS");
     return e && e[0] == '1';
@@ -325,7 +320,7 @@ extern "C" int ob_gemm_bf16(int transA, int transB, int64_t M, int64_t N,
     return ob_gemm_bf16_nt_dispatch(A, B, C, bias, residual, M, N, K, lda,
                                     ldb, ldc, strideA1, strideA2, strideB1,
                                     strideB2, strideC1, strideC2, n1, n2,
-                                    alpha, beta, out_kind, splitk, stream);
+                                    alpha, beta, out_kind, splitk, stream, M);
   const int sel = (transA ? 2 : 0) | (transB ? 1 : 0);
   switch (sel) {
     case 0: OB_BFG_OUT(false, false); break;
@@ -960,7 +955,7 @@ __global__ __launch_bounds__(256, 3) void k_gemm_bf16_nt_glds(
     const __bf16* __restrict__ R, int M, int N, int K, int64_t lda,
     int64_t ldb, int64_t ldc, int64_t sA1, int64_t sA2, int64_t sB1,
     int64_t sB2, int64_t sC1, int64_t sC2, int n2, float alpha, float beta,
-    int nbn) {
+    int nbn, int Mr) {
   // 8 KB per operand per buffer, lane-linear; 3 buffers so two tiles'
   // DMAs stay in flight across the (raw) barrier — counted vmcnt(8)
   // waits only for the current tile's pieces (guide: 3-buf span +83%
@@ -1084,6 +1079,7 @@ __global__ __launch_bounds__(256, 3) void k_gemm_bf16_nt_glds(
     const int nn = nw + (TJ)*32 + il;                                         \
     _Pragma("unroll") for (int r = 0; r < 16; ++r) {                          \
       const int mm = mw + (TI)*32 + (r & 3) + 8 * (r >> 2) + 4 * kh;          \
+      if (mm >= Mr) continue;                                                 \
       float v = alpha * ACC[r];                                               \
       if (OUT == BF_OUT_F32_ATOMIC) {                                         \
         atomicAdd(&Cf[(int64_t)mm * ldc + nn], v);                            \
@@ -1114,7 +1110,7 @@ int ob_gemm_bf16_nt_dispatch(const void* A, const void* B, void* C,
                              int64_t sA2, int64_t sB1, int64_t sB2,
                              int64_t sC1, int64_t sC2, int64_t n1, int64_t n2,
                              float alpha, float beta, int out_kind, int splitk,
-                             void* stream) {
+                             void* stream, int64_t Mr) {
   const int nbm = (int)(M / 128), nbn = (int)(N / 128);
   dim3 grid(nbm * nbn, splitk, (unsigned)(n1 * n2));
   dim3 block(256);
@@ -1122,7 +1118,7 @@ int ob_gemm_bf16_nt_dispatch(const void* A, const void* B, void* C,
   k_gemm_bf16_nt_glds<OUT_><<<grid, block, 0, S(stream)>>>(                  \
       (const __bf16*)A, (const __bf16*)B, C, (const float*)bias,             \
       (const __bf16*)residual, (int)M, (int)N, (int)K, lda, ldb, ldc, sA1,   \
-      sA2, sB1, sB2, sC1, sC2, (int)n2, alpha, beta, nbn)
+      sA2, sB1, sB2, sC1, sC2, (int)n2, alpha, beta, nbn, (int)Mr)
   if (out_kind == BF_OUT_BF16) OB_NTG(BF_OUT_BF16);
   else if (out_kind == BF_OUT_F32) OB_NTG(BF_OUT_F32);
   else OB_NTG(BF_OUT_F32_ATOMIC);

@@ -167,7 +167,11 @@ extern "C" int ob_layer_create(const ob_layer_desc* d, ob_layer_t* out) {
       return ob_fail("create: ids hipMalloc failed");
     }
   }
-  l->v_pad = (V + 7) / 8 * 8;
+  // pad vocab to a 128 multiple: makes the lm_head GEMM family
+  // (logits fwd, d_lnout, dW_lm via transposes) interior shapes for
+  // the glds path; shadow pad rows stay zero so padded logits/grads
+  // are exactly zero
+  l->v_pad = (V + 127) / 128 * 128;
   if (d->dtype == 1) {
     int64_t o2 = 0;
     if (d->kind == OB_KIND_BLOCK) {
@@ -180,7 +184,7 @@ extern "C" int ob_layer_create(const ob_layer_desc* d, ob_layer_t* out) {
       l->sh_mp = o2; o2 += 4 * H * H;
       l->sh_mp_t = o2; o2 += 4 * H * H;
     } else if (d->kind == OB_KIND_FINAL) {
-      l->sh_lm = o2; o2 += V * H;
+      l->sh_lm = o2; o2 += l->v_pad * H;   // pad rows zeroed at alloc
       l->sh_lm_t = o2; o2 += H * l->v_pad;
     }
     l->shadow_count = o2;
@@ -199,6 +203,10 @@ extern "C" int ob_layer_create(const ob_layer_desc* d, ob_layer_t* out) {
       if (ws_ensure(&g_ws.t1, &g_ws.sz_t1, (4 * H * BS + 1) / 2)) return 1;
       if (ws_ensure(&g_ws.t2, &g_ws.sz_t2, (4 * H * BS + 1) / 2)) return 1;
     }
+  }
+  if (d->kind == OB_KIND_FINAL && d->dtype == 1) {
+    if (ws_ensure(&g_ws.t1, &g_ws.sz_t1, (l->v_pad * BS + 1) / 2)) return 1;
+    if (ws_ensure(&g_ws.t2, &g_ws.sz_t2, (4 * H * BS + 1) / 2)) return 1;
   }
   if (d->kind != OB_KIND_EMBED) {
     const int64_t need = BSH;
@@ -518,9 +526,10 @@ static int final_forward_bf16(ob_layer* l, int slot, const __bf16* in,
                             st + l->o_rstd1, BS, H, 1e-5f, stream))
     return 1;
   __bf16* logits = (__bf16*)(st + l->o_logits);
-  if (gemm_bf(0, 1, BS, V, H, 1.f, lnf, H, 0, 0, l->shadows + l->sh_lm, H, 0,
-              0, logits, l->v_pad, 0, 0, 1, 1, nullptr, nullptr, 0, 1,
-              stream))
+  // N = v_pad (zero-padded shadow rows -> padded logits are exactly 0)
+  if (gemm_bf(0, 1, BS, l->v_pad, H, 1.f, lnf, H, 0, 0,
+              l->shadows + l->sh_lm, H, 0, 0, logits, l->v_pad, 0, 0, 1, 1,
+              nullptr, nullptr, 0, 1, stream))
     return 1;
   OB_HIP(hipMemsetAsync(out, 0, sizeof(float), S(stream)));
   if (ob_ce_fwd_bf16(logits, labs, st + l->o_lse, out, B, Sq, V, l->v_pad,
@@ -634,10 +643,21 @@ static int final_backward_bf16(ob_layer* l, int slot, const float* dout,
   if (ob_ce_bwd_bf16(logits, labs, st + l->o_lse, dout, B, Sq, V, l->v_pad,
                      stream))
     return 1;
-  if (gemm_bf(1, 0, V, H, BS, 1.f, logits, l->v_pad, 0, 0, lnf, H, 0, 0,
-              g + 2 * H, H, 0, 0, 1, 1, nullptr, nullptr, 2, 1, stream))
-    return 1;
-  if (gemm_bf(0, 1, BS, H, V, 1.f, logits, l->v_pad, 0, 0,
+  // dW_lm: transpose dlogits and lnf so the GEMM runs on the glds path;
+  // tile over v_pad rows, store-guard at V (pad rows of dlogits are 0
+  // anyway, but they have no grad slot).
+  {
+    __bf16* DLT = (__bf16*)g_ws.t1;
+    __bf16* LNT = (__bf16*)g_ws.t2;
+    if (ob_transpose_bf16(logits, DLT, BS, l->v_pad, stream)) return 1;
+    if (ob_transpose_bf16(lnf, LNT, BS, H, stream)) return 1;
+    if (ob_gemm_bf16_nt_dispatch(DLT, LNT, g + 2 * H, nullptr, nullptr,
+                                 l->v_pad, H, BS, BS, BS, H, 0, 0, 0, 0, 0,
+                                 0, 1, 1, 1.f, 0.f, 2, 2, stream, V))
+      return 1;
+  }
+  // d_lnout: K = v_pad (padded dlogits cols and shadow^T cols are zero)
+  if (gemm_bf(0, 1, BS, H, l->v_pad, 1.f, logits, l->v_pad, 0, 0,
               l->shadows + l->sh_lm_t, l->v_pad, 0, 0, DLN, H, 0, 0, 1, 1,
               nullptr, nullptr, 0, 1, stream))
     return 1;
