@@ -578,6 +578,7 @@ __global__ __launch_bounds__(256) void k_softmax_fwd_bf16(
   }
 }
 
+__device__ __forceinline__ uint4 bf_pack8_(const float* v);
 __device__ __forceinline__ uint4 bf_pack8(const float* v) {
   uint4 o;
   o.x = bf_bits((__bf16)v[0]) | (bf_bits((__bf16)v[1]) << 16);
@@ -585,6 +586,10 @@ __device__ __forceinline__ uint4 bf_pack8(const float* v) {
   o.z = bf_bits((__bf16)v[4]) | (bf_bits((__bf16)v[5]) << 16);
   o.w = bf_bits((__bf16)v[6]) | (bf_bits((__bf16)v[7]) << 16);
   return o;
+}
+
+__device__ __forceinline__ uint4 bf_pack8_(const float* v) {
+  return bf_pack8(v);
 }
 
 __global__ __launch_bounds__(256) void k_softmax_fwd_bf16_v8(
@@ -703,38 +708,56 @@ extern "C" int ob_softmax_causal_bwd_bf16(const void* P, void* dP,
 #define BGELU_K 0.7978845608028654f
 #define BGELU_C 0.044715f
 
+// 8-wide (uint4) gelu: bf16 scalar loads cost ~2-2.5x (guide G13)
 __global__ __launch_bounds__(256) void k_gelu_fwd_bf16(
-    const __bf16* __restrict__ u, __bf16* __restrict__ g, int64_t n) {
-  for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < n;
+    const __bf16* __restrict__ u, __bf16* __restrict__ g, int64_t n8) {
+  for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < n8;
        i += (int64_t)gridDim.x * 256) {
-    const float x = bf2f(u[i]);
-    const float t = tanhf(BGELU_K * (x + BGELU_C * x * x * x));
-    g[i] = (__bf16)(0.5f * x * (1.f + t));
+    const uint4 in = *reinterpret_cast<const uint4*>(u + i * 8);
+    float out[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float x = bf2f(bf_extract(in, j));
+      const float t = tanhf(BGELU_K * (x + BGELU_C * x * x * x));
+      out[j] = 0.5f * x * (1.f + t);
+    }
+    *reinterpret_cast<uint4*>(g + i * 8) = bf_pack8_(out);
   }
 }
 __global__ __launch_bounds__(256) void k_gelu_bwd_bf16(
     const __bf16* __restrict__ u, const __bf16* __restrict__ dg,
-    __bf16* __restrict__ du, int64_t n) {
-  for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < n;
+    __bf16* __restrict__ du, int64_t n8) {
+  for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < n8;
        i += (int64_t)gridDim.x * 256) {
-    const float x = bf2f(u[i]);
-    const float t = tanhf(BGELU_K * (x + BGELU_C * x * x * x));
-    const float d = 0.5f * (1.f + t) +
-                    0.5f * x * (1.f - t * t) * BGELU_K * (1.f + 3.f * BGELU_C * x * x);
-    du[i] = (__bf16)(bf2f(dg[i]) * d);
+    const uint4 uin = *reinterpret_cast<const uint4*>(u + i * 8);
+    const uint4 gin = *reinterpret_cast<const uint4*>(dg + i * 8);
+    float out[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float x = bf2f(bf_extract(uin, j));
+      const float t = tanhf(BGELU_K * (x + BGELU_C * x * x * x));
+      const float d = 0.5f * (1.f + t) +
+                      0.5f * x * (1.f - t * t) * BGELU_K *
+                          (1.f + 3.f * BGELU_C * x * x);
+      out[j] = bf2f(bf_extract(gin, j)) * d;
+    }
+    *reinterpret_cast<uint4*>(du + i * 8) = bf_pack8_(out);
   }
 }
 extern "C" int ob_gelu_fwd_bf16(const void* u, void* g, int64_t n,
                                 void* stream) {
-  k_gelu_fwd_bf16<<<(int)bmin64((n + 255) / 256, 2048), 256, 0, S(stream)>>>(
-      (const __bf16*)u, (__bf16*)g, n);
+  if (n % 8) return ob_fail("gelu_bf16: n must be a multiple of 8");
+  k_gelu_fwd_bf16<<<(int)bmin64((n / 8 + 255) / 256, 2048), 256, 0,
+                    S(stream)>>>((const __bf16*)u, (__bf16*)g, n / 8);
   OB_LAUNCH_CHECK();
   return 0;
 }
 extern "C" int ob_gelu_bwd_bf16(const void* u, const void* dg, void* du,
                                 int64_t n, void* stream) {
-  k_gelu_bwd_bf16<<<(int)bmin64((n + 255) / 256, 2048), 256, 0, S(stream)>>>(
-      (const __bf16*)u, (const __bf16*)dg, (__bf16*)du, n);
+  if (n % 8) return ob_fail("gelu_bf16: n must be a multiple of 8");
+  k_gelu_bwd_bf16<<<(int)bmin64((n / 8 + 255) / 256, 2048), 256, 0,
+                    S(stream)>>>((const __bf16*)u, (const __bf16*)dg,
+                                 (__bf16*)du, n / 8);
   OB_LAUNCH_CHECK();
   return 0;
 }
@@ -908,12 +931,19 @@ __global__ __launch_bounds__(256) void k_ce_bwd_bf16(
     const float l = lse[row];
     const bool valid = s_pos < Sq - 1;
     const int64_t lab = valid ? labels[row + 1] : -1;
-    for (int64_t c = (int64_t)blockIdx.x * 256 + threadIdx.x; c < ld;
-         c += (int64_t)gridDim.x * 256) {
-      float g = 0.f;
-      if (valid && c < V)
-        g = scale * (__expf(bf2f(lr[c]) - l) - (c == lab ? 1.f : 0.f));
-      lr[c] = (__bf16)g;
+    for (int64_t c8 = ((int64_t)blockIdx.x * 256 + threadIdx.x) * 8;
+         c8 < ld; c8 += (int64_t)gridDim.x * 256 * 8) {
+      const uint4 in = *reinterpret_cast<const uint4*>(lr + c8);
+      float out[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int64_t c = c8 + j;
+        out[j] = (valid && c < V)
+                     ? scale * (__expf(bf2f(bf_extract(in, j)) - l) -
+                                (c == lab ? 1.f : 0.f))
+                     : 0.f;
+      }
+      *reinterpret_cast<uint4*>(lr + c8) = bf_pack8(out);
     }
   }
 }
@@ -921,7 +951,8 @@ extern "C" int ob_ce_bwd_bf16(void* logits, const void* labels,
                               const void* lse, const void* dloss, int64_t B,
                               int64_t Sq, int64_t V, int64_t ld,
                               void* stream) {
-  dim3 grid((unsigned)bmin64((ld + 255) / 256, 256),
+  if (ld % 8) return ob_fail("ce_bwd_bf16: ld must be a multiple of 8");
+  dim3 grid((unsigned)bmin64((ld / 8 + 255) / 256, 256),
             (unsigned)bmin64(B * Sq, 16384));
   k_ce_bwd_bf16<<<grid, 256, 0, S(stream)>>>(
       (__bf16*)logits, (const int64_t*)labels, (const float*)lse,
