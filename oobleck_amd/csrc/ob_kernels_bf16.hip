@@ -325,10 +325,24 @@ extern "C" int ob_gemm_bf16(int transA, int transB, int64_t M, int64_t N,
         float alpha, float beta, int out_kind, int splitk, void* stream,
         int64_t Mr, int BN);
     if (M % 256 == 0 && K % 64 == 0 && K >= 128) {
-      const int bn2 = (N % 256 == 0) ? 256 : (N % 128 == 0 ? 128 : 0);
+      // pick BN by scheduling-round efficiency (blocks / ceil-to-256):
+      // e.g. fc (M=8192,N=3072): BN=256 -> 384 blocks = 1.5 rounds (75%),
+      // BN=128 -> 768 blocks = 3 full rounds (100%)
+      int bn2 = 0;
+      double best_eff = 0.0;
+      const int64_t zb = n1 * n2 * (splitk < 1 ? 1 : splitk);
+      for (int cand : {256, 128}) {
+        if (N % cand) continue;
+        const int64_t blocks = (M / 256) * (N / cand) * zb;
+        const double eff =
+            (double)blocks / (((blocks + 255) / 256) * 256);
+        if (eff > best_eff + 1e-9) {
+          best_eff = eff;
+          bn2 = cand;
+        }
+      }
       if (bn2) {
-        const int64_t tiles =
-            (M / 256) * (N / bn2) * n1 * n2 * (splitk < 1 ? 1 : splitk);
+        const int64_t tiles = (M / 256) * (N / bn2) * zb;
         if (tiles >= 192)
           return ob_gemm_bf16_nt256_dispatch(
               A, B, C, bias, residual, M, N, K, lda, ldb, ldc, strideA1,
@@ -1361,15 +1375,26 @@ __global__ __launch_bounds__(512, 2) void k_gemm_bf16_nt_256(
     }                                                                         \
   }
 
+  // counted vmcnt + raw barriers: at 1 block/CU there is no co-resident
+  // block to hide the drain, so never wait the in-flight next tile.
+  // per-wave glds per tile: 4 (A) + BN/128*... = 8 (BN=256) or 6 (BN=128).
   OB_N2_GLDS(0, kbeg)
-  __syncthreads();
   int cur = 0;
   for (int kt = kbeg; kt + BK < kend; kt += BK) {
     OB_N2_GLDS(cur ^ 1, kt + BK)
+    if (BN == 256)
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
     OB_N2_MFMA(cur)
-    __syncthreads();
+    __builtin_amdgcn_s_barrier();
     cur ^= 1;
   }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
   OB_N2_MFMA(cur)
 #undef OB_N2_GLDS
 #undef OB_N2_FRAG
