@@ -324,7 +324,7 @@ extern "C" int ob_gemm_bf16(int transA, int transB, int64_t M, int64_t N,
         int64_t sB2, int64_t sC1, int64_t sC2, int64_t n1, int64_t n2,
         float alpha, float beta, int out_kind, int splitk, void* stream,
         int64_t Mr, int BN);
-    if (M % 256 == 0 && K % 64 == 0 && K >= 128) {
+    if (M % 256 == 0 && K % 64 == 0 && K >= 1024) {
       // pick BN by scheduling-round efficiency (blocks / ceil-to-256):
       // e.g. fc (M=8192,N=3072): BN=256 -> 384 blocks = 1.5 rounds (75%),
       // BN=128 -> 768 blocks = 3 full rounds (100%)
@@ -1009,6 +1009,14 @@ extern "C" int ob_ce_bwd_bf16(void* logits, const void* labels,
 // on HBM latency that 256-cycle bf16 MFMA segments cannot hide).
 // ---------------------------------------------------------------------------
 
+__device__ __forceinline__ int bf_xcd_swz(int bid, int nwg) {
+  // T1: give each XCD a contiguous run of tiles so shared operand panels
+  // stay in its private L2 (bijective variant for nwg % 8 != 0).
+  const int q = nwg / 8, r = nwg % 8;
+  const int xcd = bid % 8, idx = bid / 8;
+  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+}
+
 __device__ __forceinline__ int bf_swz_key(int row) {
   return (row & 3) ^ ((row >> 2) & 3);
 }
@@ -1028,7 +1036,7 @@ __global__ __launch_bounds__(256, 3) void k_gemm_bf16_nt_glds(
   __shared__ __bf16 As[3][128 * 32];
   __shared__ __bf16 Bs[3][128 * 32];
 
-  const int tile = blockIdx.x;
+  const int tile = bf_xcd_swz(blockIdx.x, gridDim.x);
   const int bm = tile / nbn, bn = tile % nbn;
   const int m0 = bm * 128, n0 = bn * 128;
 
@@ -1296,7 +1304,7 @@ __global__ __launch_bounds__(512, 2) void k_gemm_bf16_nt_256(
   __shared__ __bf16 As[2][256 * 64];
   __shared__ __bf16 Bs[2][BN * 64];
 
-  const int tile = blockIdx.x;
+  const int tile = bf_xcd_swz(blockIdx.x, gridDim.x);
   const int bm = tile / nbn, bn = tile % nbn;
   const int m0 = bm * 256, n0 = bn * BN;
 
