@@ -126,9 +126,10 @@ def measure_roofline(args, device):
     achieved_tf = flops / avg_s / 1e12
     peak = BF16_MFMA_PEAK_TF if bf16 else F32_MFMA_PEAK_TF
     traffic = None
-    pmc_file = os.path.join(os.path.dirname(os.path.abspath(__file__)),
-                            "profiles", "pmc_gemm_fc.json")
-    if not bf16 and os.path.exists(pmc_file):
+    pmc_file = os.path.join(
+        os.path.dirname(os.path.abspath(__file__)), "profiles",
+        "pmc_gemm_fc_bf16.json" if bf16 else "pmc_gemm_fc.json")
+    if os.path.exists(pmc_file):
         with open(pmc_file) as f:
             traffic = json.load(f).get("hbm_bytes_per_launch")
     return {

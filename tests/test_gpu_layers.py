@@ -187,3 +187,43 @@ def test_gpt2_small_block_shape_runs():
     assert torch.isfinite(out).all()
     assert torch.isfinite(din).all()
     assert torch.isfinite(layer.flat_grad).all()
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", ["f32", "bf16"])
+def test_xl_dims_block_parity(dtype):
+    """One GPT-2-XL-dimension block (H=1600, 25 heads — examples/gpt3.yaml,
+    BASELINE config[3]) forward+backward parity at B=2, S=256."""
+    from oobleck_amd.config import ModelConfig
+    from oobleck_amd.layer import Layer
+    xl = dict(n_embd=1600, n_head=25, n_layer=2, n_positions=1024,
+              vocab_size=50257)
+    mc = ModelConfig(**xl)
+    oc = OracleConfig(**xl)
+    B, S = 2, 256
+    from oracle.gpt2_oracle import init_layer_params
+    flat = init_layer_params(oc, 1, 99)
+    layer = Layer(1, mc, B, S, 1, DEV, dtype=dtype)
+    layer.flat_param.copy_(flat.to(DEV))
+    layer.refresh_weights()
+    g = torch.Generator().manual_seed(12)
+    x = torch.randn(B, S, 1600, generator=g) * 0.5
+    dout = torch.randn(B, S, 1600, generator=g) * 0.1
+    adt = torch.float32 if dtype == "f32" else torch.bfloat16
+    xg = x.to(DEV).to(adt)
+    out = torch.empty_like(xg)
+    layer.forward_slot(0, xg, out)
+    din = torch.empty_like(xg)
+    layer.backward_slot(0, dout.to(DEV).to(adt), din)
+    torch.cuda.synchronize()
+    ref_out, ref_dx, (ref_grad,) = stage_forward_backward(
+        oc, [flat], [1], x, dout=dout)
+    if dtype == "f32":
+        torch.testing.assert_close(out.cpu(), ref_out, rtol=1e-4, atol=1e-4)
+        torch.testing.assert_close(layer.flat_grad.cpu(), ref_grad,
+                                   rtol=1e-3, atol=1e-3)
+    else:
+        def rel_l2(a, b):
+            return ((a - b).norm() / b.norm().clamp_min(1e-12)).item()
+        assert rel_l2(out.float().cpu(), ref_out) < 5e-2
+        assert rel_l2(layer.flat_grad.cpu(), ref_grad) < 8e-2
