@@ -150,16 +150,16 @@ def measure_cpu_baseline():
     ~10-30 s sample: one fwd+bwd microbatch at B=1, S=512."""
     from oracle.gpt2_oracle import OracleConfig, stage_forward_backward
     from oracle.gpt2_oracle import init_layer_params
-    cores = os.cpu_count() or 8
+    # cap threads: oversubscribing all 256 host cores on these small
+    # matmuls is slower AND blows the bench time budget
+    cores = min(32, os.cpu_count() or 8)
     torch.set_num_threads(cores)
     oc = OracleConfig()
     flats = [init_layer_params(oc, oc.layer_kind(i), 42 + i)
              for i in range(oc.n_layers_total)]
-    B, S = 1, 512
+    B, S = 1, 256
     g = torch.Generator().manual_seed(0)
     ids = torch.randint(0, oc.vocab_size, (B, S), generator=g)
-    stage_forward_backward(oc, flats, list(range(oc.n_layers_total)), ids,
-                           labels=ids.clone())  # warm-up
     t0 = time.perf_counter()
     stage_forward_backward(oc, flats, list(range(oc.n_layers_total)), ids,
                            labels=ids.clone())
@@ -168,7 +168,8 @@ def measure_cpu_baseline():
         "value": round(B * S / dt, 2), "unit": "tokens/s", "cores": cores,
         "kind": "port",
         "sample": f"oracle GPT-2-small fwd+bwd, B={B} S={S} ({B*S} tokens), "
-                  f"1 timed iter after 1 warmup, torch {torch.__version__} CPU",
+                  f"1 timed iter (cold), torch {torch.__version__} CPU, "
+                  f"{cores} threads",
     }
 
 
