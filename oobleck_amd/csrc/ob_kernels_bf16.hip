@@ -66,6 +66,15 @@ __device__ __forceinline__ __bf16 bf_extract(const uint4& v, int j) {
   return bits_bf((j & 1) ? (word >> 16) : word);
 }
 
+__device__ __forceinline__ uint4 bf_pack8(const float* v) {
+  uint4 o;
+  o.x = bf_bits((__bf16)v[0]) | (bf_bits((__bf16)v[1]) << 16);
+  o.y = bf_bits((__bf16)v[2]) | (bf_bits((__bf16)v[3]) << 16);
+  o.z = bf_bits((__bf16)v[4]) | (bf_bits((__bf16)v[5]) << 16);
+  o.w = bf_bits((__bf16)v[6]) | (bf_bits((__bf16)v[7]) << 16);
+  return o;
+}
+
 // direct staging: img[row][k] = src[row][k], k contiguous in memory.
 // 256 threads; NR = tile rows (64 or 128).
 template <bool EDGE, int NR>
@@ -508,6 +517,7 @@ extern "C" int ob_layernorm_fwd_bf16(const void* x, const void* w,
 }
 
 #define BLN_CHUNK 16
+// thread owns 8 CONTIGUOUS columns (uint4 loads); H <= 2048
 template <bool DX_ACCUM>
 __global__ __launch_bounds__(256) void k_ln_bwd_bf16(
     const __bf16* __restrict__ x, const float* __restrict__ w,
@@ -516,46 +526,56 @@ __global__ __launch_bounds__(256) void k_ln_bwd_bf16(
     float* __restrict__ dw, float* __restrict__ db, int64_t rows, int H) {
   __shared__ float lds4[4];
   const int64_t r0 = (int64_t)blockIdx.x * BLN_CHUNK;
+  const int c8 = threadIdx.x * 8;
+  const bool live = c8 < H;
+  float wv[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) wv[j] = live ? w[c8 + j] : 0.f;
   float accw[8] = {0}, accb[8] = {0};
   const int64_t rend = bmin64(rows, r0 + BLN_CHUNK);
   for (int64_t row = r0; row < rend; ++row) {
     const __bf16* xr = x + row * H;
     const __bf16* dyr = dy + row * H;
     const float mu = mean[row], rs = rstd[row];
+    uint4 xin = {0, 0, 0, 0}, din = {0, 0, 0, 0};
+    if (live) {
+      xin = *reinterpret_cast<const uint4*>(xr + c8);
+      din = *reinterpret_cast<const uint4*>(dyr + c8);
+    }
     float s1 = 0.f, s2 = 0.f;
+    float xh[8], dv[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      const int c = threadIdx.x + j * 256;
-      if (c < H) {
-        const float xhat = (bf2f(xr[c]) - mu) * rs;
-        const float dyv = bf2f(dyr[c]);
-        const float dyw = dyv * w[c];
-        s1 += dyw * xhat;
-        s2 += dyw;
-        accw[j] += dyv * xhat;
-        accb[j] += dyv;
-      }
+      xh[j] = (bf2f(bf_extract(xin, j)) - mu) * rs;
+      dv[j] = bf2f(bf_extract(din, j));
+      const float dyw = dv[j] * wv[j];
+      s1 += dyw * xh[j];
+      s2 += dyw;
+      accw[j] += dv[j] * xh[j];
+      accb[j] += dv[j];
     }
+    if (!live) s1 = s2 = 0.f;
     const float m1 = bblock_sum256(s1, lds4) / H;
     const float m2 = bblock_sum256(s2, lds4) / H;
-    __bf16* dxr = dx + row * H;
+    if (live) {
+      __bf16* dxr = dx + row * H;
+      float out[8];
+      uint4 dprev = {0, 0, 0, 0};
+      if (DX_ACCUM) dprev = *reinterpret_cast<const uint4*>(dxr + c8);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      const int c = threadIdx.x + j * 256;
-      if (c < H) {
-        const float xhat = (bf2f(xr[c]) - mu) * rs;
-        const float v = rs * (bf2f(dyr[c]) * w[c] - m2 - xhat * m1);
-        dxr[c] = (__bf16)(DX_ACCUM ? bf2f(dxr[c]) + v : v);
+      for (int j = 0; j < 8; ++j) {
+        const float v = rs * (dv[j] * wv[j] - m2 - xh[j] * m1);
+        out[j] = DX_ACCUM ? bf2f(bf_extract(dprev, j)) + v : v;
       }
+      *reinterpret_cast<uint4*>(dxr + c8) = bf_pack8(out);
     }
     __syncthreads();
   }
+  if (live) {
 #pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    const int c = threadIdx.x + j * 256;
-    if (c < H) {
-      atomicAdd(&dw[c], accw[j]);
-      atomicAdd(&db[c], accb[j]);
+    for (int j = 0; j < 8; ++j) {
+      atomicAdd(&dw[c8 + j], accw[j]);
+      atomicAdd(&db[c8 + j], accb[j]);
     }
   }
 }
@@ -566,6 +586,7 @@ extern "C" int ob_layernorm_bwd_bf16(const void* x, const void* w,
                                      void* db, int64_t rows, int64_t H,
                                      int dx_accum, void* stream) {
   if (H > 2048) return ob_fail("ln_bwd_bf16: H > 2048 unsupported");
+  if (H % 8) return ob_fail("ln_bwd_bf16: H must be a multiple of 8");
   const int grid = (int)((rows + BLN_CHUNK - 1) / BLN_CHUNK);
   if (dx_accum)
     k_ln_bwd_bf16<true><<<grid, 256, 0, S(stream)>>>(
@@ -612,15 +633,7 @@ __global__ __launch_bounds__(256) void k_softmax_fwd_bf16(
   }
 }
 
-__device__ __forceinline__ uint4 bf_pack8_(const float* v);
-__device__ __forceinline__ uint4 bf_pack8(const float* v) {
-  uint4 o;
-  o.x = bf_bits((__bf16)v[0]) | (bf_bits((__bf16)v[1]) << 16);
-  o.y = bf_bits((__bf16)v[2]) | (bf_bits((__bf16)v[3]) << 16);
-  o.z = bf_bits((__bf16)v[4]) | (bf_bits((__bf16)v[5]) << 16);
-  o.w = bf_bits((__bf16)v[6]) | (bf_bits((__bf16)v[7]) << 16);
-  return o;
-}
+
 
 __device__ __forceinline__ uint4 bf_pack8_(const float* v) {
   return bf_pack8(v);
@@ -813,8 +826,8 @@ __global__ __launch_bounds__(256) void k_colsum_bf16_v8(
     int64_t N) {
   const int64_t c8 = ((int64_t)blockIdx.x * 256 + threadIdx.x) * 8;
   if (c8 >= N) return;
-  const int64_t r0 = (int64_t)blockIdx.y * 64;
-  const int64_t r1 = bmin64(M, r0 + 64);
+  const int64_t r0 = (int64_t)blockIdx.y * 16;
+  const int64_t r1 = bmin64(M, r0 + 16);
   float acc[8] = {0};
   for (int64_t r = r0; r < r1; ++r) {
     const uint4 v = *reinterpret_cast<const uint4*>(X + r * N + c8);
@@ -828,7 +841,7 @@ extern "C" int ob_colsum_bf16(const void* X, void* db, int64_t M, int64_t N,
                               void* stream) {
   if (N % 8 == 0) {
     // short row chunks keep the grid chip-filling even at N = H (96 lanes)
-    dim3 grid((unsigned)((N / 8 + 255) / 256), (unsigned)((M + 63) / 64));
+    dim3 grid((unsigned)((N / 8 + 255) / 256), (unsigned)((M + 15) / 16));
     k_colsum_bf16_v8<<<grid, 256, 0, S(stream)>>>((const __bf16*)X,
                                                   (float*)db, M, N);
   } else {
