@@ -826,8 +826,8 @@ __global__ __launch_bounds__(256) void k_colsum_bf16_v8(
     int64_t N) {
   const int64_t c8 = ((int64_t)blockIdx.x * 256 + threadIdx.x) * 8;
   if (c8 >= N) return;
-  const int64_t r0 = (int64_t)blockIdx.y * 16;
-  const int64_t r1 = bmin64(M, r0 + 16);
+  const int64_t r0 = (int64_t)blockIdx.y * 64;
+  const int64_t r1 = bmin64(M, r0 + 64);
   float acc[8] = {0};
   for (int64_t r = r0; r < r1; ++r) {
     const uint4 v = *reinterpret_cast<const uint4*>(X + r * N + c8);
@@ -841,7 +841,7 @@ extern "C" int ob_colsum_bf16(const void* X, void* db, int64_t M, int64_t N,
                               void* stream) {
   if (N % 8 == 0) {
     // short row chunks keep the grid chip-filling even at N = H (96 lanes)
-    dim3 grid((unsigned)((N / 8 + 255) / 256), (unsigned)((M + 15) / 16));
+    dim3 grid((unsigned)((N / 8 + 255) / 256), (unsigned)((M + 63) / 64));
     k_colsum_bf16_v8<<<grid, 256, 0, S(stream)>>>((const __bf16*)X,
                                                   (float*)db, M, N);
   } else {
