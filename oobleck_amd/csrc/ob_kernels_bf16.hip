@@ -517,7 +517,6 @@ extern "C" int ob_layernorm_fwd_bf16(const void* x, const void* w,
 }
 
 #define BLN_CHUNK 16
-// thread owns 8 CONTIGUOUS columns (uint4 loads); H <= 2048
 template <bool DX_ACCUM>
 __global__ __launch_bounds__(256) void k_ln_bwd_bf16(
     const __bf16* __restrict__ x, const float* __restrict__ w,
@@ -526,56 +525,46 @@ __global__ __launch_bounds__(256) void k_ln_bwd_bf16(
     float* __restrict__ dw, float* __restrict__ db, int64_t rows, int H) {
   __shared__ float lds4[4];
   const int64_t r0 = (int64_t)blockIdx.x * BLN_CHUNK;
-  const int c8 = threadIdx.x * 8;
-  const bool live = c8 < H;
-  float wv[8];
-#pragma unroll
-  for (int j = 0; j < 8; ++j) wv[j] = live ? w[c8 + j] : 0.f;
   float accw[8] = {0}, accb[8] = {0};
   const int64_t rend = bmin64(rows, r0 + BLN_CHUNK);
   for (int64_t row = r0; row < rend; ++row) {
     const __bf16* xr = x + row * H;
     const __bf16* dyr = dy + row * H;
     const float mu = mean[row], rs = rstd[row];
-    uint4 xin = {0, 0, 0, 0}, din = {0, 0, 0, 0};
-    if (live) {
-      xin = *reinterpret_cast<const uint4*>(xr + c8);
-      din = *reinterpret_cast<const uint4*>(dyr + c8);
-    }
     float s1 = 0.f, s2 = 0.f;
-    float xh[8], dv[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      xh[j] = (bf2f(bf_extract(xin, j)) - mu) * rs;
-      dv[j] = bf2f(bf_extract(din, j));
-      const float dyw = dv[j] * wv[j];
-      s1 += dyw * xh[j];
-      s2 += dyw;
-      accw[j] += dv[j] * xh[j];
-      accb[j] += dv[j];
+      const int c = threadIdx.x + j * 256;
+      if (c < H) {
+        const float xhat = (bf2f(xr[c]) - mu) * rs;
+        const float dyv = bf2f(dyr[c]);
+        const float dyw = dyv * w[c];
+        s1 += dyw * xhat;
+        s2 += dyw;
+        accw[j] += dyv * xhat;
+        accb[j] += dyv;
+      }
     }
-    if (!live) s1 = s2 = 0.f;
     const float m1 = bblock_sum256(s1, lds4) / H;
     const float m2 = bblock_sum256(s2, lds4) / H;
-    if (live) {
-      __bf16* dxr = dx + row * H;
-      float out[8];
-      uint4 dprev = {0, 0, 0, 0};
-      if (DX_ACCUM) dprev = *reinterpret_cast<const uint4*>(dxr + c8);
+    __bf16* dxr = dx + row * H;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const float v = rs * (dv[j] * wv[j] - m2 - xh[j] * m1);
-        out[j] = DX_ACCUM ? bf2f(bf_extract(dprev, j)) + v : v;
+    for (int j = 0; j < 8; ++j) {
+      const int c = threadIdx.x + j * 256;
+      if (c < H) {
+        const float xhat = (bf2f(xr[c]) - mu) * rs;
+        const float v = rs * (bf2f(dyr[c]) * w[c] - m2 - xhat * m1);
+        dxr[c] = (__bf16)(DX_ACCUM ? bf2f(dxr[c]) + v : v);
       }
-      *reinterpret_cast<uint4*>(dxr + c8) = bf_pack8(out);
     }
     __syncthreads();
   }
-  if (live) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      atomicAdd(&dw[c8 + j], accw[j]);
-      atomicAdd(&db[c8 + j], accb[j]);
+  for (int j = 0; j < 8; ++j) {
+    const int c = threadIdx.x + j * 256;
+    if (c < H) {
+      atomicAdd(&dw[c], accw[j]);
+      atomicAdd(&db[c], accb[j]);
     }
   }
 }
