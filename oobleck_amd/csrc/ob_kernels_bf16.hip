@@ -219,7 +219,8 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
       acc01 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b1, acc01, 0, 0, 0);\
       acc11 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b1, acc11, 0, 0, 0);\
     }                                                                         \
-  }
+  }                                                                      \
+  __builtin_amdgcn_s_setprio(0);
 
   OB_BF_STAGE(0, kbeg)
   __syncthreads();
@@ -1099,6 +1100,7 @@ __global__ __launch_bounds__(256, 3) void k_gemm_bf16_nt_glds(
       IMG + (ROW)*32 + (((SLOT) ^ bf_swz_key(ROW)) * 8)))
 
 #define OB_NT_MFMA(BUF)                                                       \
+  __builtin_amdgcn_s_setprio(1);                                              \
   _Pragma("unroll") for (int ks = 0; ks < 2; ++ks) {                          \
     const int slot = ks * 2 + kh;                                             \
     const int ra0 = wr * 64 + il, ra1 = wr * 64 + 32 + il;                    \
@@ -1371,6 +1373,7 @@ __global__ __launch_bounds__(512, 2) void k_gemm_bf16_nt_256(
       IMG + (ROW)*64 + (((SLOT) ^ bf_swz_key8(ROW)) * 8)))
 
 #define OB_N2_MFMA(BUF)                                                       \
+  __builtin_amdgcn_s_setprio(1);                                              \
   _Pragma("unroll") for (int ks = 0; ks < 4; ++ks) {                          \
     const int slot = ks * 2 + kh;                                             \
     bf16x8 bfr[FN];                                                           \
@@ -1383,7 +1386,8 @@ __global__ __launch_bounds__(512, 2) void k_gemm_bf16_nt_256(
           __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bfr[ni], acc[mi][ni],   \
                                                   0, 0, 0);                   \
     }                                                                         \
-  }
+  }                                                                           \
+  __builtin_amdgcn_s_setprio(0);
 
   // counted vmcnt + raw barriers: at 1 block/CU there is no co-resident
   // block to hide the drain, so never wait the in-flight next tile.
