@@ -480,6 +480,12 @@ __global__ __launch_bounds__(256, 2) void k_gemm_f32_p2(
 #undef OB_EPI
 }
 
+__global__ void k_gemm_f32_tn_glds(const float* __restrict__ A,
+                                   const float* __restrict__ B,
+                                   float* __restrict__ C, int M, int N, int K,
+                                   int64_t lda, int64_t ldb, int64_t ldc,
+                                   float alpha, int nbn);
+
 extern "C" int ob_gemm_f32(int transA, int transB, int64_t M, int64_t N,
                            int64_t K, float alpha, const void* A, int64_t lda,
                            int64_t strideA1, int64_t strideA2, const void* B,
@@ -535,6 +541,19 @@ extern "C" int ob_gemm_f32(int transA, int transB, int64_t M, int64_t N,
       else OB_GEMM_LAUNCH4(TA_, TB_, AT_, false, 128);                      \
     }                                                                       \
   } while (0)
+  // fp32 TN fast path: both operands k-major -> glds-direct kernel
+  // (interior, atomic out, 16-byte-aligned bases and leading dims)
+  if (transA && !transB && atomic && !edge && BN == 128 &&
+      (reinterpret_cast<uintptr_t>(A) | reinterpret_cast<uintptr_t>(B)) %
+              16 == 0 &&
+      (lda | ldb) % 4 == 0 && n1 * n2 == 1 && K >= 2 * GEMM_BK && !bias &&
+      !residual) {
+    k_gemm_f32_tn_glds<<<grid, block, 0, S(stream)>>>(
+        (const float*)A, (const float*)B, (float*)C, (int)M, (int)N, (int)K,
+        lda, ldb, ldc, alpha, nbn);
+    OB_LAUNCH_CHECK();
+    return 0;
+  }
   const int sel = (transA ? 4 : 0) | (transB ? 2 : 0) | (atomic ? 1 : 0);
   switch (sel) {
     case 0: OB_GEMM_LAUNCH2(false, false, false); break;
@@ -1046,4 +1065,106 @@ extern "C" int ob_adamw_step(void* p, const void* g, void* m, void* v,
       beta1, beta2, eps);
   OB_LAUNCH_CHECK();
   return 0;
+}
+
+// ---------------------------------------------------------------------------
+// fp32 TN (weight-grad) kernel with async glds staging: in the TN case BOTH
+// operands are stored k-major ([K=BS, dim]), so both stage DIRECTLY into
+// unpadded k-major LDS images ([32][128] f32: ds_read_b32 fragment reads
+// are conflict-free without padding) via global_load_lds — no transpose
+// staging, no register round-trip.  2 LDS buffers, counted vmcnt + raw
+// barriers (loads for tile t+1 fly under tile t's 4096-cycle MFMA block).
+// Interior shapes, atomic-f32 output (the dW pattern), split-K supported.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256, 2) void k_gemm_f32_tn_glds(
+    const float* __restrict__ A, const float* __restrict__ B,
+    float* __restrict__ C, int M, int N, int K, int64_t lda, int64_t ldb,
+    int64_t ldc, float alpha, int nbn) {
+  __shared__ float As[2][GEMM_BK * 128];
+  __shared__ float Bs[2][GEMM_BK * 128];
+
+  const int tile = blockIdx.x;
+  const int bm = tile / nbn, bn = tile % nbn;
+  const int m0 = bm * 128, n0 = bn * 128;
+
+  const int splitk = gridDim.y;
+  const int kchunk = ((K + splitk * GEMM_BK - 1) / (splitk * GEMM_BK)) * GEMM_BK;
+  const int kbeg = blockIdx.y * kchunk;
+  const int kend = min(K, kbeg + kchunk);
+  if (kbeg >= kend) return;
+
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int wr = w >> 1, wc = w & 1;
+  const int il = lane & 31, kh = lane >> 5;
+
+  // glds: 1 KiB per wave instruction = 2 rows of 512 B; per wave per tile
+  // per operand: 4 instructions (8 of 32 k-rows).
+  const int g_row = lane >> 5;        // 0..1 within the instruction
+  const int g_m4 = (lane & 31) * 4;   // 16-B column chunk
+#define OB_TN_GLDS(BUF, KT)                                                   \
+  {                                                                           \
+    _Pragma("unroll") for (int i = 0; i < 4; ++i) {                           \
+      const int k = w * 8 + i * 2 + g_row;                                    \
+      const float* asrc = A + (int64_t)((KT) + k) * lda + m0 + g_m4;          \
+      const float* bsrc = B + (int64_t)((KT) + k) * ldb + n0 + g_m4;          \
+      auto al = (__attribute__((address_space(3))) void*)                     \
+          (&As[BUF][(w * 8 + i * 2) * 128]);                                  \
+      auto bl = (__attribute__((address_space(3))) void*)                     \
+          (&Bs[BUF][(w * 8 + i * 2) * 128]);                                  \
+      __builtin_amdgcn_global_load_lds(                                       \
+          (const __attribute__((address_space(1))) void*)asrc, al, 16, 0, 0); \
+      __builtin_amdgcn_global_load_lds(                                       \
+          (const __attribute__((address_space(1))) void*)bsrc, bl, 16, 0, 0); \
+    }                                                                         \
+  }
+
+  f32x16 acc00 = {}, acc01 = {}, acc10 = {}, acc11 = {};
+#define OB_TN_MFMA(BUF)                                                       \
+  __builtin_amdgcn_s_setprio(1);                                              \
+  _Pragma("unroll") for (int kk = 0; kk < GEMM_BK / 2; ++kk) {                \
+    const float* ar = As[BUF] + (kk * 2 + kh) * 128 + wr * 64;                \
+    const float* br = Bs[BUF] + (kk * 2 + kh) * 128 + wc * 64;                \
+    const float a0 = ar[il], a1 = ar[32 + il];                                \
+    const float b0 = br[il], b1 = br[32 + il];                                \
+    acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);     \
+    acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);     \
+    acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);     \
+    acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);     \
+  }                                                                           \
+  __builtin_amdgcn_s_setprio(0);
+
+  OB_TN_GLDS(0, kbeg)
+  int cur = 0;
+  for (int kt = kbeg; kt + GEMM_BK < kend; kt += GEMM_BK) {
+    OB_TN_GLDS(cur ^ 1, kt + GEMM_BK)
+    asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
+    OB_TN_MFMA(cur)
+    __builtin_amdgcn_s_barrier();
+    cur ^= 1;
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+  OB_TN_MFMA(cur)
+#undef OB_TN_GLDS
+#undef OB_TN_MFMA
+
+  const int mw = m0 + wr * 64, nw = n0 + wc * 64;
+#define OB_TN_EPI(ACC, TI, TJ)                                                \
+  {                                                                           \
+    const int nn = nw + (TJ)*32 + il;                                         \
+    _Pragma("unroll") for (int r = 0; r < 16; ++r) {                          \
+      const int mm = mw + (TI)*32 + (r & 3) + 8 * (r >> 2) + 4 * kh;          \
+      atomicAdd(&C[(int64_t)mm * ldc + nn], alpha * ACC[r]);                  \
+    }                                                                         \
+  }
+  OB_TN_EPI(acc00, 0, 0)
+  OB_TN_EPI(acc01, 0, 1)
+  OB_TN_EPI(acc10, 1, 0)
+  OB_TN_EPI(acc11, 1, 1)
+#undef OB_TN_EPI
 }
