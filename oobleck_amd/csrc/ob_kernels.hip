@@ -541,9 +541,14 @@ extern "C" int ob_gemm_f32(int transA, int transB, int64_t M, int64_t N,
       else OB_GEMM_LAUNCH4(TA_, TB_, AT_, false, 128);                      \
     }                                                                       \
   } while (0)
-  // fp32 TN fast path: both operands k-major -> glds-direct kernel
-  // (interior, atomic out, 16-byte-aligned bases and leading dims)
-  if (transA && !transB && atomic && !edge && BN == 128 &&
+  // fp32 TN glds-direct kernel: measured perf-neutral on the dW shapes
+  // (85 TF either way — fp32 dW is tail/occupancy-bound, not
+  // staging-bound), so it stays opt-in for experiments.
+  static const bool tn_glds = [] {
+    const char* e = getenv("OB_F32_TN_GLDS");
+    return e && e[0] == '1';
+  }();
+  if (tn_glds && transA && !transB && atomic && !edge && BN == 128 &&
       (reinterpret_cast<uintptr_t>(A) | reinterpret_cast<uintptr_t>(B)) %
               16 == 0 &&
       (lda | ldb) % 4 == 0 && n1 * n2 == 1 && K >= 2 * GEMM_BK && !bias &&
