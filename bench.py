@@ -256,7 +256,8 @@ def main():
         }
         if not args.skip_roofline:
             result["roofline"] = measure_roofline(args, device)
-        if not args.skip_cpu_baseline:
+        if not args.skip_cpu_baseline and world == 1:
+            # contract: the CPU-baseline leg runs on rank 0 at N=1 only
             result["cpu_baseline"] = measure_cpu_baseline()
         print(json.dumps(result))
     dist.destroy_process_group()
