@@ -61,6 +61,10 @@ def _configure(lib: ctypes.CDLL) -> ctypes.CDLL:
     lib.ob_f32_to_bf16.argtypes = [vp, vp, i64, vp]
     lib.ob_f32_to_bf16_t.argtypes = [vp, vp, i64, i64, vp]
     lib.ob_bf16_to_f32.argtypes = [vp, vp, i64, vp]
+    lib.ob_flash_fwd_bf16.argtypes = [vp, vp, vp, vp, i64, i64, i64, i64,
+                                      f32, vp]
+    lib.ob_transpose_bf16_b.argtypes = [vp, vp, i64, i64, i64, i64, i64,
+                                        i64, i64, vp]
     return lib
 
 

@@ -74,6 +74,12 @@ int ob_ce_bwd_bf16(void* logits, const void* labels, const void* lse,
                    int64_t ld, void* stream);
 int ob_transpose_bf16(const void* in, void* out, int64_t R, int64_t C,
                       void* stream);
+int ob_flash_fwd_bf16(const void* qkv, const void* VT, void* O, void* lse,
+                      int64_t B, int64_t Sq, int64_t H, int64_t nh,
+                      float scale, void* stream);
+int ob_transpose_bf16_b(const void* in, void* out, int64_t R, int64_t C,
+                        int64_t sIn1, int64_t sIn2, int64_t ldin, int64_t n1,
+                        int64_t n2, void* stream);
 }
 // fast NT glds dispatch (interior 128-tiled M/N; Mr = store-row guard)
 int ob_gemm_bf16_nt_dispatch(const void* A, const void* B, void* C,

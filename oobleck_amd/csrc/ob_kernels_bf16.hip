@@ -1473,3 +1473,217 @@ int ob_gemm_bf16_nt256_dispatch(const void* A, const void* B, void* C,
   OB_LAUNCH_CHECK();
   return 0;
 }
+
+// ---------------------------------------------------------------------------
+// Flash-style fused causal attention, bf16, head_dim = 64 (§7 perf queue
+// item 3; SURVEY.md "hard part (i)").  Forward: per (b, h, 128-row q-tile)
+// block of 4 waves; each wave owns 32 q rows.  QK^T is computed SWAPPED
+// (mfma(A=K, B=Q)) so each lane holds the scores of ONE q column (q =
+// lane&31, 16 kv rows per accumulator) — the online-softmax row reduce is
+// lane-local + one shfl_xor(32) partner combine.  P converts to the PV
+// A-fragment with 8 pack-to-bf16 ops + 2 permlane32_swap per k-step (the
+// kh halves exchange kv{4..7}/{8..11} exactly as the fragment k-runs
+// need).  V is consumed through a materialized V^T ([z][64][S], cheap
+// batched transpose) so the PV B-fragment reads are k-contiguous.  O
+// accumulates in fp32; per-tile rescale factors broadcast wave-locally
+// through LDS (no barriers anywhere in the loop).  Writes O (bf16) and
+// the per-row LSE (fp32) for the backward recompute.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ unsigned bf_pk2(float a, float b) {
+  return bf_bits((__bf16)a) | (bf_bits((__bf16)b) << 16);
+}
+
+__global__ __launch_bounds__(256, 2) void k_flash_fwd_bf16(
+    const __bf16* __restrict__ qkv, const __bf16* __restrict__ VT,
+    __bf16* __restrict__ Obase, float* __restrict__ lse, int Sq, int H,
+    int nh, float scale) {
+  __shared__ float bcast[128];
+  const int z = blockIdx.z;
+  const int b = z / nh, h = z % nh;
+  const int64_t qoff = (int64_t)b * Sq * 3 * H + h * 64;
+  const __bf16* Qp = qkv + qoff;
+  const __bf16* Kp = Qp + H;
+  const __bf16* VTp = VT + (int64_t)z * 64 * Sq;
+  __bf16* Op = Obase + (int64_t)b * Sq * H + h * 64;
+  float* lsep = lse + (int64_t)z * Sq;
+
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int il = lane & 31, kh = lane >> 5;
+  const int q0 = blockIdx.x * 128 + w * 32;
+  const int myq = q0 + il;
+
+  // Q fragments, resident for the whole block: B[k=d-run][j=q]
+  bf16x8 qf[4];
+#pragma unroll
+  for (int s = 0; s < 4; ++s)
+    qf[s] = *reinterpret_cast<const bf16x8*>(
+        Qp + (int64_t)myq * 3 * H + s * 16 + kh * 8);
+
+  f32x16 o0 = {}, o1 = {};
+  float m = -INFINITY, l = 0.f;
+
+  const int ntiles = (blockIdx.x * 128 + 128) / 32;
+  for (int kvt = 0; kvt < ntiles; ++kvt) {
+    const int kv0 = kvt * 32;
+    if (kv0 > q0 + 31) continue;  // wave-uniform: fully masked for this wave
+
+    // K fragments: A[i=kv][k=d-run]
+    f32x16 sacc = {};
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      const bf16x8 kf = *reinterpret_cast<const bf16x8*>(
+          Kp + (int64_t)(kv0 + il) * 3 * H + s * 16 + kh * 8);
+      sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[s], sacc, 0, 0, 0);
+    }
+
+    // online softmax over this lane's 16 kv entries (column q = myq)
+    float sv[16];
+    float mt = -INFINITY;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int kv = kv0 + (r & 3) + 8 * (r >> 2) + 4 * kh;
+      sv[r] = (kv <= myq) ? sacc[r] * scale : -INFINITY;
+      mt = fmaxf(mt, sv[r]);
+    }
+    mt = fmaxf(mt, __shfl_xor(mt, 32, 64));
+    const float mnew = fmaxf(m, mt);
+    const float af = __expf(m - mnew);  // exp(-inf - x) = 0 on the 1st tile
+    m = mnew;
+    float psum = 0.f;
+    float pv[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      pv[r] = (sv[r] == -INFINITY) ? 0.f : __expf(sv[r] - mnew);
+      psum += pv[r];
+    }
+    psum += __shfl_xor(psum, 32, 64);
+    l = l * af + psum;
+
+    // rescale O by alpha per q-row (wave-local LDS broadcast, no barrier:
+    // within-wave ds ordering is by lgkmcnt)
+    if (kh == 0) bcast[w * 32 + il] = af;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const float a = bcast[w * 32 + (r & 3) + 8 * (r >> 2) + 4 * kh];
+      o0[r] *= a;
+      o1[r] *= a;
+    }
+
+    // P -> A-fragments: pack + kh-half exchange (see header comment)
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      unsigned x0 = bf_pk2(pv[t * 8 + 0], pv[t * 8 + 1]);
+      unsigned x1 = bf_pk2(pv[t * 8 + 2], pv[t * 8 + 3]);
+      unsigned y0 = bf_pk2(pv[t * 8 + 4], pv[t * 8 + 5]);
+      unsigned y1 = bf_pk2(pv[t * 8 + 6], pv[t * 8 + 7]);
+      {
+        auto r2 = __builtin_amdgcn_permlane32_swap(x0, y0, false, false);
+        x0 = r2[0];
+        y0 = r2[1];
+      }
+      {
+        auto r2 = __builtin_amdgcn_permlane32_swap(x1, y1, false, false);
+        x1 = r2[0];
+        y1 = r2[1];
+      }
+      uint4 fr;
+      fr.x = x0;
+      fr.y = x1;
+      fr.z = y0;
+      fr.w = y1;
+      const bf16x8 pa = __builtin_bit_cast(bf16x8, fr);
+      // PV: B[k=kv-run][j=d] from V^T rows
+      const bf16x8 v0 = *reinterpret_cast<const bf16x8*>(
+          VTp + (int64_t)il * Sq + kv0 + t * 16 + kh * 8);
+      const bf16x8 v1 = *reinterpret_cast<const bf16x8*>(
+          VTp + (int64_t)(32 + il) * Sq + kv0 + t * 16 + kh * 8);
+      o0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, v0, o0, 0, 0, 0);
+      o1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, v1, o1, 0, 0, 0);
+    }
+  }
+
+  // epilogue: O /= l ; lse = m + log(l)
+  if (kh == 0) {
+    bcast[w * 32 + il] = 1.f / l;
+    lsep[myq] = m + __logf(l);
+  }
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = (r & 3) + 8 * (r >> 2) + 4 * kh;
+    const float inv = bcast[w * 32 + row];
+    Op[(int64_t)(q0 + row) * H + il] = (__bf16)(o0[r] * inv);
+    Op[(int64_t)(q0 + row) * H + 32 + il] = (__bf16)(o1[r] * inv);
+  }
+}
+
+extern "C" int ob_flash_fwd_bf16(const void* qkv, const void* VT, void* O,
+                                 void* lse, int64_t B, int64_t Sq, int64_t H,
+                                 int64_t nh, float scale, void* stream) {
+  if (H / nh != 64) return ob_fail("flash_fwd: head_dim must be 64");
+  if (Sq % 128) return ob_fail("flash_fwd: S must be a multiple of 128");
+  dim3 grid((unsigned)(Sq / 128), 1, (unsigned)(B * nh));
+  k_flash_fwd_bf16<<<grid, 256, 0, S(stream)>>>(
+      (const __bf16*)qkv, (const __bf16*)VT, (__bf16*)O, (float*)lse,
+      (int)Sq, (int)H, (int)nh, scale);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+// batched strided transpose: out[z][C][R] = in[z-slice][R][C] where the
+// input slice z starts at in + z1*(z/n2? no) — two-level (b,h) strides like
+// the GEMMs.  Used to materialize V^T (and the backward's Q^T/K^T/dO^T).
+__global__ __launch_bounds__(256) void k_transpose_bf16_b(
+    const __bf16* __restrict__ in, __bf16* __restrict__ out, int64_t R,
+    int64_t C, int64_t sIn1, int64_t sIn2, int64_t ldin, int n2) {
+  __shared__ __bf16 tile[64 * 65];
+  const int z = blockIdx.z;
+  const int i1 = z / n2, i2 = z % n2;
+  const __bf16* src = in + (int64_t)i1 * sIn1 + (int64_t)i2 * sIn2;
+  __bf16* dst = out + (int64_t)z * R * C;
+  const int64_t r0 = (int64_t)blockIdx.y * 64, c0 = (int64_t)blockIdx.x * 64;
+  {
+    const int rr = threadIdx.x >> 3;
+    const int c8 = (threadIdx.x & 7) * 8;
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int r = rr + it * 32;
+      uint4 v = {0, 0, 0, 0};
+      if (r0 + r < R && c0 + c8 + 7 < C)
+        v = *reinterpret_cast<const uint4*>(src + (r0 + r) * ldin + c0 + c8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) tile[r * 65 + c8 + j] = bf_extract(v, j);
+    }
+  }
+  __syncthreads();
+  {
+    const int cc = threadIdx.x >> 3;
+    const int r8 = (threadIdx.x & 7) * 8;
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int c = cc + it * 32;
+      if (c0 + c >= C) continue;
+      float tmp[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        tmp[j] = bf2f(tile[(r8 + j) * 65 + c]);
+      if (r0 + r8 + 7 < R)
+        *reinterpret_cast<uint4*>(dst + (c0 + c) * R + r0 + r8) =
+            bf_pack8(tmp);
+    }
+  }
+}
+
+extern "C" int ob_transpose_bf16_b(const void* in, void* out, int64_t R,
+                                   int64_t C, int64_t sIn1, int64_t sIn2,
+                                   int64_t ldin, int64_t n1, int64_t n2,
+                                   void* stream) {
+  if (R % 64 || C % 64) return ob_fail("transpose_b: R,C must be 64-multiples");
+  dim3 grid((unsigned)((C + 63) / 64), (unsigned)((R + 63) / 64),
+            (unsigned)(n1 * n2));
+  k_transpose_bf16_b<<<grid, 256, 0, S(stream)>>>(
+      (const __bf16*)in, (__bf16*)out, R, C, sIn1, sIn2, ldin, (int)n2);
+  OB_LAUNCH_CHECK();
+  return 0;
+}

@@ -206,3 +206,41 @@ def test_bf16_full_model_parity():
         (gpu_loss, ref_loss.item())
     for lid, layer in enumerate(layers):
         assert rel_l2(layer.flat_grad.cpu(), ref_grads[lid]) < 8e-2, lid
+
+
+@requires_gpu
+@pytest.mark.parametrize("B,nh,S", [(2, 2, 128), (1, 3, 256), (2, 2, 1024)])
+def test_flash_fwd_bf16(B, nh, S):
+    """Flash forward vs a torch fp32 reference of causal attention on the
+    same bf16-rounded inputs."""
+    import math
+    from oobleck_amd._ext import check, get_ext
+    H = nh * 64
+    g = torch.Generator().manual_seed(31)
+    qkv = (torch.randn(B, S, 3 * H, generator=g) * 0.5).to(DEV).bfloat16()
+    VT = torch.empty(B * nh, 64, S, device=DEV, dtype=torch.bfloat16)
+    # V slices: qkv[..., 2H + h*64 : ...] viewed [S, 64] at ld 3H
+    check(get_ext().ob_transpose_bf16_b(
+        ptr(qkv.flatten()[2 * H:]), ptr(VT), S, 64, S * 3 * H, 64, 3 * H,
+        B, nh, stream()), "vt")
+    O = torch.empty(B, S, H, device=DEV, dtype=torch.bfloat16)
+    lse = torch.empty(B * nh, S, device=DEV, dtype=torch.float32)
+    check(get_ext().ob_flash_fwd_bf16(
+        ptr(qkv), ptr(VT), ptr(O), ptr(lse), B, S, H, nh,
+        1.0 / math.sqrt(64.0), stream()), "flash")
+    torch.cuda.synchronize()
+
+    qf = qkv.float()
+    q = qf[..., :H].view(B, S, nh, 64).permute(0, 2, 1, 3)
+    k = qf[..., H:2 * H].view(B, S, nh, 64).permute(0, 2, 1, 3)
+    v = qf[..., 2 * H:].view(B, S, nh, 64).permute(0, 2, 1, 3)
+    w = torch.matmul(q, k.transpose(-1, -2)) / math.sqrt(64.0)
+    mask = torch.tril(torch.ones(S, S, dtype=torch.bool, device=DEV))
+    w = torch.where(mask, w, torch.tensor(float("-inf"), device=DEV))
+    lse_ref = torch.logsumexp(w, dim=-1)            # [B, nh, S]
+    P = torch.softmax(w, dim=-1)
+    O_ref = torch.matmul(P, v).permute(0, 2, 1, 3).reshape(B, S, H)
+    err = (O.float() - O_ref).norm() / O_ref.norm()
+    assert err < 2e-2, err.item()
+    torch.testing.assert_close(lse.view(B, nh, S), lse_ref,
+                               rtol=1e-3, atol=1e-3)
