@@ -80,6 +80,12 @@ int ob_flash_fwd_bf16(const void* qkv, const void* VT, void* O, void* lse,
 int ob_transpose_bf16_b(const void* in, void* out, int64_t R, int64_t C,
                         int64_t sIn1, int64_t sIn2, int64_t ldin, int64_t n1,
                         int64_t n2, void* stream);
+int ob_flash_dsum_bf16(const void* O, const void* dO, void* D, int64_t B,
+                       int64_t Sq, int64_t H, int64_t nh, void* stream);
+int ob_flash_bwd_bf16(const void* qkv, const void* QT, const void* KT,
+                      const void* dOT, const void* dO, const void* lse,
+                      const void* D, void* dqkv, int64_t B, int64_t Sq,
+                      int64_t H, int64_t nh, float scale, void* stream);
 }
 // fast NT glds dispatch (interior 128-tiled M/N; Mr = store-row guard)
 int ob_gemm_bf16_nt_dispatch(const void* A, const void* B, void* C,

@@ -1687,3 +1687,268 @@ extern "C" int ob_transpose_bf16_b(const void* in, void* out, int64_t R,
   OB_LAUNCH_CHECK();
   return 0;
 }
+
+// ---------------------------------------------------------------------------
+// Flash attention backward (recompute from Q, K, LSE; no stored P):
+//   D[z][q]   = sum_d dO[q,d] * O[q,d]                    (k_flash_dsum)
+//   dKdV      : blocks own 128 kv rows, loop q-tiles:
+//               S = QK^T (acc col = kv), P = exp(scale*S - lse[q]),
+//               dP = dO V^T-free (mfma(dO, V)), dS = P*(dP - D[q]),
+//               dV += P^T dO (via dO^T), dK += scale * dS^T Q (via Q^T)
+//   dQ        : blocks own 128 q rows, loop kv-tiles (fwd orientation):
+//               S^T, P^T, dP^T (mfma(V, dO)), dS^T, dQ += scale*dS K (via K^T)
+// Q^T/K^T/dO^T are materialized batched transposes ([z][64][S]).
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void k_flash_dsum(
+    const __bf16* __restrict__ O, const __bf16* __restrict__ dO,
+    float* __restrict__ D, int Sq, int H, int nh) {
+  // one wave per row: block 256 = 4 rows
+  const int z = blockIdx.z;
+  const int b = z / nh, h = z % nh;
+  const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (row >= Sq) return;
+  const int lane = threadIdx.x & 63;
+  const int64_t off = (int64_t)b * Sq * H + (int64_t)row * H + h * 64 + lane;
+  const float v = bf2f(O[off]) * bf2f(dO[off]);
+  float s = v;
+#pragma unroll
+  for (int o = 32; o > 0; o >>= 1) s += __shfl_down(s, o, 64);
+  if (lane == 0) D[(int64_t)z * Sq + row] = s;
+}
+
+extern "C" int ob_flash_dsum_bf16(const void* O, const void* dO, void* D,
+                                  int64_t B, int64_t Sq, int64_t H, int64_t nh,
+                                  void* stream) {
+  dim3 grid((unsigned)((Sq + 3) / 4), 1, (unsigned)(B * nh));
+  k_flash_dsum<<<grid, 256, 0, S(stream)>>>(
+      (const __bf16*)O, (const __bf16*)dO, (float*)D, (int)Sq, (int)H,
+      (int)nh);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+// pack-and-exchange: acc-held values (rows pattern) -> A-fragment k-runs
+__device__ __forceinline__ bf16x8 bf_dance(const float* pv8) {
+  unsigned x0 = bf_pk2(pv8[0], pv8[1]);
+  unsigned x1 = bf_pk2(pv8[2], pv8[3]);
+  unsigned y0 = bf_pk2(pv8[4], pv8[5]);
+  unsigned y1 = bf_pk2(pv8[6], pv8[7]);
+  {
+    auto r2 = __builtin_amdgcn_permlane32_swap(x0, y0, false, false);
+    x0 = r2[0];
+    y0 = r2[1];
+  }
+  {
+    auto r2 = __builtin_amdgcn_permlane32_swap(x1, y1, false, false);
+    x1 = r2[0];
+    y1 = r2[1];
+  }
+  uint4 fr;
+  fr.x = x0;
+  fr.y = x1;
+  fr.z = y0;
+  fr.w = y1;
+  return __builtin_bit_cast(bf16x8, fr);
+}
+
+__global__ __launch_bounds__(256, 2) void k_flash_bwd_dkdv(
+    const __bf16* __restrict__ qkv, const __bf16* __restrict__ QT,
+    const __bf16* __restrict__ dOT, const __bf16* __restrict__ dO,
+    const float* __restrict__ lse, const float* __restrict__ D,
+    __bf16* __restrict__ dqkv, int Sq, int H, int nh, float scale) {
+  __shared__ float lse_s[32], d_s[32];
+  const int z = blockIdx.z;
+  const int b = z / nh, h = z % nh;
+  const int64_t base = (int64_t)b * Sq * 3 * H + h * 64;
+  const __bf16* Qp = qkv + base;
+  const __bf16* Kp = Qp + H;
+  const __bf16* Vp = Qp + 2 * H;
+  const __bf16* dOp = dO + (int64_t)b * Sq * H + h * 64;
+  const __bf16* QTp = QT + (int64_t)z * 64 * Sq;
+  const __bf16* dOTp = dOT + (int64_t)z * 64 * Sq;
+  const float* lsep = lse + (int64_t)z * Sq;
+  const float* Dp = D + (int64_t)z * Sq;
+
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int il = lane & 31, kh = lane >> 5;
+  const int kv0 = blockIdx.x * 128 + w * 32;
+  const int mykv = kv0 + il;
+
+  // resident K and V fragments (j = kv columns for S and dP)
+  bf16x8 kf[4], vf[4];
+#pragma unroll
+  for (int s = 0; s < 4; ++s) {
+    kf[s] = *reinterpret_cast<const bf16x8*>(
+        Kp + (int64_t)mykv * 3 * H + s * 16 + kh * 8);
+    vf[s] = *reinterpret_cast<const bf16x8*>(
+        Vp + (int64_t)mykv * 3 * H + s * 16 + kh * 8);
+  }
+
+  f32x16 dv0 = {}, dv1 = {}, dk0 = {}, dk1 = {};
+
+  const int qt0 = (blockIdx.x * 128) / 32;
+  const int nqt = Sq / 32;
+  for (int qt = qt0; qt < nqt; ++qt) {
+    const int q0 = qt * 32;
+    if (q0 + 31 < kv0) continue;  // wave-uniform fully-masked
+    // lse/D broadcast for this q-tile (wave-local LDS is NOT enough here:
+    // the arrays are indexed by q-row across all lanes; load per wave into
+    // registers via per-lane reads instead)
+    // S-acc: mfma(A=Q(i=q), B=K(j=kv))  -> col = kv, rows = q
+    f32x16 sacc = {}, dpacc = {};
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      const bf16x8 af = *reinterpret_cast<const bf16x8*>(
+          Qp + (int64_t)(q0 + il) * 3 * H + s * 16 + kh * 8);
+      const bf16x8 df = *reinterpret_cast<const bf16x8*>(
+          dOp + (int64_t)(q0 + il) * H + s * 16 + kh * 8);
+      sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, kf[s], sacc, 0, 0, 0);
+      dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(df, vf[s], dpacc, 0, 0,
+                                                      0);
+    }
+    // P and dS per element: rows are q
+    float pv[16], dsv[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int q = q0 + (r & 3) + 8 * (r >> 2) + 4 * kh;
+      const bool ok = q >= mykv;
+      const float p = ok ? __expf(sacc[r] * scale - lsep[q]) : 0.f;
+      pv[r] = p;
+      dsv[r] = ok ? p * (dpacc[r] - Dp[q]) : 0.f;
+    }
+    // dV += P^T dO   (A = P^T frag: i = kv = lane col; k = q-run)
+    // dK += scale * dS^T Q
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      const bf16x8 pa = bf_dance(pv + t * 8);
+      const bf16x8 da = bf_dance(dsv + t * 8);
+      const bf16x8 do0 = *reinterpret_cast<const bf16x8*>(
+          dOTp + (int64_t)il * Sq + q0 + t * 16 + kh * 8);
+      const bf16x8 do1 = *reinterpret_cast<const bf16x8*>(
+          dOTp + (int64_t)(32 + il) * Sq + q0 + t * 16 + kh * 8);
+      const bf16x8 qt0f = *reinterpret_cast<const bf16x8*>(
+          QTp + (int64_t)il * Sq + q0 + t * 16 + kh * 8);
+      const bf16x8 qt1f = *reinterpret_cast<const bf16x8*>(
+          QTp + (int64_t)(32 + il) * Sq + q0 + t * 16 + kh * 8);
+      dv0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, do0, dv0, 0, 0, 0);
+      dv1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, do1, dv1, 0, 0, 0);
+      dk0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, qt0f, dk0, 0, 0, 0);
+      dk1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, qt1f, dk1, 0, 0, 0);
+    }
+  }
+
+  // write dK (scaled) and dV into dqkv slices: lane col = d, rows = kv
+  __bf16* dKp = dqkv + base + H;
+  __bf16* dVp = dqkv + base + 2 * H;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int kv = kv0 + (r & 3) + 8 * (r >> 2) + 4 * kh;
+    dVp[(int64_t)kv * 3 * H + il] = (__bf16)dv0[r];
+    dVp[(int64_t)kv * 3 * H + 32 + il] = (__bf16)dv1[r];
+    dKp[(int64_t)kv * 3 * H + il] = (__bf16)(scale * dk0[r]);
+    dKp[(int64_t)kv * 3 * H + 32 + il] = (__bf16)(scale * dk1[r]);
+  }
+  (void)lse_s;
+  (void)d_s;
+}
+
+__global__ __launch_bounds__(256, 2) void k_flash_bwd_dq(
+    const __bf16* __restrict__ qkv, const __bf16* __restrict__ KT,
+    const __bf16* __restrict__ dO, const float* __restrict__ lse,
+    const float* __restrict__ D, __bf16* __restrict__ dqkv, int Sq, int H,
+    int nh, float scale) {
+  const int z = blockIdx.z;
+  const int b = z / nh, h = z % nh;
+  const int64_t base = (int64_t)b * Sq * 3 * H + h * 64;
+  const __bf16* Qp = qkv + base;
+  const __bf16* Kp = Qp + H;
+  const __bf16* Vp = Qp + 2 * H;
+  const __bf16* dOp = dO + (int64_t)b * Sq * H + h * 64;
+  const __bf16* KTp = KT + (int64_t)z * 64 * Sq;
+  const float* lsep = lse + (int64_t)z * Sq;
+  const float* Dp = D + (int64_t)z * Sq;
+
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int il = lane & 31, kh = lane >> 5;
+  const int q0 = blockIdx.x * 128 + w * 32;
+  const int myq = q0 + il;
+  const float mylse = lsep[myq];
+  const float myD = Dp[myq];
+
+  bf16x8 qf[4], df[4];
+#pragma unroll
+  for (int s = 0; s < 4; ++s) {
+    qf[s] = *reinterpret_cast<const bf16x8*>(
+        Qp + (int64_t)myq * 3 * H + s * 16 + kh * 8);
+    df[s] = *reinterpret_cast<const bf16x8*>(
+        dOp + (int64_t)myq * H + s * 16 + kh * 8);
+  }
+
+  f32x16 dq0 = {}, dq1 = {};
+  const int ntiles = (blockIdx.x * 128 + 128) / 32;
+  for (int kvt = 0; kvt < ntiles; ++kvt) {
+    const int kv0 = kvt * 32;
+    if (kv0 > q0 + 31) continue;
+    f32x16 sacc = {}, dpacc = {};
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      const bf16x8 kf = *reinterpret_cast<const bf16x8*>(
+          Kp + (int64_t)(kv0 + il) * 3 * H + s * 16 + kh * 8);
+      const bf16x8 vf = *reinterpret_cast<const bf16x8*>(
+          Vp + (int64_t)(kv0 + il) * 3 * H + s * 16 + kh * 8);
+      sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[s], sacc, 0, 0, 0);
+      dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, df[s], dpacc, 0, 0,
+                                                      0);
+    }
+    float dsv[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int kv = kv0 + (r & 3) + 8 * (r >> 2) + 4 * kh;
+      const bool ok = kv <= myq;
+      const float p = ok ? __expf(sacc[r] * scale - mylse) : 0.f;
+      dsv[r] = ok ? p * (dpacc[r] - myD) : 0.f;
+    }
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      const bf16x8 da = bf_dance(dsv + t * 8);
+      const bf16x8 kt0 = *reinterpret_cast<const bf16x8*>(
+          KTp + (int64_t)il * Sq + kv0 + t * 16 + kh * 8);
+      const bf16x8 kt1 = *reinterpret_cast<const bf16x8*>(
+          KTp + (int64_t)(32 + il) * Sq + kv0 + t * 16 + kh * 8);
+      dq0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, kt0, dq0, 0, 0, 0);
+      dq1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, kt1, dq1, 0, 0, 0);
+    }
+  }
+  __bf16* dQp = dqkv + base;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int q = q0 + (r & 3) + 8 * (r >> 2) + 4 * kh;
+    dQp[(int64_t)q * 3 * H + il] = (__bf16)(scale * dq0[r]);
+    dQp[(int64_t)q * 3 * H + 32 + il] = (__bf16)(scale * dq1[r]);
+  }
+}
+
+extern "C" int ob_flash_bwd_bf16(const void* qkv, const void* QT,
+                                 const void* KT, const void* dOT,
+                                 const void* dO, const void* lse,
+                                 const void* D, void* dqkv, int64_t B,
+                                 int64_t Sq, int64_t H, int64_t nh,
+                                 float scale, void* stream) {
+  if (H / nh != 64) return ob_fail("flash_bwd: head_dim must be 64");
+  if (Sq % 128) return ob_fail("flash_bwd: S must be a multiple of 128");
+  dim3 grid((unsigned)(Sq / 128), 1, (unsigned)(B * nh));
+  k_flash_bwd_dkdv<<<grid, 256, 0, S(stream)>>>(
+      (const __bf16*)qkv, (const __bf16*)QT, (const __bf16*)dOT,
+      (const __bf16*)dO, (const float*)lse, (const float*)D, (__bf16*)dqkv,
+      (int)Sq, (int)H, (int)nh, scale);
+  OB_LAUNCH_CHECK();
+  k_flash_bwd_dq<<<grid, 256, 0, S(stream)>>>(
+      (const __bf16*)qkv, (const __bf16*)KT, (const __bf16*)dO,
+      (const float*)lse, (const float*)D, (__bf16*)dqkv, (int)Sq, (int)H,
+      (int)nh, scale);
+  OB_LAUNCH_CHECK();
+  return 0;
+}

@@ -73,7 +73,23 @@ struct ob_layer {
   int64_t sh_qkv = 0, sh_qkv_t = 0, sh_ap = 0, sh_ap_t = 0, sh_fc = 0,
           sh_fc_t = 0, sh_mp = 0, sh_mp_t = 0, sh_lm = 0, sh_lm_t = 0;
   int64_t shadow_count = 0;
+  // fused flash-attention path (bf16, head_dim 64, S % 128 == 0): the
+  // o_p slot holds only the per-row LSE ([Bm*nh, S] floats) instead of
+  // the materialized [Bm*nh, S, S] P.
+  bool flash = false;
 };
+
+// flash applies when head_dim == 64 and S % 128 == 0 (GPT-2 small and
+// XL both qualify); OB_BF16_FLASH=0 falls back to the materialized-P
+// path.  Decided at create time: it sizes the stash.
+static bool flash_eligible(const ob_layer_desc* d) {
+  static const bool off = [] {
+    const char* e = getenv("OB_BF16_FLASH");
+    return e && e[0] == '0';
+  }();
+  return !off && d->dtype == 1 && d->n_embd / d->n_head == 64 &&
+         d->seq_len % 128 == 0;
+}
 
 // parameter offsets (canonical layout, oracle/gpt2_oracle.py::layer_param_spec)
 struct BlockParams {
@@ -131,7 +147,8 @@ extern "C" int ob_layer_create(const ob_layer_desc* d, ob_layer_t* out) {
       l->o_rstd1 = o; o += BS;
       l->o_ln1 = o; o += BSH;
       l->o_qkv = o; o += BS * 3 * H;
-      l->o_p = o; o += Bm * nh * Sq * Sq;
+      l->flash = flash_eligible(d);
+      l->o_p = o; o += l->flash ? Bm * nh * Sq : Bm * nh * Sq * Sq;
       l->o_attnm = o; o += BSH;
       l->o_hmid = o; o += BSH;
       l->o_mean2 = o; o += BS;
@@ -196,7 +213,9 @@ extern "C" int ob_layer_create(const ob_layer_desc* d, ob_layer_t* out) {
   }
   // grow the shared backward workspace
   if (d->kind == OB_KIND_BLOCK) {
-    if (ws_ensure(&g_ws.dp, &g_ws.sz_dp, Bm * nh * Sq * Sq)) return 1;
+    if (!l->flash &&
+        ws_ensure(&g_ws.dp, &g_ws.sz_dp, Bm * nh * Sq * Sq))
+      return 1;
     if (ws_ensure(&g_ws.dqkv, &g_ws.sz_dqkv, BS * 3 * H)) return 1;
     if (ws_ensure(&g_ws.b4h, &g_ws.sz_b4h, BS * 4 * H)) return 1;
     if (d->dtype == 1) {
@@ -449,6 +468,8 @@ static int dw_bf16(const __bf16* act, int64_t actw, const __bf16* dY,
                  pick_splitk(actw, dyw, BS), stream);
 }
 
+static bool use_flash(const ob_layer* l) { return l->flash; }
+
 static int block_forward_bf16(ob_layer* l, int slot, const __bf16* in,
                               __bf16* out, void* stream) {
   const int64_t Sq = l->d.seq_len, H = l->d.n_embd, nh = l->d.n_head;
@@ -471,19 +492,33 @@ static int block_forward_bf16(ob_layer* l, int slot, const __bf16* in,
   if (gemm_bf(0, 1, BS, 3 * H, H, 1.f, ln1, H, 0, 0, sh + l->sh_qkv_t, H, 0,
               0, qkv, 3 * H, 0, 0, 1, 1, p + bp.b_qkv, nullptr, 0, 1, stream))
     return 1;
-  __bf16* P = (__bf16*)(st + l->o_p);
-  if (gemm_bf(0, 1, Sq, Sq, hd, 1.f, qkv, 3 * H, Sq * 3 * H, hd, qkv + H,
-              3 * H, Sq * 3 * H, hd, P, Sq, nh * Sq * Sq, Sq * Sq, B, nh,
-              nullptr, nullptr, 0, 1, stream))
-    return 1;
-  if (ob_softmax_causal_fwd_bf16(P, B * nh, Sq, 1.f / sqrtf((float)hd),
-                                 stream))
-    return 1;
   __bf16* am = (__bf16*)(st + l->o_attnm);
-  if (gemm_bf(0, 0, Sq, hd, Sq, 1.f, P, Sq, nh * Sq * Sq, Sq * Sq,
-              qkv + 2 * H, 3 * H, Sq * 3 * H, hd, am, H, Sq * H, hd, B, nh,
-              nullptr, nullptr, 0, 1, stream))
-    return 1;
+  if (use_flash(l)) {
+    // V^T materialization (workspace, transient within this call), then
+    // the fused kernel writes O and the per-row LSE (stored where the
+    // materialized-P path keeps P).
+    __bf16* VTw = (__bf16*)g_ws.t2;
+    if (ob_transpose_bf16_b(qkv + 2 * H, VTw, Sq, 64, Sq * 3 * H, 64, 3 * H,
+                            B, nh, stream))
+      return 1;
+    float* lseP = st + l->o_p;
+    if (ob_flash_fwd_bf16(qkv, VTw, am, lseP, B, Sq, H, nh,
+                          1.f / sqrtf((float)hd), stream))
+      return 1;
+  } else {
+    __bf16* P = (__bf16*)(st + l->o_p);
+    if (gemm_bf(0, 1, Sq, Sq, hd, 1.f, qkv, 3 * H, Sq * 3 * H, hd, qkv + H,
+                3 * H, Sq * 3 * H, hd, P, Sq, nh * Sq * Sq, Sq * Sq, B, nh,
+                nullptr, nullptr, 0, 1, stream))
+      return 1;
+    if (ob_softmax_causal_fwd_bf16(P, B * nh, Sq, 1.f / sqrtf((float)hd),
+                                   stream))
+      return 1;
+    if (gemm_bf(0, 0, Sq, hd, Sq, 1.f, P, Sq, nh * Sq * Sq, Sq * Sq,
+                qkv + 2 * H, 3 * H, Sq * 3 * H, hd, am, H, Sq * H, hd, B, nh,
+                nullptr, nullptr, 0, 1, stream))
+      return 1;
+  }
   __bf16* hmid = (__bf16*)(st + l->o_hmid);
   if (gemm_bf(0, 1, BS, H, H, 1.f, am, H, 0, 0, sh + l->sh_ap_t, H, 0, 0,
               hmid, H, 0, 0, 1, 1, p + bp.b_attnproj, x, 0, 1, stream))
@@ -595,23 +630,45 @@ static int block_backward_bf16(ob_layer* l, int slot, const __bf16* dout,
     return 1;
   if (ob_colsum_bf16(din, g + bp.b_attnproj, BS, H, stream)) return 1;
   // ---- attention core ----
-  if (gemm_bf(0, 1, Sq, Sq, hd, 1.f, DATT, H, Sq * H, hd, qkv + 2 * H, 3 * H,
-              Sq * 3 * H, hd, DP, Sq, nh * Sq * Sq, Sq * Sq, B, nh, nullptr,
-              nullptr, 0, 1, stream))
-    return 1;
-  if (ob_softmax_causal_bwd_bf16(P, DP, B * nh, Sq, stream)) return 1;
-  if (gemm_bf(0, 0, Sq, hd, Sq, scale, DP, Sq, nh * Sq * Sq, Sq * Sq,
-              qkv + H, 3 * H, Sq * 3 * H, hd, DQKV, 3 * H, Sq * 3 * H, hd, B,
-              nh, nullptr, nullptr, 0, 1, stream))
-    return 1;
-  if (gemm_bf(1, 0, Sq, hd, Sq, scale, DP, Sq, nh * Sq * Sq, Sq * Sq, qkv,
-              3 * H, Sq * 3 * H, hd, DQKV + H, 3 * H, Sq * 3 * H, hd, B, nh,
-              nullptr, nullptr, 0, 1, stream))
-    return 1;
-  if (gemm_bf(1, 0, Sq, hd, Sq, 1.f, P, Sq, nh * Sq * Sq, Sq * Sq, DATT, H,
-              Sq * H, hd, DQKV + 2 * H, 3 * H, Sq * 3 * H, hd, B, nh,
-              nullptr, nullptr, 0, 1, stream))
-    return 1;
+  if (use_flash(l)) {
+    const int64_t BSH = BS * H;
+    __bf16* QTw = (__bf16*)g_ws.t1;
+    __bf16* KTw = QTw + BSH;
+    __bf16* dOTw = KTw + BSH;
+    float* Dbuf = g_ws.t1 + (3 * BSH + 1) / 2;
+    const float* lseP = st + l->o_p;
+    if (ob_transpose_bf16_b(qkv, QTw, Sq, 64, Sq * 3 * H, 64, 3 * H, B, nh,
+                            stream))
+      return 1;
+    if (ob_transpose_bf16_b(qkv + H, KTw, Sq, 64, Sq * 3 * H, 64, 3 * H, B,
+                            nh, stream))
+      return 1;
+    if (ob_transpose_bf16_b(DATT, dOTw, Sq, 64, Sq * H, 64, H, B, nh,
+                            stream))
+      return 1;
+    if (ob_flash_dsum_bf16(am, DATT, Dbuf, B, Sq, H, nh, stream)) return 1;
+    if (ob_flash_bwd_bf16(qkv, QTw, KTw, dOTw, DATT, lseP, Dbuf, DQKV, B,
+                          Sq, H, nh, scale, stream))
+      return 1;
+  } else {
+    if (gemm_bf(0, 1, Sq, Sq, hd, 1.f, DATT, H, Sq * H, hd, qkv + 2 * H,
+                3 * H, Sq * 3 * H, hd, DP, Sq, nh * Sq * Sq, Sq * Sq, B, nh,
+                nullptr, nullptr, 0, 1, stream))
+      return 1;
+    if (ob_softmax_causal_bwd_bf16(P, DP, B * nh, Sq, stream)) return 1;
+    if (gemm_bf(0, 0, Sq, hd, Sq, scale, DP, Sq, nh * Sq * Sq, Sq * Sq,
+                qkv + H, 3 * H, Sq * 3 * H, hd, DQKV, 3 * H, Sq * 3 * H, hd,
+                B, nh, nullptr, nullptr, 0, 1, stream))
+      return 1;
+    if (gemm_bf(1, 0, Sq, hd, Sq, scale, DP, Sq, nh * Sq * Sq, Sq * Sq, qkv,
+                3 * H, Sq * 3 * H, hd, DQKV + H, 3 * H, Sq * 3 * H, hd, B,
+                nh, nullptr, nullptr, 0, 1, stream))
+      return 1;
+    if (gemm_bf(1, 0, Sq, hd, Sq, 1.f, P, Sq, nh * Sq * Sq, Sq * Sq, DATT,
+                H, Sq * H, hd, DQKV + 2 * H, 3 * H, Sq * 3 * H, hd, B, nh,
+                nullptr, nullptr, 0, 1, stream))
+      return 1;
+  }
   // ---- QKV projection ----
   if (ob_colsum_bf16(DQKV, g + bp.b_qkv, BS, 3 * H, stream)) return 1;
   if (dw_bf16(ln1, H, DQKV, 3 * H, BS, g + bp.w_qkv, 3 * H, stream))
