@@ -334,6 +334,27 @@ extern "C" int ob_gemm_bf16(int transA, int transB, int64_t M, int64_t N,
         int64_t sB2, int64_t sC1, int64_t sC2, int64_t n1, int64_t n2,
         float alpha, float beta, int out_kind, int splitk, void* stream,
         int64_t Mr, int BN);
+    // 8-phase 256^2 kernel: wins when its grid fills whole scheduling
+    // rounds (measured: lmhead 692 vs 618 TF, 4096^3 1058 vs 995; loses
+    // on partial rounds — fc 1.5 rounds: 529 vs 554).  tiles >= 192 and
+    // round-efficiency >= 0.9 required.
+    if (M % 256 == 0 && N % 256 == 0 && K % 64 == 0) {
+      extern int ob_gemm_bf16_nt_8ph(
+          const void* A, const void* B, void* C, const void* bias,
+          const void* residual, int64_t M, int64_t N, int64_t K, int64_t lda,
+          int64_t ldb, int64_t ldc, int64_t sA1, int64_t sA2, int64_t sB1,
+          int64_t sB2, int64_t sC1, int64_t sC2, int64_t n1, int64_t n2,
+          float alpha, float beta, int out_kind, int splitk, void* stream,
+          int64_t Mr);
+      const int64_t zb = n1 * n2 * (splitk < 1 ? 1 : splitk);
+      const int64_t tiles = (M / 256) * (N / 256) * zb;
+      const double eff = (double)tiles / (((tiles + 255) / 256) * 256);
+      if (tiles >= 192 && eff >= 0.9)
+        return ob_gemm_bf16_nt_8ph(A, B, C, bias, residual, M, N, K, lda,
+                                   ldb, ldc, strideA1, strideA2, strideB1,
+                                   strideB2, strideC1, strideC2, n1, n2,
+                                   alpha, beta, out_kind, splitk, stream, M);
+    }
     if (M % 256 == 0 && K % 64 == 0 && K >= 1024) {
       // pick BN by scheduling-round efficiency (blocks / ceil-to-256):
       // e.g. fc (M=8192,N=3072): BN=256 -> 384 blocks = 1.5 rounds (75%),
@@ -1949,6 +1970,280 @@ extern "C" int ob_flash_bwd_bf16(const void* qkv, const void* QT,
       (const __bf16*)qkv, (const __bf16*)KT, (const __bf16*)dO,
       (const float*)lse, (const float*)D, (__bf16*)dqkv, (int)Sq, (int)H,
       (int)nh, scale);
+  OB_LAUNCH_CHECK();
+  return 0;
+}
+
+// ---------------------------------------------------------------------------
+// 256x256 8-phase NT GEMM (the guide's fast template: glds staging,
+// st_16x32 LDS swizzle, per-phase ds_read || glds || MFMA interleave,
+// counted vmcnt once per K-tile, raw barriers, 1 block/CU).
+//   tile 256x256, BK=64, 512 threads = 8 waves as 2(M) x 4(N);
+//   per-wave output 128x64; MFMA 16x16x32 bf16 (fp32 acc).
+//   LDS = 2 buffers x 4 images x [128 rows][64 k] bf16 = 128 KiB.
+//   images: 0 = A rows 0-127, 1 = A rows 128-255, 2 = B rows 0-127,
+//   3 = B rows 128-255 (B = [N,K] k-major, NT).
+//   st_16x32 swizzle: image byte ^= ((byte>>9)&1)<<5, applied via the
+//   SOURCE-side address on the lane-linear glds write and the matching
+//   XOR on ds_read (glds cannot scatter).
+//   K-tile = 4 phases, one output quadrant (4 M-frags x 2 N-frags) each;
+//   staging runs 3 images ahead: phase 1 stages the next tile's last
+//   image, phase 4 stages tile+2's first three (into the buffer whose
+//   reads completed at phase 3's closing barrier), so the per-K-tile
+//   wait is vmcnt(6) = "3 images still flying".
+// ---------------------------------------------------------------------------
+
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+// SWZ 0: within-row 8-slot XOR, key (row>>1)&7 (the nt256 kernel's
+// measured-0-conflict pattern); SWZ 1: 16-slot XOR with the bit-6 flip.
+template <int OUT, int SWZ = 0>
+__global__ __launch_bounds__(512, 1) void k_gemm_bf16_nt_8ph(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    void* __restrict__ Cv, const float* __restrict__ bias,
+    const __bf16* __restrict__ R, int M, int N, int K, int64_t lda,
+    int64_t ldb, int64_t ldc, int64_t sA1, int64_t sA2, int64_t sB1,
+    int64_t sB2, int64_t sC1, int64_t sC2, int n2, float alpha, float beta,
+    int nbn, int Mr) {
+  __shared__ __bf16 lds[2][4][128 * 64];
+
+  const int tile = bf_xcd_swz(blockIdx.x, gridDim.x);
+  const int bm = tile / nbn, bn = tile % nbn;
+  const int m0 = bm * 256, n0 = bn * 256;
+
+  const int z = blockIdx.z;
+  const int i1 = z / n2, i2 = z % n2;
+  A += (int64_t)i1 * sA1 + (int64_t)i2 * sA2;
+  B += (int64_t)i1 * sB1 + (int64_t)i2 * sB2;
+  float* Cf = reinterpret_cast<float*>(Cv);
+  __bf16* Cb = reinterpret_cast<__bf16*>(Cv);
+  const int64_t coff = (int64_t)i1 * sC1 + (int64_t)i2 * sC2;
+  Cf += coff;
+  Cb += coff;
+  if (R) R += coff;
+
+  const int splitk = gridDim.y;
+  constexpr int BK = 64;
+  const int kchunk = ((K + splitk * BK - 1) / (splitk * BK)) * BK;
+  const int kbeg = blockIdx.y * kchunk;
+  const int kend = min(K, kbeg + kchunk);
+  if (kbeg >= kend) return;
+  const int NT = (kend - kbeg) / BK;  // K % 64 == 0 gated at dispatch
+
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;     // 0..7
+  const int wr = w >> 2, wc = w & 3;  // 2(M) x 4(N)
+  const int l16 = lane & 15, lg = lane >> 4;
+
+  // ---- LDS swizzle (T2, conflict-free variant): element index
+  //   idx = row*64 XOR (ke XOR ((row&15)<<3))
+  // The (row&8)<<3 term crosses the 64-element row boundary (it flips
+  // idx bit 6 = the bank-row half), so the 16 distinct rows of a
+  // ds_read_b128 fragment group land on 16 distinct 16-B slots of the
+  // 256-B bank row: conflict-free (the plain in-row XOR caps at 4-way).
+  // glds writes are lane-linear, so the permutation moves to the SOURCE
+  // address: lane l of block blk (8 image rows, 1 KiB) fetches
+  //   row = blk*8 + (l>>4)*2 + (((l>>3)&1) ^ (blk&1))
+  //   ce  = (l&7)*8 ^ ((row&7)<<3)
+  // (the inverse of idx at a = blk*512 + l*8).
+  // stage image IMG (0/1: A half, 2/3: B half) of K-tile at element KT
+#define P8_GLDS1(BUF, IMG, KT)                                                \
+  {                                                                           \
+    const __bf16* const src0 = (IMG) < 2 ? A : B;                             \
+    const int64_t ld = (IMG) < 2 ? lda : ldb;                                 \
+    const int r0 = ((IMG)&1) * 128 + ((IMG) < 2 ? m0 : n0);                   \
+    _Pragma("unroll") for (int v = 0; v < 2; ++v) {                           \
+      const int blk = w * 2 + v;                                              \
+      const int row =                                                         \
+          SWZ ? blk * 8 + ((lane >> 4) << 1) + (((lane >> 3) & 1) ^ (blk & 1))\
+              : blk * 8 + (lane >> 3);                                        \
+      const int ce = SWZ ? ((lane & 7) * 8) ^ ((row & 7) << 3)                \
+                         : ((lane & 7) * 8) ^ (((row >> 1) & 7) << 3);        \
+      const __bf16* src = src0 + (int64_t)(r0 + row) * ld + (KT) + ce;        \
+      auto lbase = (__attribute__((address_space(3))) void*)                  \
+          (&lds[BUF][IMG][blk * 8 * 64]);                                     \
+      __builtin_amdgcn_global_load_lds(                                       \
+          (const __attribute__((address_space(1))) void*)src, lbase, 16, 0,   \
+          0);                                                                 \
+    }                                                                         \
+  }
+
+  // ---- swizzled fragment read: row r, k-run element ke (lane's 8-run)
+#define P8_FRAG(BUF, IMG, ROW, KE)                                           \
+  (*reinterpret_cast<const bf16x8*>(                                         \
+      &lds[BUF][IMG][SWZ ? (((ROW)*64) ^ ((KE) ^ ((((ROW)&15)) << 3)))       \
+                         : ((ROW)*64 + ((KE) ^ (((((ROW) >> 1)) & 7)         \
+                                                << 3)))]))
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) acc[mi][ni] = (f32x4){};
+
+  bf16x8 Ar[4][2];     // one A half (mi, kk): a=0 read at phase 1 (used
+                       // p1, p2), overwritten by a=1 at phase 3
+  bf16x8 Br[2][2][2];  // BOTH N-quarters (b, ni, kk): b=0 at phase 1,
+                       // b=1 at phase 2, both live through phase 4 (the
+                       // B image is then free for phase-3 staging)
+  const int imA = wr;            // this wave's A image
+  const int imB = 2 + (wc >> 1); // this wave's B image
+  const int arow0 = l16;                    // + a*64 + mi*16
+  const int brow0 = (wc & 1) * 64 + l16;    // + b*32 + ni*16
+
+#define P8_RD_A(BUF, a)                                                      \
+  _Pragma("unroll") for (int mi = 0; mi < 4; ++mi)                           \
+      _Pragma("unroll") for (int kk = 0; kk < 2; ++kk) Ar[mi][kk] =          \
+          P8_FRAG(BUF, imA, arow0 + (a)*64 + mi * 16, kk * 32 + lg * 8);
+#define P8_RD_B(BUF, b)                                                      \
+  _Pragma("unroll") for (int ni = 0; ni < 2; ++ni)                           \
+      _Pragma("unroll") for (int kk = 0; kk < 2; ++kk) Br[b][ni][kk] =       \
+          P8_FRAG(BUF, imB, brow0 + (b)*32 + ni * 16, kk * 32 + lg * 8);
+
+  // kk outer: 8 independent MFMAs between the two updates of each acc
+#define P8_MFMA(a, b)                                                        \
+  __builtin_amdgcn_s_setprio(1);                                             \
+  _Pragma("unroll") for (int kk = 0; kk < 2; ++kk)                           \
+      _Pragma("unroll") for (int mi = 0; mi < 4; ++mi)                       \
+          _Pragma("unroll") for (int ni = 0; ni < 2; ++ni)                   \
+              acc[(a)*4 + mi][(b)*2 + ni] =                                  \
+      __builtin_amdgcn_mfma_f32_16x16x32_bf16(                               \
+          Ar[mi][kk], Br[b][ni][kk], acc[(a)*4 + mi][(b)*2 + ni], 0, 0,     \
+          0);                                                                \
+  __builtin_amdgcn_s_setprio(0);
+
+#define P8_FENCE asm volatile("" ::: "memory")
+#define P8_BAR                                                               \
+  P8_FENCE;                                                                  \
+  __builtin_amdgcn_s_barrier();                                              \
+  P8_FENCE
+#define P8_LGKM asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory")
+
+  // ---- prologue: tile 0 fully, then tile 1's images in steady-state
+  // issue order (img2=B0 as a phase-3 would, img0/img1=A as a phase-4)
+  P8_GLDS1(0, 0, kbeg)
+  P8_GLDS1(0, 1, kbeg)
+  P8_GLDS1(0, 2, kbeg)
+  P8_GLDS1(0, 3, kbeg)
+  if (NT > 1) {
+    P8_GLDS1(1, 2, kbeg + BK)
+    P8_GLDS1(1, 0, kbeg + BK)
+    P8_GLDS1(1, 1, kbeg + BK)
+    asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  } else {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  }
+  P8_BAR;
+
+  for (int ti = 0; ti < NT; ++ti) {
+    const int buf = ti & 1;
+    const int kt = kbeg + ti * BK;
+    // phase 1: quad (0,0); stage next tile's last image (other buffer)
+    P8_RD_A(buf, 0)
+    P8_RD_B(buf, 0)
+    if (ti + 1 < NT) P8_GLDS1(buf ^ 1, 3, kt + BK)
+    P8_BAR;
+    P8_LGKM;
+    P8_MFMA(0, 0)
+    P8_BAR;
+    // phase 2: quad (0,1) — last B reads of this tile's buffer
+    P8_RD_B(buf, 1)
+    P8_BAR;
+    P8_LGKM;
+    P8_MFMA(0, 1)
+    P8_BAR;
+    // phase 3: quad (1,0) — last A reads; stage tile+2's first B image
+    // into this buffer (B reads completed at phase 2's closing barrier)
+    P8_RD_A(buf, 1)
+    if (ti + 2 < NT) P8_GLDS1(buf, 2, kt + 2 * BK)
+    P8_BAR;
+    P8_LGKM;
+    P8_MFMA(1, 0)
+    P8_BAR;
+    // phase 4: quad (1,1); stage tile+2's A images (A reads ended at
+    // phase 3's closing barrier)
+    if (ti + 2 < NT) {
+      P8_GLDS1(buf, 0, kt + 2 * BK)
+      P8_GLDS1(buf, 1, kt + 2 * BK)
+    }
+    P8_MFMA(1, 1)
+    if (ti + 1 < NT) {
+      if (ti + 2 < NT)
+        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    P8_BAR;
+  }
+#undef P8_GLDS1
+#undef P8_FRAG
+#undef P8_RD_A
+#undef P8_RD_B
+#undef P8_MFMA
+
+  // ---- epilogue: D frag (mi, ni): row = mi*16 + lg*4 + r, col = ni*16+l16
+  const int mw = m0 + wr * 128, nw = n0 + wc * 64;
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      const int nn = nw + ni * 16 + l16;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int mm = mw + mi * 16 + lg * 4 + r;
+        if (mm >= Mr) continue;
+        float v = alpha * acc[mi][ni][r];
+        if (OUT == BF_OUT_F32_ATOMIC) {
+          atomicAdd(&Cf[(int64_t)mm * ldc + nn], v);
+        } else {
+          if (bias) v += bias[nn];
+          if (R) v += bf2f(R[(int64_t)mm * ldc + nn]);
+          if (OUT == BF_OUT_F32) {
+            if (beta != 0.f) v += beta * Cf[(int64_t)mm * ldc + nn];
+            Cf[(int64_t)mm * ldc + nn] = v;
+          } else {
+            if (beta != 0.f) v += beta * bf2f(Cb[(int64_t)mm * ldc + nn]);
+            Cb[(int64_t)mm * ldc + nn] = (__bf16)v;
+          }
+        }
+      }
+    }
+  }
+}
+
+extern "C" int ob_gemm_bf16_nt_8ph(const void* A, const void* B, void* C,
+                                   const void* bias, const void* residual,
+                                   int64_t M, int64_t N, int64_t K,
+                                   int64_t lda, int64_t ldb, int64_t ldc,
+                                   int64_t sA1, int64_t sA2, int64_t sB1,
+                                   int64_t sB2, int64_t sC1, int64_t sC2,
+                                   int64_t n1, int64_t n2, float alpha,
+                                   float beta, int out_kind, int splitk,
+                                   void* stream, int64_t Mr) {
+  if (M % 256 || N % 256 || K % 64)
+    return ob_fail("nt_8ph: M,N %% 256, K %% 64 required");
+  const int nbm = (int)(M / 256), nbn = (int)(N / 256);
+  dim3 grid(nbm * nbn, splitk < 1 ? 1 : splitk, (unsigned)(n1 * n2));
+  dim3 block(512);
+  static const int swz = [] {
+    const char* e = getenv("OB_P8_SWZ");
+    return e ? atoi(e) : 0;
+  }();
+#define OB_P8(OUT_, SWZ_)                                                    \
+  k_gemm_bf16_nt_8ph<OUT_, SWZ_><<<grid, block, 0, S(stream)>>>(             \
+      (const __bf16*)A, (const __bf16*)B, C, (const float*)bias,             \
+      (const __bf16*)residual, (int)M, (int)N, (int)K, lda, ldb, ldc, sA1,   \
+      sA2, sB1, sB2, sC1, sC2, (int)n2, alpha, beta, nbn, (int)Mr)
+#define OB_P8S(OUT_)                                                         \
+  do {                                                                       \
+    if (swz == 1) OB_P8(OUT_, 1);                                            \
+    else OB_P8(OUT_, 0);                                                     \
+  } while (0)
+  if (out_kind == BF_OUT_BF16) OB_P8S(BF_OUT_BF16);
+  else if (out_kind == BF_OUT_F32) OB_P8S(BF_OUT_F32);
+  else OB_P8S(BF_OUT_F32_ATOMIC);
+#undef OB_P8S
+#undef OB_P8
   OB_LAUNCH_CHECK();
   return 0;
 }
