@@ -1778,7 +1778,6 @@ __global__ __launch_bounds__(256, 2) void k_flash_bwd_dkdv(
     const __bf16* __restrict__ dOT, const __bf16* __restrict__ dO,
     const float* __restrict__ lse, const float* __restrict__ D,
     __bf16* __restrict__ dqkv, int Sq, int H, int nh, float scale) {
-  __shared__ float lse_s[32], d_s[32];
   const int z = blockIdx.z;
   const int b = z / nh, h = z % nh;
   const int64_t base = (int64_t)b * Sq * 3 * H + h * 64;
@@ -1814,9 +1813,12 @@ __global__ __launch_bounds__(256, 2) void k_flash_bwd_dkdv(
   for (int qt = qt0; qt < nqt; ++qt) {
     const int q0 = qt * 32;
     if (q0 + 31 < kv0) continue;  // wave-uniform fully-masked
-    // lse/D broadcast for this q-tile (wave-local LDS is NOT enough here:
-    // the arrays are indexed by q-row across all lanes; load per wave into
-    // registers via per-lane reads instead)
+    // lse/D for this q-tile: one coalesced load per half-wave (lane il
+    // holds row q0+il), fanned out by bpermute below — NOT per-element
+    // global loads (32 scalar loads/tile measured as the kernel's
+    // dominant stall).
+    const float lse_t = lsep[q0 + il];
+    const float d_t = Dp[q0 + il];
     // S-acc: mfma(A=Q(i=q), B=K(j=kv))  -> col = kv, rows = q
     f32x16 sacc = {}, dpacc = {};
 #pragma unroll
@@ -1833,11 +1835,14 @@ __global__ __launch_bounds__(256, 2) void k_flash_bwd_dkdv(
     float pv[16], dsv[16];
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
-      const int q = q0 + (r & 3) + 8 * (r >> 2) + 4 * kh;
+      const int qoff = (r & 3) + 8 * (r >> 2) + 4 * kh;
+      const int q = q0 + qoff;
       const bool ok = q >= mykv;
-      const float p = ok ? __expf(sacc[r] * scale - lsep[q]) : 0.f;
+      const float lse_q = __shfl(lse_t, qoff, 64);
+      const float d_q = __shfl(d_t, qoff, 64);
+      const float p = ok ? __expf(sacc[r] * scale - lse_q) : 0.f;
       pv[r] = p;
-      dsv[r] = ok ? p * (dpacc[r] - Dp[q]) : 0.f;
+      dsv[r] = ok ? p * (dpacc[r] - d_q) : 0.f;
     }
     // dV += P^T dO   (A = P^T frag: i = kv = lane col; k = q-run)
     // dK += scale * dS^T Q
@@ -1871,8 +1876,6 @@ __global__ __launch_bounds__(256, 2) void k_flash_bwd_dkdv(
     dKp[(int64_t)kv * 3 * H + il] = (__bf16)(scale * dk0[r]);
     dKp[(int64_t)kv * 3 * H + 32 + il] = (__bf16)(scale * dk1[r]);
   }
-  (void)lse_s;
-  (void)d_s;
 }
 
 __global__ __launch_bounds__(256, 2) void k_flash_bwd_dq(

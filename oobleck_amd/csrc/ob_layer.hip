@@ -184,11 +184,12 @@ extern "C" int ob_layer_create(const ob_layer_desc* d, ob_layer_t* out) {
       return ob_fail("create: ids hipMalloc failed");
     }
   }
-  // pad vocab to a 128 multiple: makes the lm_head GEMM family
+  // pad vocab to a 256 multiple: makes the lm_head GEMM family
   // (logits fwd, d_lnout, dW_lm via transposes) interior shapes for
-  // the glds path; shadow pad rows stay zero so padded logits/grads
-  // are exactly zero
-  l->v_pad = (V + 127) / 128 * 128;
+  // the glds path AND the 256^2 8-phase kernel (gpt2's 50257 at 128
+  // padding lands on 50304 % 256 != 0); shadow pad rows stay zero so
+  // padded logits/grads are exactly zero
+  l->v_pad = (V + 255) / 256 * 256;
   if (d->dtype == 1) {
     int64_t o2 = 0;
     if (d->kind == OB_KIND_BLOCK) {
