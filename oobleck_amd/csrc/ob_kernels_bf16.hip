@@ -539,6 +539,111 @@ extern "C" int ob_layernorm_fwd_bf16(const void* x, const void* w,
 }
 
 #define BLN_CHUNK 16
+#define BLN_WROWS 8
+// wave-per-row ln backward (H <= 1024): each of the block's 4 waves owns
+// its rows outright -- the row reductions are wave shfl ladders, no LDS
+// and no per-row __syncthreads (the block-per-row kernel serialized two
+// block reductions per row); 16-B vector loads throughout.  dw/db fold
+// lane accumulators -> LDS (once) -> one global atomic per column per
+// block.
+template <bool DX_ACCUM>
+__global__ __launch_bounds__(256) void k_ln_bwd_bf16_w(
+    const __bf16* __restrict__ x, const float* __restrict__ w,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const __bf16* __restrict__ dy, __bf16* __restrict__ dx,
+    float* __restrict__ dw, float* __restrict__ db, int64_t rows, int H) {
+  __shared__ float redw[4][1024];
+  __shared__ float redb[4][1024];
+  const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+  const int c0 = lane * 8, c1 = (lane + 64) * 8;
+  const bool has1 = c1 < H;
+  float accw[2][8] = {}, accb[2][8] = {};
+  float wv0[8], wv1[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    wv0[j] = w[c0 + j];
+    wv1[j] = has1 ? w[c1 + j] : 0.f;
+  }
+  const int64_t r0 = (int64_t)blockIdx.x * (4 * BLN_WROWS) + wid;
+  for (int i = 0; i < BLN_WROWS; ++i) {
+    const int64_t row = r0 + (int64_t)i * 4;
+    if (row >= rows) break;
+    const __bf16* xr = x + row * H;
+    const __bf16* dyr = dy + row * H;
+    const float mu = mean[row], rs = rstd[row];
+    const uint4 xv0 = *reinterpret_cast<const uint4*>(xr + c0);
+    const uint4 yv0 = *reinterpret_cast<const uint4*>(dyr + c0);
+    uint4 xv1 = {}, yv1 = {};
+    if (has1) {
+      xv1 = *reinterpret_cast<const uint4*>(xr + c1);
+      yv1 = *reinterpret_cast<const uint4*>(dyr + c1);
+    }
+    float s1 = 0.f, s2 = 0.f;
+    float xh0[8], dy0[8], xh1[8], dy1[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      xh0[j] = (bf2f(bf_extract(xv0, j)) - mu) * rs;
+      dy0[j] = bf2f(bf_extract(yv0, j));
+      const float dyw = dy0[j] * wv0[j];
+      s1 += dyw * xh0[j];
+      s2 += dyw;
+      accw[0][j] += dy0[j] * xh0[j];
+      accb[0][j] += dy0[j];
+    }
+    if (has1) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        xh1[j] = (bf2f(bf_extract(xv1, j)) - mu) * rs;
+        dy1[j] = bf2f(bf_extract(yv1, j));
+        const float dyw = dy1[j] * wv1[j];
+        s1 += dyw * xh1[j];
+        s2 += dyw;
+        accw[1][j] += dy1[j] * xh1[j];
+        accb[1][j] += dy1[j];
+      }
+    }
+#pragma unroll
+    for (int o = 32; o > 0; o >>= 1) {
+      s1 += __shfl_xor(s1, o, 64);
+      s2 += __shfl_xor(s2, o, 64);
+    }
+    const float m1 = s1 / H, m2 = s2 / H;
+    __bf16 ox[16];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float v = rs * (dy0[j] * wv0[j] - m2 - xh0[j] * m1);
+      ox[j] = (__bf16)(DX_ACCUM ? bf2f(dx[row * H + c0 + j]) + v : v);
+    }
+    if (has1) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float v = rs * (dy1[j] * wv1[j] - m2 - xh1[j] * m1);
+        ox[8 + j] = (__bf16)(DX_ACCUM ? bf2f(dx[row * H + c1 + j]) + v : v);
+      }
+    }
+    *reinterpret_cast<uint4*>(dx + row * H + c0) =
+        *reinterpret_cast<const uint4*>(&ox[0]);
+    if (has1)
+      *reinterpret_cast<uint4*>(dx + row * H + c1) =
+          *reinterpret_cast<const uint4*>(&ox[8]);
+  }
+  // dw/db: lane accs -> LDS (per wave) -> wave 0 folds -> one atomic/col
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    redw[wid][c0 + j] = accw[0][j];
+    redb[wid][c0 + j] = accb[0][j];
+    if (has1) {
+      redw[wid][c1 + j] = accw[1][j];
+      redb[wid][c1 + j] = accb[1][j];
+    }
+  }
+  __syncthreads();
+  for (int c = threadIdx.x; c < H; c += 256) {
+    atomicAdd(&dw[c], redw[0][c] + redw[1][c] + redw[2][c] + redw[3][c]);
+    atomicAdd(&db[c], redb[0][c] + redb[1][c] + redb[2][c] + redb[3][c]);
+  }
+}
+
 template <bool DX_ACCUM>
 __global__ __launch_bounds__(256) void k_ln_bwd_bf16(
     const __bf16* __restrict__ x, const float* __restrict__ w,
@@ -598,6 +703,21 @@ extern "C" int ob_layernorm_bwd_bf16(const void* x, const void* w,
                                      int dx_accum, void* stream) {
   if (H > 2048) return ob_fail("ln_bwd_bf16: H > 2048 unsupported");
   if (H % 8) return ob_fail("ln_bwd_bf16: H must be a multiple of 8");
+  if (H >= 512 && H <= 1024) {
+    const int grid = (int)((rows + 4 * BLN_WROWS - 1) / (4 * BLN_WROWS));
+    if (dx_accum)
+      k_ln_bwd_bf16_w<true><<<grid, 256, 0, S(stream)>>>(
+          (const __bf16*)x, (const float*)w, (const float*)mean,
+          (const float*)rstd, (const __bf16*)dy, (__bf16*)dx, (float*)dw,
+          (float*)db, rows, (int)H);
+    else
+      k_ln_bwd_bf16_w<false><<<grid, 256, 0, S(stream)>>>(
+          (const __bf16*)x, (const float*)w, (const float*)mean,
+          (const float*)rstd, (const __bf16*)dy, (__bf16*)dx, (float*)dw,
+          (float*)db, rows, (int)H);
+    OB_LAUNCH_CHECK();
+    return 0;
+  }
   const int grid = (int)((rows + BLN_CHUNK - 1) / BLN_CHUNK);
   if (dx_accum)
     k_ln_bwd_bf16<true><<<grid, 256, 0, S(stream)>>>(
@@ -848,12 +968,68 @@ __global__ __launch_bounds__(256) void k_colsum_bf16_v8(
 #pragma unroll
   for (int j = 0; j < 8; ++j) atomicAdd(&db[c8 + j], acc[j]);
 }
+
+// two-level column sum (v9): block = 8 col-threads x 32 row-threads over
+// a 64-col x 128-row tile; each thread keeps 4 independent uint4 loads
+// in flight (the v8 kernel's single serial load stream measured ~640
+// GB/s); wave shfl + LDS tree reduction, ONE global atomic per column
+// per block.
+__global__ __launch_bounds__(256) void k_colsum_bf16_v9(
+    const __bf16* __restrict__ X, float* __restrict__ db, int64_t M,
+    int64_t N) {
+  __shared__ float red[4][64];
+  const int tid = threadIdx.x;
+  const int cg = tid & 7, rt = tid >> 3;       // col-thread, row-thread
+  const int lane = tid & 63, w = tid >> 6;
+  const int64_t c0 = (int64_t)blockIdx.x * 64 + cg * 8;
+  const int64_t r0 = (int64_t)blockIdx.y * 128 + rt;
+  float acc[8] = {0};
+  if (c0 + 7 < N) {
+    const __bf16* Xp = X + r0 * N + c0;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int64_t r = r0 + i * 32;
+      if (r < M) {
+        const uint4 v = *reinterpret_cast<const uint4*>(Xp + (int64_t)i * 32 * N);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc[j] += bf2f(bf_extract(v, j));
+      }
+    }
+  } else if (c0 < N) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int64_t r = r0 + i * 32;
+      if (r < M)
+        for (int j = 0; j < (int)(N - c0); ++j)
+          acc[j] += bf2f(X[r * N + c0 + j]);
+    }
+  }
+  // wave: fold the 8 row-slots (lanes cg+8*s, s=0..7)
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    acc[j] += __shfl_down(acc[j], 32, 64);
+    acc[j] += __shfl_down(acc[j], 16, 64);
+    acc[j] += __shfl_down(acc[j], 8, 64);
+  }
+  if (lane < 8) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) red[w][lane * 8 + j] = acc[j];
+  }
+  __syncthreads();
+  // first 64 threads: fold 4 waves, one atomic per column
+  if (tid < 64) {
+    const int64_t c = (int64_t)blockIdx.x * 64 + tid;
+    if (c < N)
+      atomicAdd(&db[c],
+                red[0][tid] + red[1][tid] + red[2][tid] + red[3][tid]);
+  }
+}
+
 extern "C" int ob_colsum_bf16(const void* X, void* db, int64_t M, int64_t N,
                               void* stream) {
   if (N % 8 == 0) {
-    // short row chunks keep the grid chip-filling even at N = H (96 lanes)
-    dim3 grid((unsigned)((N / 8 + 255) / 256), (unsigned)((M + 63) / 64));
-    k_colsum_bf16_v8<<<grid, 256, 0, S(stream)>>>((const __bf16*)X,
+    dim3 grid((unsigned)((N + 63) / 64), (unsigned)((M + 127) / 128));
+    k_colsum_bf16_v9<<<grid, 256, 0, S(stream)>>>((const __bf16*)X,
                                                   (float*)db, M, N);
   } else {
     dim3 grid((unsigned)((N + 255) / 256), (unsigned)((M + 255) / 256));

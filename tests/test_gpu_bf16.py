@@ -341,3 +341,47 @@ def test_bf16_block_layer_parity_flash():
     assert rel_l2(out.float().cpu(), ref_out) < 5e-2
     assert rel_l2(din.float().cpu(), ref_dx) < 5e-2
     assert rel_l2(layer.flat_grad.cpu(), ref_grad) < 5e-2
+
+
+@requires_gpu
+@pytest.mark.parametrize("R,H", [(512, 768), (300, 1024), (1000, 512)])
+def test_ln_bwd_bf16_wave(R, H):
+    """Wave-per-row ln backward (H>=512 path) vs torch fp32 on the same
+    bf16-rounded inputs."""
+    from oobleck_amd._ext import check, get_ext
+    g = torch.Generator().manual_seed(5)
+    x = (torch.randn(R, H, generator=g)).cuda().bfloat16()
+    dy = (torch.randn(R, H, generator=g) * 0.5).cuda().bfloat16()
+    w = torch.randn(H, generator=g).cuda().float()
+    b = torch.randn(H, generator=g).cuda().float()
+    xf = x.float().requires_grad_(True)
+    mu = xf.mean(-1, keepdim=True)
+    var = xf.var(-1, unbiased=False, keepdim=True)
+    y = (xf - mu) / torch.sqrt(var + 0.0) * w + b
+    y.backward(dy.float())
+    mean = x.float().mean(-1).contiguous()
+    rstd = (1.0 / x.float().var(-1, unbiased=False).sqrt()).contiguous()
+    dx = torch.empty_like(x)
+    dw = torch.zeros(H, device=DEV, dtype=torch.float32)
+    db = torch.zeros(H, device=DEV, dtype=torch.float32)
+    check(get_ext().ob_layernorm_bwd_bf16(
+        ptr(x), ptr(w), ptr(mean), ptr(rstd), ptr(dy), ptr(dx), ptr(dw),
+        ptr(db), R, H, 0, stream()), "lnbwd")
+    torch.cuda.synchronize()
+    assert rel_l2(dx.float(), xf.grad) < 3e-2
+    assert rel_l2(dw, (dy.float() * (x.float() - mu) *
+                       (var + 0.0).rsqrt()).sum(0)) < 2e-2
+    assert rel_l2(db, dy.float().sum(0)) < 2e-2
+
+
+@requires_gpu
+@pytest.mark.parametrize("M,N", [(8192, 768), (1000, 3072), (129, 520),
+                                 (64, 40)])
+def test_colsum_bf16_v9(M, N):
+    from oobleck_amd._ext import check, get_ext
+    g = torch.Generator().manual_seed(6)
+    x = (torch.randn(M, N, generator=g)).cuda().bfloat16()
+    db = torch.zeros(N, device=DEV, dtype=torch.float32)
+    check(get_ext().ob_colsum_bf16(ptr(x), ptr(db), M, N, stream()), "cs")
+    torch.cuda.synchronize()
+    assert rel_l2(db, x.float().sum(0)) < 2e-2
