@@ -87,6 +87,15 @@ int ob_flash_bwd_bf16(const void* qkv, const void* QT, const void* KT,
                       const void* D, void* dqkv, int64_t B, int64_t Sq,
                       int64_t H, int64_t nh, float scale, void* stream);
 }
+extern "C" int ob_gemm_bf16_nt_8ph(const void* A, const void* B, void* C,
+                                   const void* bias, const void* residual,
+                                   int64_t M, int64_t N, int64_t K,
+                                   int64_t lda, int64_t ldb, int64_t ldc,
+                                   int64_t sA1, int64_t sA2, int64_t sB1,
+                                   int64_t sB2, int64_t sC1, int64_t sC2,
+                                   int64_t n1, int64_t n2, float alpha,
+                                   float beta, int out_kind, int splitk,
+                                   void* stream, int64_t Mr);
 // fast NT glds dispatch (interior 128-tiled M/N; Mr = store-row guard)
 int ob_gemm_bf16_nt_dispatch(const void* A, const void* B, void* C,
                              const void* bias, const void* residual,
