@@ -348,20 +348,11 @@ extern "C" int ob_gemm_bf16(int transA, int transB, int64_t M, int64_t N,
           int64_t sB2, int64_t sC1, int64_t sC2, int64_t n1, int64_t n2,
           float alpha, float beta, int out_kind, int splitk, void* stream,
           int64_t Mr);
-      int sk = splitk < 1 ? 1 : splitk;
-      if (out_kind == BF_OUT_F32_ATOMIC) {
-        const int64_t t1tiles = (M / 256) * (N / 256) * n1 * n2;
-        double best = 0.0;
-        for (int c = 1; c <= 16; ++c) {
-          if (K / ((int64_t)c * 64) < 8) break;  // keep >=8 K-tiles/chunk
-          const int64_t t = t1tiles * c;
-          const double eff = (double)t / (((t + 255) / 256) * 256);
-          if (eff > best + 1e-9) {
-            best = eff;
-            sk = c;
-          }
-        }
-      }
+      // NOTE: an auto-splitk search for atomic outputs was tried here
+      // (round-filling sk up to 16) and REGRESSED the step 360k->347k
+      // tok/s: split-k multiplies the f32 atomic output traffic by sk,
+      // which dominates on the (W1, W2, BS) weight-grad shapes.
+      const int sk = splitk < 1 ? 1 : splitk;
       const int64_t tiles = (M / 256) * (N / 256) * n1 * n2 * sk;
       const double eff = (double)tiles / (((tiles + 255) / 256) * 256);
       if (tiles >= 192 && eff >= 0.9)

@@ -709,34 +709,10 @@ static int final_backward_bf16(ob_layer* l, int slot, const float* dout,
     __bf16* LNT = (__bf16*)g_ws.t2;
     if (ob_transpose_bf16(logits, DLT, BS, l->v_pad, stream)) return 1;
     if (ob_transpose_bf16(lnf, LNT, BS, H, stream)) return 1;
-    // 8-phase kernel when the (v_pad, H) grid is 256-tileable: pick the
-    // K split for round fill (atomic out: grads accumulate, split free)
-    if (l->v_pad % 256 == 0 && H % 256 == 0 && BS % 64 == 0) {
-      const int64_t t1 = (l->v_pad / 256) * (H / 256);
-      int sk = 1;
-      double best = 0.0;
-      for (int c = 1; c <= 16; ++c) {
-        if (BS / ((int64_t)c * 64) < 8) break;
-        const int64_t t = t1 * c;
-        const double eff = (double)t / (((t + 255) / 256) * 256);
-        if (eff > best + 1e-9) {
-          best = eff;
-          sk = c;
-        }
-      }
-      if (t1 * sk >= 192 && best >= 0.9) {
-        if (ob_gemm_bf16_nt_8ph(DLT, LNT, g + 2 * H, nullptr, nullptr,
-                                l->v_pad, H, BS, BS, BS, H, 0, 0, 0, 0, 0, 0,
-                                1, 1, 1.f, 0.f, 2, sk, stream, V))
-          return 1;
-        goto dwlm_done;
-      }
-    }
     if (ob_gemm_bf16_nt_dispatch(DLT, LNT, g + 2 * H, nullptr, nullptr,
                                  l->v_pad, H, BS, BS, BS, H, 0, 0, 0, 0, 0,
                                  0, 1, 1, 1.f, 0.f, 2, 2, stream, V))
       return 1;
-  dwlm_done:;
   }
   // d_lnout: K = v_pad (padded dlogits cols and shadow^T cols are zero)
   if (gemm_bf(0, 1, BS, H, l->v_pad, 1.f, logits, l->v_pad, 0, 0,
