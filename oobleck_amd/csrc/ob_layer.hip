@@ -31,10 +31,36 @@ struct Workspace {
   float* b4h = nullptr;    // [B, S, 4H]
   float* t1 = nullptr;     // bf16 [dim, BS] transposed operand (as floats)
   float* t2 = nullptr;
+  float* s1 = nullptr;     // side-stream dW transpose buffers (bf16 as
+  float* s2 = nullptr;     // floats): the dW family runs on g_side
   int64_t sz_dp = 0, sz_bsh = 0, sz_bsh2 = 0, sz_dqkv = 0, sz_b4h = 0,
-          sz_t1 = 0, sz_t2 = 0;
+          sz_t1 = 0, sz_t2 = 0, sz_s1 = 0, sz_s2 = 0;
 };
 static Workspace g_ws;
+
+// side stream + events for overlapping the weight-grad family (dW
+// transposes + atomic GEMMs + bias colsums: HBM-heavy, independent of
+// the dX chain) under the main stream's dX/attention work.  All
+// enqueued host-side in order, so event reuse across calls is safe.
+struct SideSync {
+  hipStream_t stream = nullptr;
+  hipEvent_t e1 = nullptr, e2 = nullptr, e3 = nullptr;  // main -> side
+  hipEvent_t s2 = nullptr, sf = nullptr;                // side -> main
+  bool ready = false;
+};
+static SideSync g_side;
+
+static int side_init() {
+  if (g_side.ready) return 0;
+  OB_HIP(hipStreamCreateWithFlags(&g_side.stream, hipStreamNonBlocking));
+  OB_HIP(hipEventCreateWithFlags(&g_side.e1, hipEventDisableTiming));
+  OB_HIP(hipEventCreateWithFlags(&g_side.e2, hipEventDisableTiming));
+  OB_HIP(hipEventCreateWithFlags(&g_side.e3, hipEventDisableTiming));
+  OB_HIP(hipEventCreateWithFlags(&g_side.s2, hipEventDisableTiming));
+  OB_HIP(hipEventCreateWithFlags(&g_side.sf, hipEventDisableTiming));
+  g_side.ready = true;
+  return 0;
+}
 
 static int ws_ensure(float** buf, int64_t* cur, int64_t need) {
   if (need <= *cur) return 0;
@@ -222,6 +248,9 @@ extern "C" int ob_layer_create(const ob_layer_desc* d, ob_layer_t* out) {
     if (d->dtype == 1) {
       if (ws_ensure(&g_ws.t1, &g_ws.sz_t1, (4 * H * BS + 1) / 2)) return 1;
       if (ws_ensure(&g_ws.t2, &g_ws.sz_t2, (4 * H * BS + 1) / 2)) return 1;
+      if (ws_ensure(&g_ws.s1, &g_ws.sz_s1, (4 * H * BS + 1) / 2)) return 1;
+      if (ws_ensure(&g_ws.s2, &g_ws.sz_s2, (4 * H * BS + 1) / 2)) return 1;
+      if (side_init()) return 1;
     }
   }
   if (d->kind == OB_KIND_FINAL && d->dtype == 1) {
@@ -453,20 +482,27 @@ static int gemm_bf(int tA, int tB, int64_t M, int64_t N, int64_t K,
 // weight-grad GEMM for bf16: materialize X^T / dY^T (cheap HBM transpose)
 // so the GEMM runs on the fast NT glds path instead of the
 // transpose-staged TN case (~150 TF measured there).
-static int dw_bf16(const __bf16* act, int64_t actw, const __bf16* dY,
-                   int64_t dyw, int64_t BS, float* gout, int64_t ldc,
-                   void* stream) {
+static int dw_bf16_ws(const __bf16* act, int64_t actw, const __bf16* dY,
+                      int64_t dyw, int64_t BS, float* gout, int64_t ldc,
+                      float* w1, float* w2, void* stream) {
   if ((actw % 128) || (dyw % 128) || (BS % 128))
     return gemm_bf(1, 0, actw, dyw, BS, 1.f, act, actw, 0, 0, dY, dyw, 0, 0,
                    gout, ldc, 0, 0, 1, 1, nullptr, nullptr, 2,
                    pick_splitk(actw, dyw, BS), stream);
-  __bf16* XT = (__bf16*)g_ws.t1;
-  __bf16* DYT = (__bf16*)g_ws.t2;
+  __bf16* XT = (__bf16*)w1;
+  __bf16* DYT = (__bf16*)w2;
   if (ob_transpose_bf16(act, XT, BS, actw, stream)) return 1;
   if (ob_transpose_bf16(dY, DYT, BS, dyw, stream)) return 1;
   return gemm_bf(0, 1, actw, dyw, BS, 1.f, XT, BS, 0, 0, DYT, BS, 0, 0, gout,
                  ldc, 0, 0, 1, 1, nullptr, nullptr, 2,
                  pick_splitk(actw, dyw, BS), stream);
+}
+
+static int dw_bf16(const __bf16* act, int64_t actw, const __bf16* dY,
+                   int64_t dyw, int64_t BS, float* gout, int64_t ldc,
+                   void* stream) {
+  return dw_bf16_ws(act, actw, dY, dyw, BS, gout, ldc, g_ws.t1, g_ws.t2,
+                    stream);
 }
 
 static bool use_flash(const ob_layer* l) { return l->flash; }
@@ -602,19 +638,32 @@ static int block_backward_bf16(ob_layer* l, int slot, const __bf16* dout,
   __bf16* DQKV = (__bf16*)g_ws.dqkv;
   __bf16* DP = (__bf16*)g_ws.dp;
 
+  // The dW family (transposes + atomic GEMMs + bias colsums) has no
+  // consumer inside this call and no effect on the dX chain: it runs on
+  // g_side, overlapped under the dX/attention work, fenced by events at
+  // the producer edges and JOINED at the end of the call (the g_ws
+  // buffers it reads are reused by the next microbatch's backward).
+  void* const side = (void*)g_side.stream;
+  OB_HIP(hipEventRecord(g_side.e1, S(stream)));  // entry state (dout ready)
+  OB_HIP(hipStreamWaitEvent(g_side.stream, g_side.e1, 0));
+
   OB_HIP(hipMemcpyAsync(din, dout, BS * H * sizeof(__bf16),
                         hipMemcpyDeviceToDevice, S(stream)));
   // ---- MLP ----
   if (gemm_bf(0, 1, BS, 4 * H, H, 1.f, dout, H, 0, 0, sh + l->sh_mp, H, 0, 0,
               DY4, 4 * H, 0, 0, 1, 1, nullptr, nullptr, 0, 1, stream))
     return 1;
-  if (dw_bf16(gact, 4 * H, dout, H, BS, g + bp.w_mlpproj, H, stream))
+  if (dw_bf16_ws(gact, 4 * H, dout, H, BS, g + bp.w_mlpproj, H, g_ws.s1,
+                 g_ws.s2, side))
     return 1;
-  if (ob_colsum_bf16(dout, g + bp.b_mlpproj, BS, H, stream)) return 1;
+  if (ob_colsum_bf16(dout, g + bp.b_mlpproj, BS, H, side)) return 1;
   if (ob_gelu_bwd_bf16(u, DY4, DY4, BS * 4 * H, stream)) return 1;
-  if (dw_bf16(ln2, H, DY4, 4 * H, BS, g + bp.w_fc, 4 * H, stream))
+  OB_HIP(hipEventRecord(g_side.e2, S(stream)));  // DY4 post-gelu
+  OB_HIP(hipStreamWaitEvent(g_side.stream, g_side.e2, 0));
+  if (dw_bf16_ws(ln2, H, DY4, 4 * H, BS, g + bp.w_fc, 4 * H, g_ws.s1,
+                 g_ws.s2, side))
     return 1;
-  if (ob_colsum_bf16(DY4, g + bp.b_fc, BS, 4 * H, stream)) return 1;
+  if (ob_colsum_bf16(DY4, g + bp.b_fc, BS, 4 * H, side)) return 1;
   if (gemm_bf(0, 1, BS, H, 4 * H, 1.f, DY4, 4 * H, 0, 0, sh + l->sh_fc,
               4 * H, 0, 0, DLN, H, 0, 0, 1, 1, nullptr, nullptr, 0, 1,
               stream))
@@ -624,12 +673,18 @@ static int block_backward_bf16(ob_layer* l, int slot, const __bf16* dout,
                             g + bp.ln2_b, BS, H, 1, stream))
     return 1;
   // ---- attention projection ----
+  OB_HIP(hipEventRecord(g_side.e3, S(stream)));  // din post-ln2-bwd
+  OB_HIP(hipStreamWaitEvent(g_side.stream, g_side.e3, 0));
   if (gemm_bf(0, 1, BS, H, H, 1.f, din, H, 0, 0, sh + l->sh_ap, H, 0, 0,
               DATT, H, 0, 0, 1, 1, nullptr, nullptr, 0, 1, stream))
     return 1;
-  if (dw_bf16(am, H, din, H, BS, g + bp.w_attnproj, H, stream))
+  if (dw_bf16_ws(am, H, din, H, BS, g + bp.w_attnproj, H, g_ws.s1, g_ws.s2,
+                 side))
     return 1;
-  if (ob_colsum_bf16(din, g + bp.b_attnproj, BS, H, stream)) return 1;
+  if (ob_colsum_bf16(din, g + bp.b_attnproj, BS, H, side)) return 1;
+  // din is re-updated by the final ln1 backward: the main stream must
+  // not reach it before the side finished reading din
+  OB_HIP(hipEventRecord(g_side.s2, g_side.stream));
   // ---- attention core ----
   if (use_flash(l)) {
     const int64_t BSH = BS * H;
@@ -671,17 +726,25 @@ static int block_backward_bf16(ob_layer* l, int slot, const __bf16* dout,
       return 1;
   }
   // ---- QKV projection ----
-  if (ob_colsum_bf16(DQKV, g + bp.b_qkv, BS, 3 * H, stream)) return 1;
-  if (dw_bf16(ln1, H, DQKV, 3 * H, BS, g + bp.w_qkv, 3 * H, stream))
+  OB_HIP(hipEventRecord(g_side.e1, S(stream)));  // DQKV ready
+  OB_HIP(hipStreamWaitEvent(g_side.stream, g_side.e1, 0));
+  if (ob_colsum_bf16(DQKV, g + bp.b_qkv, BS, 3 * H, side)) return 1;
+  if (dw_bf16_ws(ln1, H, DQKV, 3 * H, BS, g + bp.w_qkv, 3 * H, g_ws.s1,
+                 g_ws.s2, side))
     return 1;
   if (gemm_bf(0, 1, BS, H, 3 * H, 1.f, DQKV, 3 * H, 0, 0, sh + l->sh_qkv,
               3 * H, 0, 0, DLN, H, 0, 0, 1, 1, nullptr, nullptr, 0, 1,
               stream))
     return 1;
+  // ln1 backward accumulates into din: wait for the side's din readers
+  OB_HIP(hipStreamWaitEvent(S(stream), g_side.s2, 0));
   if (ob_layernorm_bwd_bf16(x, p + bp.ln1_w, st + l->o_mean1,
                             st + l->o_rstd1, DLN, din, g + bp.ln1_w,
                             g + bp.ln1_b, BS, H, 1, stream))
     return 1;
+  // join: the next call reuses DY4/DQKV/din workspaces on the main stream
+  OB_HIP(hipEventRecord(g_side.sf, g_side.stream));
+  OB_HIP(hipStreamWaitEvent(S(stream), g_side.sf, 0));
   return 0;
 }
 
