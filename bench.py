@@ -181,9 +181,11 @@ def main():
     ap.add_argument("--seq-len", type=int, default=1024)
     ap.add_argument("--microbatch", type=int, default=8)
     ap.add_argument("--global-batch", type=int, default=128)
-    ap.add_argument("--dtype", choices=["f32", "bf16"], default="f32",
-                    help="compute dtype (f32 = the reference dtype; "
-                         "bf16 = mixed precision, SURVEY §8 f4)")
+    ap.add_argument("--dtype", choices=["f32", "bf16"], default="bf16",
+                    help="compute dtype (default bf16 mixed precision with "
+                         "fp32 masters/grads — north_star's 'MFMA-bf16 "
+                         "roofline' target dtype; f32 = the reference's "
+                         "own compute dtype, kept as the parity anchor)")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     ap.add_argument("--skip-roofline", action="store_true")
     ap.add_argument("--probe", choices=["gemm"], default=None,

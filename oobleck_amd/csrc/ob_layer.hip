@@ -256,6 +256,9 @@ extern "C" int ob_layer_create(const ob_layer_desc* d, ob_layer_t* out) {
   if (d->kind == OB_KIND_FINAL && d->dtype == 1) {
     if (ws_ensure(&g_ws.t1, &g_ws.sz_t1, (l->v_pad * BS + 1) / 2)) return 1;
     if (ws_ensure(&g_ws.t2, &g_ws.sz_t2, (4 * H * BS + 1) / 2)) return 1;
+    if (ws_ensure(&g_ws.s1, &g_ws.sz_s1, (l->v_pad * BS + 1) / 2)) return 1;
+    if (ws_ensure(&g_ws.s2, &g_ws.sz_s2, (4 * H * BS + 1) / 2)) return 1;
+    if (side_init()) return 1;
   }
   if (d->kind != OB_KIND_EMBED) {
     const int64_t need = BSH;
@@ -764,17 +767,21 @@ static int final_backward_bf16(ob_layer* l, int slot, const float* dout,
   if (ob_ce_bwd_bf16(logits, labs, st + l->o_lse, dout, B, Sq, V, l->v_pad,
                      stream))
     return 1;
-  // dW_lm: transpose dlogits and lnf so the GEMM runs on the glds path;
-  // tile over v_pad rows, store-guard at V (pad rows of dlogits are 0
-  // anyway, but they have no grad slot).
+  // dW_lm on the side stream (overlaps d_lnout + ln backward below):
+  // transpose dlogits and lnf so the GEMM runs on the glds path; tile
+  // over v_pad rows, store-guard at V (pad rows of dlogits are 0 anyway,
+  // but they have no grad slot).
   {
-    __bf16* DLT = (__bf16*)g_ws.t1;
-    __bf16* LNT = (__bf16*)g_ws.t2;
-    if (ob_transpose_bf16(logits, DLT, BS, l->v_pad, stream)) return 1;
-    if (ob_transpose_bf16(lnf, LNT, BS, H, stream)) return 1;
+    OB_HIP(hipEventRecord(g_side.e1, S(stream)));  // dlogits ready
+    OB_HIP(hipStreamWaitEvent(g_side.stream, g_side.e1, 0));
+    void* const side = (void*)g_side.stream;
+    __bf16* DLT = (__bf16*)g_ws.s1;
+    __bf16* LNT = (__bf16*)g_ws.s2;
+    if (ob_transpose_bf16(logits, DLT, BS, l->v_pad, side)) return 1;
+    if (ob_transpose_bf16(lnf, LNT, BS, H, side)) return 1;
     if (ob_gemm_bf16_nt_dispatch(DLT, LNT, g + 2 * H, nullptr, nullptr,
                                  l->v_pad, H, BS, BS, BS, H, 0, 0, 0, 0, 0,
-                                 0, 1, 1, 1.f, 0.f, 2, 2, stream, V))
+                                 0, 1, 1, 1.f, 0.f, 2, 2, side, V))
       return 1;
   }
   // d_lnout: K = v_pad (padded dlogits cols and shadow^T cols are zero)
@@ -785,6 +792,9 @@ static int final_backward_bf16(ob_layer* l, int slot, const float* dout,
   if (ob_layernorm_bwd_bf16(x, p + 0, st + l->o_mean1, st + l->o_rstd1, DLN,
                             din, g + 0, g + H, BS, H, 0, stream))
     return 1;
+  // join: the next backward reuses s1/s2 and reads g on the main stream
+  OB_HIP(hipEventRecord(g_side.sf, g_side.stream));
+  OB_HIP(hipStreamWaitEvent(S(stream), g_side.sf, 0));
   return 0;
 }
 
