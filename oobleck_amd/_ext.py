@@ -65,6 +65,8 @@ def _configure(lib: ctypes.CDLL) -> ctypes.CDLL:
                                       f32, vp]
     lib.ob_transpose_bf16_b.argtypes = [vp, vp, i64, i64, i64, i64, i64,
                                         i64, i64, vp]
+    lib.ob_layernorm_fwd_bf16.argtypes = [vp, vp, vp, vp, vp, vp, i64, i64,
+                                          f32, vp]
     lib.ob_layernorm_bwd_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp,
                                           i64, i64, i32, vp]
     lib.ob_colsum_bf16.argtypes = [vp, vp, i64, i64, vp]

@@ -385,3 +385,26 @@ def test_colsum_bf16_v9(M, N):
     check(get_ext().ob_colsum_bf16(ptr(x), ptr(db), M, N, stream()), "cs")
     torch.cuda.synchronize()
     assert rel_l2(db, x.float().sum(0)) < 2e-2
+
+
+@requires_gpu
+@pytest.mark.parametrize("R,H", [(512, 768), (301, 1024), (77, 520)])
+def test_ln_fwd_bf16_wave(R, H):
+    from oobleck_amd._ext import check, get_ext
+    g = torch.Generator().manual_seed(15)
+    x = (torch.randn(R, H, generator=g)).cuda().bfloat16()
+    w = torch.randn(H, generator=g).cuda().float()
+    b = torch.randn(H, generator=g).cuda().float()
+    y = torch.empty_like(x)
+    mean = torch.empty(R, device=DEV, dtype=torch.float32)
+    rstd = torch.empty(R, device=DEV, dtype=torch.float32)
+    check(get_ext().ob_layernorm_fwd_bf16(
+        ptr(x), ptr(w), ptr(b), ptr(y), ptr(mean), ptr(rstd), R, H, 1e-5,
+        stream()), "lnfwd")
+    torch.cuda.synchronize()
+    xf = x.float()
+    mu = xf.mean(-1, keepdim=True)
+    var = xf.var(-1, unbiased=False, keepdim=True)
+    ref = (xf - mu) * (var + 1e-5).rsqrt() * w + b
+    assert rel_l2(y.float(), ref) < 2e-2
+    torch.testing.assert_close(mean, mu.squeeze(-1), rtol=1e-3, atol=1e-3)
