@@ -361,7 +361,11 @@ extern "C" int ob_gemm_bf16(int transA, int transB, int64_t M, int64_t N,
                                    strideB2, strideC1, strideC2, n1, n2,
                                    alpha, beta, out_kind, sk, stream, M);
     }
-    if (M % 256 == 0 && K % 64 == 0 && K >= 1024) {
+    static const bool no256 = [] {
+      const char* e = getenv("OB_BF16_NO256");
+      return e && e[0] == '1';
+    }();
+    if (!no256 && M % 256 == 0 && K % 64 == 0 && K >= 1024) {
       // pick BN by scheduling-round efficiency (blocks / ceil-to-256):
       // e.g. fc (M=8192,N=3072): BN=256 -> 384 blocks = 1.5 rounds (75%),
       // BN=128 -> 768 blocks = 3 full rounds (100%)
