@@ -86,14 +86,6 @@ int ob_flash_bwd_bf16(const void* qkv, const void* QT, const void* KT,
                       const void* dOT, const void* dO, const void* lse,
                       const void* D, void* dqkv, int64_t B, int64_t Sq,
                       int64_t H, int64_t nh, float scale, void* stream);
-int ob_flash_bwd_dkdv_bf16(const void* qkv, const void* QT, const void* dOT,
-                           const void* dO, const void* lse, const void* D,
-                           void* dqkv, int64_t B, int64_t Sq, int64_t H,
-                           int64_t nh, float scale, void* stream);
-int ob_flash_bwd_dq_bf16(const void* qkv, const void* KT, const void* dO,
-                         const void* lse, const void* D, void* dqkv,
-                         int64_t B, int64_t Sq, int64_t H, int64_t nh,
-                         float scale, void* stream);
 }
 extern "C" int ob_gemm_bf16_nt_8ph(const void* A, const void* B, void* C,
                                    const void* bias, const void* residual,

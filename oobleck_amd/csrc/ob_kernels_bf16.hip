@@ -2234,12 +2234,12 @@ __global__ __launch_bounds__(256, 2) void k_flash_bwd_dq(
   }
 }
 
-extern "C" int ob_flash_bwd_dkdv_bf16(const void* qkv, const void* QT,
-                                      const void* dOT, const void* dO,
-                                      const void* lse, const void* D,
-                                      void* dqkv, int64_t B, int64_t Sq,
-                                      int64_t H, int64_t nh, float scale,
-                                      void* stream) {
+extern "C" int ob_flash_bwd_bf16(const void* qkv, const void* QT,
+                                 const void* KT, const void* dOT,
+                                 const void* dO, const void* lse,
+                                 const void* D, void* dqkv, int64_t B,
+                                 int64_t Sq, int64_t H, int64_t nh,
+                                 float scale, void* stream) {
   if (H / nh != 64) return ob_fail("flash_bwd: head_dim must be 64");
   if (Sq % 128) return ob_fail("flash_bwd: S must be a multiple of 128");
   dim3 grid((unsigned)(Sq / 128), 1, (unsigned)(B * nh));
@@ -2248,36 +2248,12 @@ extern "C" int ob_flash_bwd_dkdv_bf16(const void* qkv, const void* QT,
       (const __bf16*)dO, (const float*)lse, (const float*)D, (__bf16*)dqkv,
       (int)Sq, (int)H, (int)nh, scale);
   OB_LAUNCH_CHECK();
-  return 0;
-}
-
-extern "C" int ob_flash_bwd_dq_bf16(const void* qkv, const void* KT,
-                                    const void* dO, const void* lse,
-                                    const void* D, void* dqkv, int64_t B,
-                                    int64_t Sq, int64_t H, int64_t nh,
-                                    float scale, void* stream) {
-  if (H / nh != 64) return ob_fail("flash_bwd: head_dim must be 64");
-  if (Sq % 128) return ob_fail("flash_bwd: S must be a multiple of 128");
-  dim3 grid((unsigned)(Sq / 128), 1, (unsigned)(B * nh));
   k_flash_bwd_dq<<<grid, 256, 0, S(stream)>>>(
       (const __bf16*)qkv, (const __bf16*)KT, (const __bf16*)dO,
       (const float*)lse, (const float*)D, (__bf16*)dqkv, (int)Sq, (int)H,
       (int)nh, scale);
   OB_LAUNCH_CHECK();
   return 0;
-}
-
-extern "C" int ob_flash_bwd_bf16(const void* qkv, const void* QT,
-                                 const void* KT, const void* dOT,
-                                 const void* dO, const void* lse,
-                                 const void* D, void* dqkv, int64_t B,
-                                 int64_t Sq, int64_t H, int64_t nh,
-                                 float scale, void* stream) {
-  if (ob_flash_bwd_dkdv_bf16(qkv, QT, dOT, dO, lse, D, dqkv, B, Sq, H, nh,
-                             scale, stream))
-    return 1;
-  return ob_flash_bwd_dq_bf16(qkv, KT, dO, lse, D, dqkv, B, Sq, H, nh,
-                              scale, stream);
 }
 
 // ---------------------------------------------------------------------------

@@ -44,10 +44,8 @@ static Workspace g_ws;
 // enqueued host-side in order, so event reuse across calls is safe.
 struct SideSync {
   hipStream_t stream = nullptr;
-  hipStream_t attn = nullptr;  // second lane: flash-bwd dq beside dkdv
   hipEvent_t e1 = nullptr, e2 = nullptr, e3 = nullptr;  // main -> side
   hipEvent_t s2 = nullptr, sf = nullptr;                // side -> main
-  hipEvent_t a1 = nullptr, a2 = nullptr;                // main <-> attn
   bool ready = false;
 };
 static SideSync g_side;
@@ -55,14 +53,11 @@ static SideSync g_side;
 static int side_init() {
   if (g_side.ready) return 0;
   OB_HIP(hipStreamCreateWithFlags(&g_side.stream, hipStreamNonBlocking));
-  OB_HIP(hipStreamCreateWithFlags(&g_side.attn, hipStreamNonBlocking));
   OB_HIP(hipEventCreateWithFlags(&g_side.e1, hipEventDisableTiming));
   OB_HIP(hipEventCreateWithFlags(&g_side.e2, hipEventDisableTiming));
   OB_HIP(hipEventCreateWithFlags(&g_side.e3, hipEventDisableTiming));
   OB_HIP(hipEventCreateWithFlags(&g_side.s2, hipEventDisableTiming));
   OB_HIP(hipEventCreateWithFlags(&g_side.sf, hipEventDisableTiming));
-  OB_HIP(hipEventCreateWithFlags(&g_side.a1, hipEventDisableTiming));
-  OB_HIP(hipEventCreateWithFlags(&g_side.a2, hipEventDisableTiming));
   g_side.ready = true;
   return 0;
 }
@@ -711,18 +706,9 @@ static int block_backward_bf16(ob_layer* l, int slot, const __bf16* dout,
                             stream))
       return 1;
     if (ob_flash_dsum_bf16(am, DATT, Dbuf, B, Sq, H, nh, stream)) return 1;
-    // dq beside dkdv on the attn lane (independent dqkv slices); main
-    // rejoins before the DQKV consumers below
-    OB_HIP(hipEventRecord(g_side.a1, S(stream)));
-    OB_HIP(hipStreamWaitEvent(g_side.attn, g_side.a1, 0));
-    if (ob_flash_bwd_dq_bf16(qkv, KTw, DATT, lseP, Dbuf, DQKV, B, Sq, H, nh,
-                             scale, (void*)g_side.attn))
+    if (ob_flash_bwd_bf16(qkv, QTw, KTw, dOTw, DATT, lseP, Dbuf, DQKV, B,
+                          Sq, H, nh, scale, stream))
       return 1;
-    OB_HIP(hipEventRecord(g_side.a2, g_side.attn));
-    if (ob_flash_bwd_dkdv_bf16(qkv, QTw, dOTw, DATT, lseP, Dbuf, DQKV, B,
-                               Sq, H, nh, scale, stream))
-      return 1;
-    OB_HIP(hipStreamWaitEvent(S(stream), g_side.a2, 0));
   } else {
     if (gemm_bf(0, 1, Sq, Sq, hd, 1.f, DATT, H, Sq * H, hd, qkv + 2 * H,
                 3 * H, Sq * 3 * H, hd, DP, Sq, nh * Sq * Sq, Sq * Sq, B, nh,
