@@ -9,6 +9,7 @@ PKG_DIR = pathlib.Path(__file__).resolve().parent
 SO_PATH = PKG_DIR / "libob_stage.so"
 SOURCES = [PKG_DIR / "csrc" / "ob_kernels.hip",
            PKG_DIR / "csrc" / "ob_kernels_bf16.hip",
+           PKG_DIR / "csrc" / "ob_blaslt.hip",
            PKG_DIR / "csrc" / "ob_layer.hip"]
 HEADERS = [PKG_DIR.parent / "include" / "oobleck_stage.h",
            PKG_DIR / "csrc" / "ob_internal.h"]
@@ -24,7 +25,8 @@ def needs_rebuild() -> bool:
 def build(force: bool = False) -> pathlib.Path:
     if force or needs_rebuild():
         cmd = ["hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC",
-               "-shared"] + [str(s) for s in SOURCES] + ["-o", str(SO_PATH)]
+               "-shared"] + [str(s) for s in SOURCES] + \
+              ["-L/opt/rocm/lib", "-lhipblaslt", "-o", str(SO_PATH)]
         subprocess.run(cmd, check=True, capture_output=True, text=True)
     return SO_PATH
 

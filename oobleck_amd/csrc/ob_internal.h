@@ -87,6 +87,12 @@ int ob_flash_bwd_bf16(const void* qkv, const void* QT, const void* KT,
                       const void* D, void* dqkv, int64_t B, int64_t Sq,
                       int64_t H, int64_t nh, float scale, void* stream);
 }
+// hipBLASLt path for plain GEMMs (ob_blaslt.hip); returns -1 when the
+// heuristic offers no algo (caller falls back to the hand-written path)
+extern "C" int ob_gemm_lt(int tA, int tB, int64_t M, int64_t N, int64_t K,
+                          float alpha, const void* A, int64_t lda,
+                          const void* B, int64_t ldb, float beta, void* C,
+                          int64_t ldc, int c_f32, void* stream);
 extern "C" int ob_gemm_bf16_nt_8ph(const void* A, const void* B, void* C,
                                    const void* bias, const void* residual,
                                    int64_t M, int64_t N, int64_t K,

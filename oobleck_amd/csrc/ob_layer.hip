@@ -488,6 +488,18 @@ static int gemm_bf(int tA, int tB, int64_t M, int64_t N, int64_t K,
 static int dw_bf16_ws(const __bf16* act, int64_t actw, const __bf16* dY,
                       int64_t dyw, int64_t BS, float* gout, int64_t ldc,
                       float* w1, float* w2, void* stream) {
+  // direct TN via hipBLASLt with beta=1 accumulation into the fp32 flat
+  // grads: no transpose materialization, no atomics (measured 694 TF on
+  // the fc dW shape vs 436 for transpose + atomic NT)
+  static const bool no_lt = [] {
+    const char* e = getenv("OB_NO_BLASLT");
+    return e && e[0] == '1';
+  }();
+  if (!no_lt && ldc == dyw) {
+    const int r = ob_gemm_lt(1, 0, actw, dyw, BS, 1.f, act, actw, dY, dyw,
+                             1.f, gout, ldc, 1, stream);
+    if (r >= 0) return r;
+  }
   if ((actw % 128) || (dyw % 128) || (BS % 128))
     return gemm_bf(1, 0, actw, dyw, BS, 1.f, act, actw, 0, 0, dY, dyw, 0, 0,
                    gout, ldc, 0, 0, 1, 1, nullptr, nullptr, 2,
@@ -775,14 +787,27 @@ static int final_backward_bf16(ob_layer* l, int slot, const float* dout,
     OB_HIP(hipEventRecord(g_side.e1, S(stream)));  // dlogits ready
     OB_HIP(hipStreamWaitEvent(g_side.stream, g_side.e1, 0));
     void* const side = (void*)g_side.stream;
-    __bf16* DLT = (__bf16*)g_ws.s1;
-    __bf16* LNT = (__bf16*)g_ws.s2;
-    if (ob_transpose_bf16(logits, DLT, BS, l->v_pad, side)) return 1;
-    if (ob_transpose_bf16(lnf, LNT, BS, H, side)) return 1;
-    if (ob_gemm_bf16_nt_dispatch(DLT, LNT, g + 2 * H, nullptr, nullptr,
-                                 l->v_pad, H, BS, BS, BS, H, 0, 0, 0, 0, 0,
-                                 0, 1, 1, 1.f, 0.f, 2, 2, side, V))
-      return 1;
+    // direct TN via hipBLASLt: M = V with lda = v_pad skips the pad
+    // rows entirely (no store guard, no 826 MB logits transpose)
+    static const bool no_lt = [] {
+      const char* e = getenv("OB_NO_BLASLT");
+      return e && e[0] == '1';
+    }();
+    int r = -1;
+    if (!no_lt)
+      r = ob_gemm_lt(1, 0, V, H, BS, 1.f, logits, l->v_pad, lnf, H, 1.f,
+                     g + 2 * H, H, 1, side);
+    if (r > 0) return 1;
+    if (r < 0) {
+      __bf16* DLT = (__bf16*)g_ws.s1;
+      __bf16* LNT = (__bf16*)g_ws.s2;
+      if (ob_transpose_bf16(logits, DLT, BS, l->v_pad, side)) return 1;
+      if (ob_transpose_bf16(lnf, LNT, BS, H, side)) return 1;
+      if (ob_gemm_bf16_nt_dispatch(DLT, LNT, g + 2 * H, nullptr, nullptr,
+                                   l->v_pad, H, BS, BS, BS, H, 0, 0, 0, 0,
+                                   0, 0, 1, 1, 1.f, 0.f, 2, 2, side, V))
+        return 1;
+    }
   }
   // d_lnout: K = v_pad (padded dlogits cols and shadow^T cols are zero)
   if (gemm_bf(0, 1, BS, H, l->v_pad, 1.f, logits, l->v_pad, 0, 0,
