@@ -33,8 +33,10 @@ struct LtPlan {
 hipblasLtHandle_t g_lt = nullptr;
 void* g_lt_ws = nullptr;
 constexpr size_t kLtWs = 64u << 20;
-// key: tA,tB (row-major semantics), M, N, K, ldc, out f32?, beta!=0?
-using Key = std::tuple<int, int, int64_t, int64_t, int64_t, int64_t, int, int>;
+// key: tA,tB (row-major semantics), M, N, K, ldc, out f32?, beta!=0?,
+// bias epilogue?
+using Key =
+    std::tuple<int, int, int64_t, int64_t, int64_t, int64_t, int, int, int>;
 std::map<Key, LtPlan> g_plans;
 
 int lt_init() {
@@ -51,12 +53,16 @@ int lt_init() {
 // A is [M,K] (or [K,M] if tA), B is [K,N] (or [N,K] if tB), bf16,
 // fp32 accumulate; C bf16 (c_f32 == 0) or fp32 (c_f32 == 1).
 // Returns 0 on success, -1 if no algo (caller falls back), 1 on error.
-extern "C" int ob_gemm_lt(int tA, int tB, int64_t M, int64_t N, int64_t K,
-                          float alpha, const void* A, int64_t lda,
-                          const void* B, int64_t ldb, float beta, void* C,
-                          int64_t ldc, int c_f32, void* stream) {
+// bias (optional, fp32, length N) is applied via the BIAS epilogue —
+// in the column-major swap D = C-bar (N x M), whose rows are our output
+// columns, exactly the broadcast hipBLASLt defines.
+extern "C" int ob_gemm_lt_bias(int tA, int tB, int64_t M, int64_t N,
+                               int64_t K, float alpha, const void* A,
+                               int64_t lda, const void* B, int64_t ldb,
+                               float beta, void* C, int64_t ldc, int c_f32,
+                               const void* bias, void* stream) {
   if (lt_init()) return 1;
-  const Key key{tA, tB, M, N, K, ldc, c_f32, beta != 0.f};
+  const Key key{tA, tB, M, N, K, ldc, c_f32, beta != 0.f, bias != nullptr};
   auto it = g_plans.find(key);
   if (it == g_plans.end()) {
     LtPlan p;
@@ -76,6 +82,16 @@ extern "C" int ob_gemm_lt(int tA, int tB, int64_t M, int64_t N, int64_t K,
                                     &opA, sizeof(opA));
     hipblasLtMatmulDescSetAttribute(p.desc, HIPBLASLT_MATMUL_DESC_TRANSB,
                                     &opB, sizeof(opB));
+    if (bias) {
+      const hipblasLtEpilogue_t ep = HIPBLASLT_EPILOGUE_BIAS;
+      hipblasLtMatmulDescSetAttribute(p.desc, HIPBLASLT_MATMUL_DESC_EPILOGUE,
+                                      &ep, sizeof(ep));
+      const int32_t bt = HIP_R_32F;
+      hipblasLtMatmulDescSetAttribute(
+          p.desc, HIPBLASLT_MATMUL_DESC_BIAS_DATA_TYPE, &bt, sizeof(bt));
+      hipblasLtMatmulDescSetAttribute(
+          p.desc, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &bias, sizeof(bias));
+    }
     hipblasLtMatrixLayoutCreate(&p.la, HIP_R_16BF, a_rows, a_cols, ldb);
     hipblasLtMatrixLayoutCreate(&p.lb, HIP_R_16BF, b_rows, b_cols, lda);
     hipblasLtMatrixLayoutCreate(&p.lc, c_f32 ? HIP_R_32F : HIP_R_16BF, N, M,
@@ -98,10 +114,21 @@ extern "C" int ob_gemm_lt(int tA, int tB, int64_t M, int64_t N, int64_t K,
   }
   const LtPlan& p = it->second;
   if (!p.ok) return -1;
+  if (bias)
+    hipblasLtMatmulDescSetAttribute(
+        p.desc, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &bias, sizeof(bias));
   const hipblasStatus_t st = hipblasLtMatmul(
       g_lt, p.desc, &alpha, B, p.la, A, p.lb, &beta, C, p.lc, C, p.lc,
       &p.algo, g_lt_ws, kLtWs, reinterpret_cast<hipStream_t>(stream));
   if (st != HIPBLAS_STATUS_SUCCESS)
     return ob_fail("hipblasLtMatmul failed (%d)", (int)st);
   return 0;
+}
+
+extern "C" int ob_gemm_lt(int tA, int tB, int64_t M, int64_t N, int64_t K,
+                          float alpha, const void* A, int64_t lda,
+                          const void* B, int64_t ldb, float beta, void* C,
+                          int64_t ldc, int c_f32, void* stream) {
+  return ob_gemm_lt_bias(tA, tB, M, N, K, alpha, A, lda, B, ldb, beta, C,
+                         ldc, c_f32, nullptr, stream);
 }
