@@ -2077,7 +2077,8 @@ __device__ __forceinline__ bf16x8 bf_dance(const float* pv8) {
   return __builtin_bit_cast(bf16x8, fr);
 }
 
-__global__ __launch_bounds__(256, 2) void k_flash_bwd_dkdv(
+template <int MAXB>
+__global__ __launch_bounds__(256, MAXB) void k_flash_bwd_dkdv(
     const __bf16* __restrict__ qkv, const __bf16* __restrict__ QT,
     const __bf16* __restrict__ dOT, const __bf16* __restrict__ dO,
     const float* __restrict__ lse, const float* __restrict__ D,
@@ -2135,25 +2136,27 @@ __global__ __launch_bounds__(256, 2) void k_flash_bwd_dkdv(
       dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(df, vf[s], dpacc, 0, 0,
                                                       0);
     }
-    // P and dS per element: rows are q
-    float pv[16], dsv[16];
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int qoff = (r & 3) + 8 * (r >> 2) + 4 * kh;
-      const int q = q0 + qoff;
-      const bool ok = q >= mykv;
-      const float lse_q = __shfl(lse_t, qoff, 64);
-      const float d_q = __shfl(d_t, qoff, 64);
-      const float p = ok ? __expf(sacc[r] * scale - lse_q) : 0.f;
-      pv[r] = p;
-      dsv[r] = ok ? p * (dpacc[r] - d_q) : 0.f;
-    }
+    // P and dS per element (one 8-run at a time: halves the live pv/dsv
+    // registers so the kernel fits 3 waves/SIMD); rows are q
     // dV += P^T dO   (A = P^T frag: i = kv = lane col; k = q-run)
     // dK += scale * dS^T Q
 #pragma unroll
     for (int t = 0; t < 2; ++t) {
-      const bf16x8 pa = bf_dance(pv + t * 8);
-      const bf16x8 da = bf_dance(dsv + t * 8);
+      float pv[8], dsv[8];
+#pragma unroll
+      for (int r8 = 0; r8 < 8; ++r8) {
+        const int r = t * 8 + r8;
+        const int qoff = (r & 3) + 8 * (r >> 2) + 4 * kh;
+        const int q = q0 + qoff;
+        const bool ok = q >= mykv;
+        const float lse_q = __shfl(lse_t, qoff, 64);
+        const float d_q = __shfl(d_t, qoff, 64);
+        const float p = ok ? __expf(sacc[r] * scale - lse_q) : 0.f;
+        pv[r8] = p;
+        dsv[r8] = ok ? p * (dpacc[r] - d_q) : 0.f;
+      }
+      const bf16x8 pa = bf_dance(pv);
+      const bf16x8 da = bf_dance(dsv);
       const bf16x8 do0 = *reinterpret_cast<const bf16x8*>(
           dOTp + (int64_t)il * Sq + q0 + t * 16 + kh * 8);
       const bf16x8 do1 = *reinterpret_cast<const bf16x8*>(
@@ -2268,7 +2271,21 @@ extern "C" int ob_flash_bwd_bf16(const void* qkv, const void* QT,
   if (H / nh != 64) return ob_fail("flash_bwd: head_dim must be 64");
   if (Sq % 128) return ob_fail("flash_bwd: S must be a multiple of 128");
   dim3 grid((unsigned)(Sq / 128), 1, (unsigned)(B * nh));
-  k_flash_bwd_dkdv<<<grid, 256, 0, S(stream)>>>(
+  // 3 waves/SIMD (168 VGPR + 52 B scratch) measured 4% faster at step
+  // level than the spill-free 2-wave build; OB_DKDV3=0 reverts.
+  static const bool occ3 = [] {
+    const char* e = getenv("OB_DKDV3");
+    return !(e && e[0] == '0');
+  }();
+  if (occ3) {
+    k_flash_bwd_dkdv<3><<<grid, 256, 0, S(stream)>>>(
+        (const __bf16*)qkv, (const __bf16*)QT, (const __bf16*)dOT,
+        (const __bf16*)dO, (const float*)lse, (const float*)D,
+        (__bf16*)dqkv, (int)Sq, (int)H, (int)nh, scale);
+    OB_LAUNCH_CHECK();
+    return 0;
+  }
+  k_flash_bwd_dkdv<2><<<grid, 256, 0, S(stream)>>>(
       (const __bf16*)qkv, (const __bf16*)QT, (const __bf16*)dOT,
       (const __bf16*)dO, (const float*)lse, (const float*)D, (__bf16*)dqkv,
       (int)Sq, (int)H, (int)nh, scale);
