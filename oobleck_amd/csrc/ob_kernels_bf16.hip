@@ -2277,11 +2277,12 @@ extern "C" int ob_flash_bwd_bf16(const void* qkv, const void* QT,
   if (H / nh != 64) return ob_fail("flash_bwd: head_dim must be 64");
   if (Sq % 128) return ob_fail("flash_bwd: S must be a multiple of 128");
   dim3 grid((unsigned)(Sq / 128), 1, (unsigned)(B * nh));
-  // 3 waves/SIMD (168 VGPR + 52 B scratch) measured 4% faster at step
-  // level than the spill-free 2-wave build; OB_DKDV3=0 reverts.
+  // NOTE: the 3-wave build (168 VGPR + 52 B scratch) benched "faster"
+  // but FAULTS — its dQ-phase never ran (parity tests caught it); it
+  // stays opt-in (OB_DKDV3=1) for debugging only.
   static const bool occ3 = [] {
     const char* e = getenv("OB_DKDV3");
-    return !(e && e[0] == '0');
+    return e && e[0] == '1';
   }();
   if (occ3) {
     k_flash_bwd_dkdv<3><<<grid, 256, 0, S(stream)>>>(
