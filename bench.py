@@ -126,18 +126,24 @@ def measure_roofline(args, device):
     achieved_tf = flops / avg_s / 1e12
     peak = BF16_MFMA_PEAK_TF if bf16 else F32_MFMA_PEAK_TF
     traffic = None
-    pmc_file = os.path.join(
-        os.path.dirname(os.path.abspath(__file__)), "profiles",
-        "pmc_gemm_fc_bf16.json" if bf16 else "pmc_gemm_fc.json")
-    if os.path.exists(pmc_file):
-        with open(pmc_file) as f:
-            traffic = json.load(f).get("hbm_bytes_per_launch")
+    # bf16 plain GEMMs dispatch to hipBLASLt (ob_blaslt.hip) — the stale
+    # PMC traffic file measured the hand-written kernel, so it only
+    # applies to the f32 leg.
+    if not bf16:
+        pmc_file = os.path.join(
+            os.path.dirname(os.path.abspath(__file__)), "profiles",
+            "pmc_gemm_fc.json")
+        if os.path.exists(pmc_file):
+            with open(pmc_file) as f:
+                traffic = json.load(f).get("hbm_bytes_per_launch")
     return {
         "bound": "mfma", "achieved": round(achieved_tf, 2),
         "peak": peak, "unit": "TFLOP/s",
         "frac": round(achieved_tf / peak, 4),
         "traffic": traffic,
-        "kernel": ("k_gemm_bf16 (MLP fc, M=8192 N=3072 K=768, bf16 MFMA)"
+        "kernel": ("MLP fc GEMM M=8192 N=3072 K=768 bf16 (production "
+                   "dispatch: hipBLASLt Cijk; hand-written k_gemm_bf16 "
+                   "carries the fused/fallback cases)"
                    if bf16 else
                    "k_gemm_f32 (MLP fc, M=8192 N=3072 K=768, fp32 MFMA)"),
         "avg_launch_ms": round(avg_s * 1e3, 4),
