@@ -31,8 +31,19 @@ struct LtPlan {
 };
 
 hipblasLtHandle_t g_lt = nullptr;
-void* g_lt_ws = nullptr;
 constexpr size_t kLtWs = 64u << 20;
+// one workspace per stream: concurrent matmuls on different streams
+// (fwd/bwd/side) must not share scratch
+std::map<void*, void*> g_lt_ws_by_stream;
+
+void* lt_ws(void* stream) {
+  auto it = g_lt_ws_by_stream.find(stream);
+  if (it != g_lt_ws_by_stream.end()) return it->second;
+  void* ws = nullptr;
+  if (hipMalloc(&ws, kLtWs) != hipSuccess) return nullptr;
+  g_lt_ws_by_stream.emplace(stream, ws);
+  return ws;
+}
 // key: tA,tB (row-major semantics), M, N, K, ldc, out f32?, beta!=0?,
 // bias epilogue?
 using Key =
@@ -43,7 +54,6 @@ int lt_init() {
   if (g_lt) return 0;
   if (hipblasLtCreate(&g_lt) != HIPBLAS_STATUS_SUCCESS)
     return ob_fail("hipblasLtCreate failed");
-  OB_HIP(hipMalloc(&g_lt_ws, kLtWs));
   return 0;
 }
 
@@ -117,9 +127,11 @@ extern "C" int ob_gemm_lt_bias(int tA, int tB, int64_t M, int64_t N,
   if (bias)
     hipblasLtMatmulDescSetAttribute(
         p.desc, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &bias, sizeof(bias));
+  void* ws = lt_ws(stream);
+  if (!ws) return ob_fail("hipBLASLt workspace alloc failed");
   const hipblasStatus_t st = hipblasLtMatmul(
       g_lt, p.desc, &alpha, B, p.la, A, p.lb, &beta, C, p.lc, C, p.lc,
-      &p.algo, g_lt_ws, kLtWs, reinterpret_cast<hipStream_t>(stream));
+      &p.algo, ws, kLtWs, reinterpret_cast<hipStream_t>(stream));
   if (st != HIPBLAS_STATUS_SUCCESS)
     return ob_fail("hipblasLtMatmul failed (%d)", (int)st);
   return 0;

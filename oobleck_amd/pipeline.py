@@ -69,11 +69,29 @@ class PipelineExecution:
         import torch as _torch
         self._act_dtype = getattr(layers[0], "act_dtype", _torch.float32) \
             if layers else _torch.float32
-        # lazily-allocated working buffers
-        self._tmp: list[torch.Tensor | None] = [None, None]
+        # lazily-allocated working buffers (indices 0/1: forward-pass
+        # inter-layer temps, 2/3: backward-pass temps — disjoint so the
+        # dual-stream overlap below can run a forward beside a backward)
+        self._tmp: list[torch.Tensor | None] = [None, None, None, None]
         self._out_bufs: dict[int, torch.Tensor] = {}
         self._loss_bufs: dict[int, torch.Tensor] = {}
         self._din_bufs: dict[int, torch.Tensor] = {}
+        # single-stage (pp1) GPU overlap: forward of microbatch m+1 runs
+        # on its own HIP stream beside backward of microbatch m (the 1F1B
+        # schedule at stages=1 alternates F/B).  Slot reuse and the
+        # F(m)->B(m) edge are event-fenced; the optimizer/all-reduce epoch
+        # re-joins both streams (train()).  Disabled under FSDP (its
+        # collectives must keep a single well-ordered stream).
+        self._overlap = (
+            pipeline.device.type == "cuda"
+            and pipeline.is_first_stage() and pipeline.is_last_stage()
+            and not any(getattr(l, "_sharded", None) is not None
+                        for l in layers))
+        if self._overlap:
+            self._s_fwd = torch.cuda.Stream()
+            self._s_bwd = torch.cuda.Stream()
+            self._ev_fwd: dict[int, torch.cuda.Event] = {}
+            self._ev_bwd: dict[int, torch.cuda.Event] = {}
 
     @property
     def pipeline(self) -> "OobleckPipeline":
@@ -102,11 +120,27 @@ class PipelineExecution:
         if self.pipeline.is_first_stage():
             batch = next(self._data_iterator)
             dev = self.pipeline.device
-            ids = batch["input_ids"].to(dev, non_blocking=True)
-            labels = batch["labels"].to(dev, non_blocking=True)
+            import contextlib
+            ctx = (torch.cuda.stream(self._s_fwd) if self._overlap
+                   else contextlib.nullcontext())
+            with ctx:
+                ids = batch["input_ids"].to(dev, non_blocking=True)
+                labels = batch["labels"].to(dev, non_blocking=True)
             self.pipeline.pipe_buffers["inputs"][buffer_id] = (ids, labels)
 
     def forward_pass(self, buffer_id: int) -> None:
+        if self._overlap:
+            with torch.cuda.stream(self._s_fwd):
+                ev = self._ev_bwd.get(buffer_id)
+                if ev is not None:  # slot reuse: wait its previous backward
+                    self._s_fwd.wait_event(ev)
+                self._forward_impl(buffer_id)
+                fev = self._ev_fwd.setdefault(buffer_id, torch.cuda.Event())
+                fev.record(self._s_fwd)
+        else:
+            self._forward_impl(buffer_id)
+
+    def _forward_impl(self, buffer_id: int) -> None:
         x, labels = self.pipeline.pipe_buffers["inputs"][buffer_id]
         batch = x.shape[0]
         n = len(self._layers)
@@ -142,6 +176,16 @@ class PipelineExecution:
             self.pipeline.pipe_buffers["outputs"][buffer_id] = (x, labels)
 
     def backward_pass(self, buffer_id: int) -> None:
+        if self._overlap:
+            with torch.cuda.stream(self._s_bwd):
+                self._s_bwd.wait_event(self._ev_fwd[buffer_id])
+                self._backward_impl(buffer_id)
+                bev = self._ev_bwd.setdefault(buffer_id, torch.cuda.Event())
+                bev.record(self._s_bwd)
+        else:
+            self._backward_impl(buffer_id)
+
+    def _backward_impl(self, buffer_id: int) -> None:
         if self.pipeline.is_last_stage():
             dout = None  # final layer seeds dloss = 1.0 (layer.py:250-253)
         else:
@@ -162,7 +206,7 @@ class PipelineExecution:
                         self._din_bufs[buffer_id] = din
                     din = din[:batch]
             else:
-                din = self._get_tmp(i % 2, batch)
+                din = self._get_tmp(2 + i % 2, batch)
             layer.backward_slot(buffer_id, dout, din)
             dout = din
         # free forward output (reference pipeline.py:235-239)
@@ -363,6 +407,10 @@ class OobleckPipeline:
                 if handler is None:
                     raise RuntimeError(f"unknown instruction {cmd!r}")
                 handler(**cmd.kwargs)
+        if getattr(self.execution, "_overlap", False):
+            cur = torch.cuda.current_stream()
+            cur.wait_stream(self.execution._s_fwd)
+            cur.wait_stream(self.execution._s_bwd)
         for name, bufs in self.pipe_buffers.items():
             self.pipe_buffers[name] = [None] * len(bufs)
 
