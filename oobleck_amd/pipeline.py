@@ -82,8 +82,13 @@ class PipelineExecution:
         # F(m)->B(m) edge are event-fenced; the optimizer/all-reduce epoch
         # re-joins both streams (train()).  Disabled under FSDP (its
         # collectives must keep a single well-ordered stream).
+        # DISABLED pending round-2 debugging: the overlap passed the full
+        # GPU parity suite but stalled the 16-microbatch bench run;
+        # OB_PP1_OVERLAP=1 re-enables for investigation.
+        import os
         self._overlap = (
-            pipeline.device.type == "cuda"
+            os.environ.get("OB_PP1_OVERLAP", "0") == "1"
+            and pipeline.device.type == "cuda"
             and pipeline.is_first_stage() and pipeline.is_last_stage()
             and not any(getattr(l, "_sharded", None) is not None
                         for l in layers))
