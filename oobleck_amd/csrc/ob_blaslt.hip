@@ -45,9 +45,9 @@ void* lt_ws(void* stream) {
   return ws;
 }
 // key: tA,tB (row-major semantics), M, N, K, ldc, out f32?, beta!=0?,
-// bias epilogue?
-using Key =
-    std::tuple<int, int, int64_t, int64_t, int64_t, int64_t, int, int, int>;
+// bias epilogue?, A/B f32?
+using Key = std::tuple<int, int, int64_t, int64_t, int64_t, int64_t, int,
+                       int, int, int>;
 std::map<Key, LtPlan> g_plans;
 
 int lt_init() {
@@ -66,13 +66,15 @@ int lt_init() {
 // bias (optional, fp32, length N) is applied via the BIAS epilogue —
 // in the column-major swap D = C-bar (N x M), whose rows are our output
 // columns, exactly the broadcast hipBLASLt defines.
-extern "C" int ob_gemm_lt_bias(int tA, int tB, int64_t M, int64_t N,
-                               int64_t K, float alpha, const void* A,
-                               int64_t lda, const void* B, int64_t ldb,
-                               float beta, void* C, int64_t ldc, int c_f32,
-                               const void* bias, void* stream) {
+static int lt_matmul(int tA, int tB, int64_t M, int64_t N, int64_t K,
+                     float alpha, const void* A, int64_t lda, const void* B,
+                     int64_t ldb, float beta, void* C, int64_t ldc,
+                     int c_f32, const void* bias, int ab_f32, void* stream) {
   if (lt_init()) return 1;
-  const Key key{tA, tB, M, N, K, ldc, c_f32, beta != 0.f, bias != nullptr};
+  const Key key{tA,     tB,           M,
+                N,      K,            ldc,
+                c_f32,  beta != 0.f,  bias != nullptr,
+                ab_f32};
   auto it = g_plans.find(key);
   if (it == g_plans.end()) {
     LtPlan p;
@@ -102,8 +104,9 @@ extern "C" int ob_gemm_lt_bias(int tA, int tB, int64_t M, int64_t N,
       hipblasLtMatmulDescSetAttribute(
           p.desc, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &bias, sizeof(bias));
     }
-    hipblasLtMatrixLayoutCreate(&p.la, HIP_R_16BF, a_rows, a_cols, ldb);
-    hipblasLtMatrixLayoutCreate(&p.lb, HIP_R_16BF, b_rows, b_cols, lda);
+    const hipDataType abt = ab_f32 ? HIP_R_32F : HIP_R_16BF;
+    hipblasLtMatrixLayoutCreate(&p.la, abt, a_rows, a_cols, ldb);
+    hipblasLtMatrixLayoutCreate(&p.lb, abt, b_rows, b_cols, lda);
     hipblasLtMatrixLayoutCreate(&p.lc, c_f32 ? HIP_R_32F : HIP_R_16BF, N, M,
                                 ldc);
     hipblasLtMatmulPreference_t pref;
@@ -141,6 +144,25 @@ extern "C" int ob_gemm_lt(int tA, int tB, int64_t M, int64_t N, int64_t K,
                           float alpha, const void* A, int64_t lda,
                           const void* B, int64_t ldb, float beta, void* C,
                           int64_t ldc, int c_f32, void* stream) {
-  return ob_gemm_lt_bias(tA, tB, M, N, K, alpha, A, lda, B, ldb, beta, C,
-                         ldc, c_f32, nullptr, stream);
+  return lt_matmul(tA, tB, M, N, K, alpha, A, lda, B, ldb, beta, C, ldc,
+                   c_f32, nullptr, 0, stream);
+}
+
+extern "C" int ob_gemm_lt_bias(int tA, int tB, int64_t M, int64_t N,
+                               int64_t K, float alpha, const void* A,
+                               int64_t lda, const void* B, int64_t ldb,
+                               float beta, void* C, int64_t ldc, int c_f32,
+                               const void* bias, void* stream) {
+  return lt_matmul(tA, tB, M, N, K, alpha, A, lda, B, ldb, beta, C, ldc,
+                   c_f32, bias, 0, stream);
+}
+
+// fp32 operands, fp32 accumulate/out (the reference-dtype leg)
+extern "C" int ob_gemm_lt_f32(int tA, int tB, int64_t M, int64_t N,
+                              int64_t K, float alpha, const void* A,
+                              int64_t lda, const void* B, int64_t ldb,
+                              float beta, void* C, int64_t ldc,
+                              void* stream) {
+  return lt_matmul(tA, tB, M, N, K, alpha, A, lda, B, ldb, beta, C, ldc, 1,
+                   nullptr, 1, stream);
 }

@@ -98,6 +98,11 @@ extern "C" int ob_gemm_lt_bias(int tA, int tB, int64_t M, int64_t N,
                                int64_t lda, const void* B, int64_t ldb,
                                float beta, void* C, int64_t ldc, int c_f32,
                                const void* bias, void* stream);
+extern "C" int ob_gemm_lt_f32(int tA, int tB, int64_t M, int64_t N,
+                              int64_t K, float alpha, const void* A,
+                              int64_t lda, const void* B, int64_t ldb,
+                              float beta, void* C, int64_t ldc,
+                              void* stream);
 extern "C" int ob_gemm_bf16_nt_8ph(const void* A, const void* B, void* C,
                                    const void* bias, const void* residual,
                                    int64_t M, int64_t N, int64_t K,

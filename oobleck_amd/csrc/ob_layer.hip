@@ -338,6 +338,18 @@ static int gemm(int tA, int tB, int64_t M, int64_t N, int64_t K, float alpha,
                 float beta, float* C, int64_t ldc, int64_t sC1, int64_t sC2,
                 int64_t n1, int64_t n2, const float* bias, const float* R,
                 int atomic, int splitk, void* stream) {
+  // plain (unbatched, no fused epilogue) fp32 GEMMs to hipBLASLt —
+  // covers the dW family too (it accumulates through beta=1)
+  static const bool no_lt = [] {
+    const char* e = getenv("OB_NO_BLASLT");
+    return e && e[0] == '1';
+  }();
+  if (!no_lt && !bias && !R && !atomic && splitk <= 1 && n1 == 1 &&
+      n2 == 1 && M * N >= 512 * 512) {
+    const int r = ob_gemm_lt_f32(tA, tB, M, N, K, alpha, A, lda, B, ldb,
+                                 beta, C, ldc, stream);
+    if (r >= 0) return r;
+  }
   return ob_gemm_f32(tA, tB, M, N, K, alpha, A, lda, sA1, sA2, B, ldb, sB1,
                      sB2, beta, C, ldc, sC1, sC2, n1, n2, bias, R, atomic,
                      splitk, stream);
@@ -360,6 +372,15 @@ static int gemm_dx_splitk(int tB, int64_t M, int64_t N, int64_t K,
                           float alpha, const float* A, int64_t lda,
                           const float* B, int64_t ldb, float* C, int64_t ldc,
                           void* stream) {
+  static const bool no_lt = [] {
+    const char* e = getenv("OB_NO_BLASLT");
+    return e && e[0] == '1';
+  }();
+  if (!no_lt) {
+    const int r = ob_gemm_lt_f32(0, tB, M, N, K, alpha, A, lda, B, ldb, 0.f,
+                                 C, ldc, stream);
+    if (r >= 0) return r;
+  }
   const int64_t tiles = ((M + 127) / 128) * ((N + 127) / 128);
   int sk = 1;
   while (sk < 16 && tiles * sk < 1024 && (K / (sk * 2)) >= 512) sk *= 2;
