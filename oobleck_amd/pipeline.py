@@ -406,12 +406,21 @@ class OobleckPipeline:
             SendGrad: self.communication.send_gradients,
             RecvGrad: self.communication.recv_gradients,
         }
+        import os
+        import sys
+        trace = os.environ.get("OB_TRACE_SCHED", "0") == "1"
         for step_cmds in self.train_schedule:
             for cmd in step_cmds:
                 handler = instruction_map.get(type(cmd))
                 if handler is None:
                     raise RuntimeError(f"unknown instruction {cmd!r}")
+                if trace:
+                    print(f"[sched] {type(cmd).__name__} buf={cmd.buffer_id}",
+                          file=sys.stderr, flush=True)
                 handler(**cmd.kwargs)
+                if trace:
+                    print(f"[sched] done {type(cmd).__name__}",
+                          file=sys.stderr, flush=True)
         if getattr(self.execution, "_overlap", False):
             cur = torch.cuda.current_stream()
             cur.wait_stream(self.execution._s_fwd)
