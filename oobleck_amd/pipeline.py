@@ -82,9 +82,13 @@ class PipelineExecution:
         # F(m)->B(m) edge are event-fenced; the optimizer/all-reduce epoch
         # re-joins both streams (train()).  Disabled under FSDP (its
         # collectives must keep a single well-ordered stream).
-        # DISABLED pending round-2 debugging: the overlap passed the full
-        # GPU parity suite but stalled the 16-microbatch bench run;
-        # OB_PP1_OVERLAP=1 re-enables for investigation.
+        # DISABLED pending round-2 debugging: the overlap passes the full
+        # GPU parity suite and completes at 4 microbatches/step, but
+        # hangs (or degrades beyond a 180 s timeout) at the
+        # 16-microbatch bench scale.  Suspects: the pageable H2D copy on
+        # a non-default stream blocking the host per microbatch, and the
+        # slot-reuse event fences at depth.  OB_PP1_OVERLAP=1 +
+        # OB_TRACE_SCHED=1 re-enable for investigation.
         import os
         self._overlap = (
             os.environ.get("OB_PP1_OVERLAP", "0") == "1"
