@@ -534,13 +534,6 @@ static int dw_bf16_ws(const __bf16* act, int64_t actw, const __bf16* dY,
                  pick_splitk(actw, dyw, BS), stream);
 }
 
-static int dw_bf16(const __bf16* act, int64_t actw, const __bf16* dY,
-                   int64_t dyw, int64_t BS, float* gout, int64_t ldc,
-                   void* stream) {
-  return dw_bf16_ws(act, actw, dY, dyw, BS, gout, ldc, g_ws.t1, g_ws.t2,
-                    stream);
-}
-
 static bool use_flash(const ob_layer* l) { return l->flash; }
 
 static int block_forward_bf16(ob_layer* l, int slot, const __bf16* in,
