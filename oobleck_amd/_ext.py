@@ -70,6 +70,8 @@ def _configure(lib: ctypes.CDLL) -> ctypes.CDLL:
     lib.ob_layernorm_bwd_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp,
                                           i64, i64, i32, vp]
     lib.ob_colsum_bf16.argtypes = [vp, vp, i64, i64, vp]
+    lib.ob_gemm_lt.argtypes = [i32, i32, i64, i64, i64, f32, vp, i64, vp,
+                               i64, f32, vp, i64, i32, vp]
     lib.ob_gemm_bf16_nt_8ph.argtypes = [vp, vp, vp, vp, vp, i64, i64, i64,
                                         i64, i64, i64, i64, i64, i64, i64,
                                         i64, i64, i64, i64, f32, f32, i32,

@@ -408,3 +408,54 @@ def test_ln_fwd_bf16_wave(R, H):
     ref = (xf - mu) * (var + 1e-5).rsqrt() * w + b
     assert rel_l2(y.float(), ref) < 2e-2
     torch.testing.assert_close(mean, mu.squeeze(-1), rtol=1e-3, atol=1e-3)
+
+
+@requires_gpu
+def test_lt_plain_nt_parity():
+    """The hipBLASLt plain-GEMM route (M*N >= 512^2, no epilogue) vs
+    torch fp32 on bf16-rounded inputs (ob_blaslt.hip column-major swap)."""
+    M, N, K = 1024, 768, 320
+    A = rb(M, K, seed=21)
+    B = rb(N, K, seed=22)
+    C = torch.empty(M, N, device=DEV, dtype=torch.bfloat16)
+    gemm_bf16(A, B, C, transB=1, M=M, N=N, K=K, lda=K, ldb=K, ldc=N)
+    torch.cuda.synchronize()
+    ref = A.float() @ B.float().T
+    assert rel_l2(C.float(), ref) < 2e-2
+
+
+@requires_gpu
+def test_lt_bias_epilogue_parity():
+    from oobleck_amd._ext import check, get_ext
+    M, N, K = 1024, 1024, 256
+    A = rb(M, K, seed=23)
+    B = rb(N, K, seed=24)
+    bias = torch.randn(N, device=DEV, dtype=torch.float32)
+    C = torch.empty(M, N, device=DEV, dtype=torch.bfloat16)
+    gemm_bf16(A, B, C, transB=1, M=M, N=N, K=K, lda=K, ldb=K, ldc=N,
+              bias=bias)
+    torch.cuda.synchronize()
+    ref = A.float() @ B.float().T + bias
+    assert rel_l2(C.float(), ref) < 2e-2
+
+
+@requires_gpu
+def test_lt_tn_beta1_accumulate():
+    """Direct-TN dW route: C_f32 += A^T B with beta=1 (the weight-grad
+    accumulation path, ob_gemm_lt c_f32=1)."""
+    import ctypes
+    from oobleck_amd._ext import check, get_ext
+    BS, W1, W2 = 768, 512, 640
+    A = rb(BS, W1, seed=25)
+    B = rb(BS, W2, seed=26)
+    C = torch.randn(W1, W2, device=DEV, dtype=torch.float32)
+    C0 = C.clone()
+    r = get_ext().ob_gemm_lt(1, 0, W1, W2, BS, ctypes.c_float(1.0), ptr(A),
+                             W1, ptr(B), W2, ctypes.c_float(1.0), ptr(C),
+                             W2, 1, stream())
+    torch.cuda.synchronize()
+    if r == -1:
+        pytest.skip("hipBLASLt offered no algo for this shape")
+    assert r == 0
+    ref = C0 + A.float().T @ B.float()
+    assert rel_l2(C, ref) < 2e-2
