@@ -257,3 +257,73 @@ def _run_fsdp(rank: int, world: int, tmp: str):
                               "fsdp_full_shard"])
 def test_multiprocess(target, tmp_path):
     mp.spawn(target, args=(2, str(tmp_path)), nprocs=2, join=True)
+
+
+# ---------------------------------------------------------------------------
+# the 8-GPU topology (4 stages x 2 DP replicas) on gloo world 8 — the exact
+# shape bench.py builds for the driver's N=8 scaling run (TOPOLOGY[8])
+# ---------------------------------------------------------------------------
+
+def _run_4x2(rank: int, world: int, tmp: str):
+    _setup(rank, world, tmp)
+    from oobleck_amd.engine import (DataParallelEngine, even_stage_split,
+                                    make_rank_grid)
+    from oobleck_amd.pipeline import OobleckPipeline
+    from tests.oracle_layer import NoOpOptimizer, OracleLayer
+
+    mc, oc, tc0 = _configs()
+    from oobleck_amd.config import TrainingConfig
+    stages, replicas = 4, 2
+    tc = TrainingConfig(microbatch_size=B,
+                        global_microbatch_size=B * MB * replicas, seq_len=S)
+    flats = _flats(oc)
+    stage_layers = even_stage_split(mc, stages)
+
+    def loader_for(pid):
+        class Loader:
+            def __iter__(self):
+                return iter({"input_ids": i, "labels": l}
+                            for i, l in _batches(oc, MB, seed=7 + pid))
+        return Loader()
+
+    pipelines, my_pipeline = [], None
+    for pid in range(replicas):
+        ranks = [pid * stages + s for s in range(stages)]
+        grid = make_rank_grid(oc.n_layers_total, stage_layers,
+                              [[r] for r in ranks])
+        pipe = OobleckPipeline(pid, grid, mc, tc, loader_for(pid), MB,
+                               torch.device("cpu"))
+        pipe.initialize_distributed_fsdp()
+        pipe.initialize_distributed_pipeline()
+        pipelines.append(pipe)
+    for p in pipelines:
+        if p.my_pipeline:
+            p.initialize_execution(
+                layer_factory=lambda lid, pg, n_slots: OracleLayer(
+                    lid, oc, flats[lid]),
+                optimizer_factory=lambda layers: (NoOpOptimizer(layers),
+                                                  None))
+            my_pipeline = p
+    dp = DataParallelEngine(pipelines)
+    assert my_pipeline is not None
+    my_pipeline.train()
+    dp.do_allreduce(my_pipeline)
+
+    # reference: grads summed over BOTH replicas' microbatches (DP SUM)
+    losses0, g0 = _reference_grads(oc, flats, _batches(oc, MB, seed=7))
+    losses1, g1 = _reference_grads(oc, flats, _batches(oc, MB, seed=8))
+    for layer in my_pipeline.execution._layers:
+        ref = g0[layer.layer_id] + g1[layer.layer_id]
+        torch.testing.assert_close(layer.flat_grad, ref, rtol=1e-4,
+                                   atol=1e-5)
+    if my_pipeline.is_last_stage():
+        ref_losses = losses0 if rank < stages else losses1
+        total_ref = sum(l.item() for l in ref_losses)
+        got = my_pipeline.execution.total_loss.item()
+        assert abs(got - total_ref) < 1e-4 * abs(total_ref), (got, total_ref)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_multiprocess_4x2(tmp_path):
+    mp.spawn(_run_4x2, args=(8, str(tmp_path)), nprocs=8, join=True)
