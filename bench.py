@@ -30,14 +30,14 @@ BF16_MFMA_PEAK_TF = 2500.0  # gfx950 bf16 dense MFMA peak (NOT the 2:1-sparse 5 
 
 
 def build_world(args, device):
-    from oobleck_amd.config import GPT2_SMALL, TrainingConfig
+    from oobleck_amd.config import GPT2_SMALL, GPT2_XL, TrainingConfig
     from oobleck_amd.engine import (DataParallelEngine, even_stage_split,
                                     make_rank_grid)
     from oobleck_amd.layer import Layer
     from oobleck_amd.optimizer import FusedAdamW, WarmupLR
     from oobleck_amd.pipeline import OobleckPipeline, SyntheticDataLoader
 
-    mc = GPT2_SMALL
+    mc = GPT2_XL if args.model == "gpt2-xl" else GPT2_SMALL
     tc = TrainingConfig(seq_len=args.seq_len,
                         microbatch_size=args.microbatch,
                         global_microbatch_size=args.global_batch)
@@ -182,6 +182,11 @@ def measure_cpu_baseline():
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--model", choices=["gpt2", "gpt2-xl"], default="gpt2",
+                    help="gpt2 = gpt2.yaml GPT-2 small (the headline "
+                         "config); gpt2-xl = the dims of examples/"
+                         "gpt3.yaml (1600x48, head_dim 64 — BASELINE "
+                         "config[3]'s workload)")
     ap.add_argument("--steps", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--seq-len", type=int, default=1024)
@@ -254,9 +259,11 @@ def main():
             "dtype": args.dtype,  # f32 = the reference compute dtype
             "data": "synthetic",
             "config": {
-                "workload": "gpt2.yaml GPT-2-small 1F1B, global microbatch "
-                            f"128x1024 tokens, {args.dtype}",
-                "model": "gpt2",
+                "workload": (
+                    f"{'gpt3.yaml GPT-2-XL' if args.model == 'gpt2-xl' else 'gpt2.yaml GPT-2-small'}"
+                    f" 1F1B, global microbatch "
+                    f"{args.global_batch}x{args.seq_len} tokens, {args.dtype}"),
+                "model": args.model,
                 "global_batch": args.global_batch,
                 "seq_len": args.seq_len,
                 "parallelism": f"pp{stages}dp{replicas}",
