@@ -184,6 +184,19 @@ int ob_f32_to_bf16_t(const void* x, void* y, int64_t rows, int64_t cols,
                      void* stream);  /* y[c][r] = x[r][c] */
 int ob_bf16_to_f32(const void* x, void* y, int64_t n, void* stream);
 
+/* ---- in-step profiling (measurement only; off by default) ---------------
+ * When enabled, each wrapped launch region inside ob_layer_forward/backward
+ * and ob_adamw_step is bracketed by HIP events on its launch stream and
+ * accumulated per family, so the bench can report the production dispatch's
+ * in-step per-launch time (roofline) and a per-family step-time split.
+ * Families (ob_profile_read's `fam`): 0 fc_fwd_gemm (the roofline kernel),
+ * 1 gemm_fwd_other, 2 gemm_dx, 3 gemm_dw(side stream), 4 flash_fwd,
+ * 5 flash_bwd, 6 attn_matmuls(non-flash), 7 layernorm, 8 cross_entropy,
+ * 9 elementwise(gelu/colsum/embed), 10 adamw. */
+void ob_profile_enable(int on);
+void ob_profile_reset(void);
+int ob_profile_read(int fam, double* total_ms, long long* count);
+
 const char* ob_last_error(void);
 
 /* Build stamp: returns the gfx arch this library was compiled for. */

@@ -76,6 +76,12 @@ def _configure(lib: ctypes.CDLL) -> ctypes.CDLL:
                                         i64, i64, i64, i64, i64, i64, i64,
                                         i64, i64, i64, i64, f32, f32, i32,
                                         i32, vp, i64]
+    lib.ob_profile_enable.argtypes = [i32]
+    lib.ob_profile_enable.restype = None
+    lib.ob_profile_reset.argtypes = []
+    lib.ob_profile_reset.restype = None
+    lib.ob_profile_read.argtypes = [i32, ctypes.POINTER(ctypes.c_double),
+                                    ctypes.POINTER(ctypes.c_longlong)]
     lib.ob_flash_dsum_bf16.argtypes = [vp, vp, vp, i64, i64, i64, i64, vp]
     lib.ob_flash_bwd_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, i64,
                                       i64, i64, i64, f32, vp]

@@ -28,6 +28,46 @@ int ob_fail(const char* fmt, ...);
                      hipGetErrorString(e_));                          \
   } while (0)
 
+// ---------------------------------------------------------------------------
+// In-step profiler (state in ob_layer.hip).  When enabled (ob_profile_enable,
+// public ABI), every wrapped launch region is bracketed by HIP events ON THE
+// STREAM IT LAUNCHES ON and accumulated per family, so bench.py can report
+// the production dispatch's in-step per-launch times (roofline) and a
+// per-family step-time split.  Off by default: zero overhead in the timed
+// region (one branch per call).
+enum {
+  OB_PF_FC_FWD = 0,     // the roofline kernel: MLP fc forward GEMM
+  OB_PF_GEMM_FWD = 1,   // other forward GEMMs (qkv/attnproj/mlpproj/lm_head)
+  OB_PF_GEMM_DX = 2,    // backward activation-grad GEMMs
+  OB_PF_GEMM_DW = 3,    // weight-grad GEMMs (side stream)
+  OB_PF_FLASH_FWD = 4,  // flash fwd + its V^T staging
+  OB_PF_FLASH_BWD = 5,  // flash bwd (transposes + dsum + dkdv/dq)
+  OB_PF_ATTN_MAT = 6,   // non-flash attention matmuls + softmax
+  OB_PF_LN = 7,
+  OB_PF_CE = 8,
+  OB_PF_ELEM = 9,       // gelu / colsum / embed / misc elementwise
+  OB_PF_ADAMW = 10,
+  OB_PF_NFAM = 11
+};
+bool ob_prof_on();
+int ob_prof_beg(int fam, hipStream_t s);
+void ob_prof_end(int fam, int slot, hipStream_t s);
+
+#define OB_PROF(fid, strm, call)                                        \
+  ({                                                                    \
+    int _r;                                                             \
+    if (ob_prof_on()) {                                                 \
+      hipStream_t _ps = reinterpret_cast<hipStream_t>(strm);            \
+      const int _fi = (fid);                                            \
+      const int _slot = ob_prof_beg(_fi, _ps);                          \
+      _r = (call);                                                      \
+      ob_prof_end(_fi, _slot, _ps);                                     \
+    } else {                                                            \
+      _r = (call);                                                      \
+    }                                                                   \
+    _r;                                                                 \
+  })
+
 // internal launchers used by the layer orchestration (same semantics as the
 // public ob_* wrappers but C++ linkage, no error wrapping duplication).
 int ob_embed_fwd_f32(const int64_t* ids, const float* wte, const float* wpe,

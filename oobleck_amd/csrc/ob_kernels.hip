@@ -1064,10 +1064,13 @@ extern "C" int ob_adamw_step(void* p, const void* g, void* m, void* v,
   const double bc1 = 1.0 - pow((double)beta1, (double)step);
   const double bc2 = 1.0 - pow((double)beta2, (double)step);
   const int grid = (int)imin64((n + 255) / 256, 4096);
+  const int prof_slot =
+      ob_prof_on() ? ob_prof_beg(OB_PF_ADAMW, S(stream)) : -1;
   k_adamw<<<grid, 256, 0, S(stream)>>>(
       (float*)p, (const float*)g, (float*)m, (float*)v, n,
       1.f - lr * weight_decay, (float)(lr / bc1), (float)(1.0 / sqrt(bc2)),
       beta1, beta2, eps);
+  if (prof_slot >= 0) ob_prof_end(OB_PF_ADAMW, prof_slot, S(stream));
   OB_LAUNCH_CHECK();
   return 0;
 }

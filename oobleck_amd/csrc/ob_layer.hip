@@ -62,6 +62,81 @@ static int side_init() {
   return 0;
 }
 
+// ---------------------------------------------------------------------------
+// in-step profiler (ob_internal.h declares the family ids + OB_PROF macro).
+// Ring of reusable event pairs per family: a slot being reused is first
+// drained (sync + accumulate), so memory stays bounded while totals cover
+// every region.  Enabled only outside bench.py's timed region.
+// ---------------------------------------------------------------------------
+
+struct ProfFam {
+  static constexpr int RING = 64;
+  hipEvent_t beg[RING] = {}, end[RING] = {};
+  bool pending[RING] = {};
+  int head = 0;
+  double total_ms = 0.0;
+  long long count = 0;
+};
+static struct {
+  bool on = false;
+  ProfFam fam[OB_PF_NFAM];
+} g_prof;
+
+static void prof_drain_slot(ProfFam& f, int i) {
+  if (!f.pending[i]) return;
+  hipEventSynchronize(f.end[i]);
+  float ms = 0.f;
+  if (hipEventElapsedTime(&ms, f.beg[i], f.end[i]) == hipSuccess)
+    f.total_ms += ms;
+  f.pending[i] = false;
+}
+
+bool ob_prof_on() { return g_prof.on; }
+
+int ob_prof_beg(int fam, hipStream_t s) {
+  ProfFam& f = g_prof.fam[fam];
+  const int i = f.head;
+  if (!f.beg[i]) {
+    hipEventCreate(&f.beg[i]);
+    hipEventCreate(&f.end[i]);
+  }
+  prof_drain_slot(f, i);
+  hipEventRecord(f.beg[i], s);
+  return i;
+}
+
+void ob_prof_end(int fam, int slot, hipStream_t s) {
+  ProfFam& f = g_prof.fam[fam];
+  hipEventRecord(f.end[slot], s);
+  f.pending[slot] = true;
+  f.count++;
+  f.head = (slot + 1) % ProfFam::RING;
+}
+
+extern "C" void ob_profile_enable(int on) { g_prof.on = on != 0; }
+
+extern "C" void ob_profile_reset(void) {
+  for (auto& f : g_prof.fam) {
+    for (int i = 0; i < ProfFam::RING; i++)
+      if (f.pending[i]) {
+        hipEventSynchronize(f.end[i]);
+        f.pending[i] = false;
+      }
+    f.total_ms = 0.0;
+    f.count = 0;
+    f.head = 0;
+  }
+}
+
+extern "C" int ob_profile_read(int fam, double* total_ms, long long* count) {
+  if (fam < 0 || fam >= OB_PF_NFAM) return ob_fail("profile_read: bad fam");
+  ProfFam& f = g_prof.fam[fam];
+  for (int i = 0; i < ProfFam::RING; i++) prof_drain_slot(f, i);
+  *total_ms = f.total_ms;
+  *count = f.count;
+  return 0;
+}
+
 static int ws_ensure(float** buf, int64_t* cur, int64_t need) {
   if (need <= *cur) return 0;
   if (*buf) OB_HIP(hipFree(*buf));
@@ -550,13 +625,16 @@ static int block_forward_bf16(ob_layer* l, int slot, const __bf16* in,
   OB_HIP(hipMemcpyAsync(x, in, BS * H * sizeof(__bf16),
                         hipMemcpyDeviceToDevice, S(stream)));
   __bf16* ln1 = (__bf16*)(st + l->o_ln1);
-  if (ob_layernorm_fwd_bf16(x, p + bp.ln1_w, p + bp.ln1_b, ln1,
-                            st + l->o_mean1, st + l->o_rstd1, BS, H, 1e-5f,
-                            stream))
+  if (OB_PROF(OB_PF_LN, stream,
+              ob_layernorm_fwd_bf16(x, p + bp.ln1_w, p + bp.ln1_b, ln1,
+                                    st + l->o_mean1, st + l->o_rstd1, BS, H,
+                                    1e-5f, stream)))
     return 1;
   __bf16* qkv = (__bf16*)(st + l->o_qkv);
-  if (gemm_bf(0, 1, BS, 3 * H, H, 1.f, ln1, H, 0, 0, sh + l->sh_qkv_t, H, 0,
-              0, qkv, 3 * H, 0, 0, 1, 1, p + bp.b_qkv, nullptr, 0, 1, stream))
+  if (OB_PROF(OB_PF_GEMM_FWD, stream,
+              gemm_bf(0, 1, BS, 3 * H, H, 1.f, ln1, H, 0, 0,
+                      sh + l->sh_qkv_t, H, 0, 0, qkv, 3 * H, 0, 0, 1, 1,
+                      p + bp.b_qkv, nullptr, 0, 1, stream)))
     return 1;
   __bf16* am = (__bf16*)(st + l->o_attnm);
   if (use_flash(l)) {
@@ -564,45 +642,58 @@ static int block_forward_bf16(ob_layer* l, int slot, const __bf16* in,
     // the fused kernel writes O and the per-row LSE (stored where the
     // materialized-P path keeps P).
     __bf16* VTw = (__bf16*)g_ws.t2;
-    if (ob_transpose_bf16_b(qkv + 2 * H, VTw, Sq, 64, Sq * 3 * H, 64, 3 * H,
-                            B, nh, stream))
+    if (OB_PROF(OB_PF_FLASH_FWD, stream,
+                ob_transpose_bf16_b(qkv + 2 * H, VTw, Sq, 64, Sq * 3 * H, 64,
+                                    3 * H, B, nh, stream)))
       return 1;
     float* lseP = st + l->o_p;
-    if (ob_flash_fwd_bf16(qkv, VTw, am, lseP, B, Sq, H, nh,
-                          1.f / sqrtf((float)hd), stream))
+    if (OB_PROF(OB_PF_FLASH_FWD, stream,
+                ob_flash_fwd_bf16(qkv, VTw, am, lseP, B, Sq, H, nh,
+                                  1.f / sqrtf((float)hd), stream)))
       return 1;
   } else {
     __bf16* P = (__bf16*)(st + l->o_p);
-    if (gemm_bf(0, 1, Sq, Sq, hd, 1.f, qkv, 3 * H, Sq * 3 * H, hd, qkv + H,
-                3 * H, Sq * 3 * H, hd, P, Sq, nh * Sq * Sq, Sq * Sq, B, nh,
-                nullptr, nullptr, 0, 1, stream))
+    if (OB_PROF(OB_PF_ATTN_MAT, stream,
+                gemm_bf(0, 1, Sq, Sq, hd, 1.f, qkv, 3 * H, Sq * 3 * H, hd,
+                        qkv + H, 3 * H, Sq * 3 * H, hd, P, Sq, nh * Sq * Sq,
+                        Sq * Sq, B, nh, nullptr, nullptr, 0, 1, stream)))
       return 1;
-    if (ob_softmax_causal_fwd_bf16(P, B * nh, Sq, 1.f / sqrtf((float)hd),
-                                   stream))
+    if (OB_PROF(OB_PF_ATTN_MAT, stream,
+                ob_softmax_causal_fwd_bf16(P, B * nh, Sq,
+                                           1.f / sqrtf((float)hd), stream)))
       return 1;
-    if (gemm_bf(0, 0, Sq, hd, Sq, 1.f, P, Sq, nh * Sq * Sq, Sq * Sq,
-                qkv + 2 * H, 3 * H, Sq * 3 * H, hd, am, H, Sq * H, hd, B, nh,
-                nullptr, nullptr, 0, 1, stream))
+    if (OB_PROF(OB_PF_ATTN_MAT, stream,
+                gemm_bf(0, 0, Sq, hd, Sq, 1.f, P, Sq, nh * Sq * Sq, Sq * Sq,
+                        qkv + 2 * H, 3 * H, Sq * 3 * H, hd, am, H, Sq * H,
+                        hd, B, nh, nullptr, nullptr, 0, 1, stream)))
       return 1;
   }
   __bf16* hmid = (__bf16*)(st + l->o_hmid);
-  if (gemm_bf(0, 1, BS, H, H, 1.f, am, H, 0, 0, sh + l->sh_ap_t, H, 0, 0,
-              hmid, H, 0, 0, 1, 1, p + bp.b_attnproj, x, 0, 1, stream))
+  if (OB_PROF(OB_PF_GEMM_FWD, stream,
+              gemm_bf(0, 1, BS, H, H, 1.f, am, H, 0, 0, sh + l->sh_ap_t, H,
+                      0, 0, hmid, H, 0, 0, 1, 1, p + bp.b_attnproj, x, 0, 1,
+                      stream)))
     return 1;
   __bf16* ln2 = (__bf16*)(st + l->o_ln2);
-  if (ob_layernorm_fwd_bf16(hmid, p + bp.ln2_w, p + bp.ln2_b, ln2,
-                            st + l->o_mean2, st + l->o_rstd2, BS, H, 1e-5f,
-                            stream))
+  if (OB_PROF(OB_PF_LN, stream,
+              ob_layernorm_fwd_bf16(hmid, p + bp.ln2_w, p + bp.ln2_b, ln2,
+                                    st + l->o_mean2, st + l->o_rstd2, BS, H,
+                                    1e-5f, stream)))
     return 1;
   __bf16* u = (__bf16*)(st + l->o_u);
-  if (gemm_bf(0, 1, BS, 4 * H, H, 1.f, ln2, H, 0, 0, sh + l->sh_fc_t, H, 0,
-              0, u, 4 * H, 0, 0, 1, 1, p + bp.b_fc, nullptr, 0, 1, stream))
+  if (OB_PROF(OB_PF_FC_FWD, stream,
+              gemm_bf(0, 1, BS, 4 * H, H, 1.f, ln2, H, 0, 0, sh + l->sh_fc_t,
+                      H, 0, 0, u, 4 * H, 0, 0, 1, 1, p + bp.b_fc, nullptr, 0,
+                      1, stream)))
     return 1;
   __bf16* gact = (__bf16*)(st + l->o_g);
-  if (ob_gelu_fwd_bf16(u, gact, BS * 4 * H, stream)) return 1;
-  if (gemm_bf(0, 1, BS, H, 4 * H, 1.f, gact, 4 * H, 0, 0, sh + l->sh_mp_t,
-              4 * H, 0, 0, out, H, 0, 0, 1, 1, p + bp.b_mlpproj, hmid, 0, 1,
-              stream))
+  if (OB_PROF(OB_PF_ELEM, stream,
+              ob_gelu_fwd_bf16(u, gact, BS * 4 * H, stream)))
+    return 1;
+  if (OB_PROF(OB_PF_GEMM_FWD, stream,
+              gemm_bf(0, 1, BS, H, 4 * H, 1.f, gact, 4 * H, 0, 0,
+                      sh + l->sh_mp_t, 4 * H, 0, 0, out, H, 0, 0, 1, 1,
+                      p + bp.b_mlpproj, hmid, 0, 1, stream)))
     return 1;
   return 0;
 }
@@ -623,18 +714,21 @@ static int final_forward_bf16(ob_layer* l, int slot, const __bf16* in,
   OB_HIP(hipMemcpyAsync(labs, labels, BS * sizeof(int64_t),
                         hipMemcpyDeviceToDevice, S(stream)));
   __bf16* lnf = (__bf16*)(st + l->o_ln1);
-  if (ob_layernorm_fwd_bf16(x, p + 0, p + H, lnf, st + l->o_mean1,
-                            st + l->o_rstd1, BS, H, 1e-5f, stream))
+  if (OB_PROF(OB_PF_LN, stream,
+              ob_layernorm_fwd_bf16(x, p + 0, p + H, lnf, st + l->o_mean1,
+                                    st + l->o_rstd1, BS, H, 1e-5f, stream)))
     return 1;
   __bf16* logits = (__bf16*)(st + l->o_logits);
   // N = v_pad (zero-padded shadow rows -> padded logits are exactly 0)
-  if (gemm_bf(0, 1, BS, l->v_pad, H, 1.f, lnf, H, 0, 0,
-              l->shadows + l->sh_lm, H, 0, 0, logits, l->v_pad, 0, 0, 1, 1,
-              nullptr, nullptr, 0, 1, stream))
+  if (OB_PROF(OB_PF_GEMM_FWD, stream,
+              gemm_bf(0, 1, BS, l->v_pad, H, 1.f, lnf, H, 0, 0,
+                      l->shadows + l->sh_lm, H, 0, 0, logits, l->v_pad, 0, 0,
+                      1, 1, nullptr, nullptr, 0, 1, stream)))
     return 1;
   OB_HIP(hipMemsetAsync(out, 0, sizeof(float), S(stream)));
-  if (ob_ce_fwd_bf16(logits, labs, st + l->o_lse, out, B, Sq, V, l->v_pad,
-                     stream))
+  if (OB_PROF(OB_PF_CE, stream,
+              ob_ce_fwd_bf16(logits, labs, st + l->o_lse, out, B, Sq, V,
+                             l->v_pad, stream)))
     return 1;
   return 0;
 }
@@ -679,38 +773,55 @@ static int block_backward_bf16(ob_layer* l, int slot, const __bf16* dout,
   OB_HIP(hipMemcpyAsync(din, dout, BS * H * sizeof(__bf16),
                         hipMemcpyDeviceToDevice, S(stream)));
   // ---- MLP ----
-  if (gemm_bf(0, 1, BS, 4 * H, H, 1.f, dout, H, 0, 0, sh + l->sh_mp, H, 0, 0,
-              DY4, 4 * H, 0, 0, 1, 1, nullptr, nullptr, 0, 1, stream))
+  if (OB_PROF(OB_PF_GEMM_DX, stream,
+              gemm_bf(0, 1, BS, 4 * H, H, 1.f, dout, H, 0, 0, sh + l->sh_mp,
+                      H, 0, 0, DY4, 4 * H, 0, 0, 1, 1, nullptr, nullptr, 0,
+                      1, stream)))
     return 1;
-  if (dw_bf16_ws(gact, 4 * H, dout, H, BS, g + bp.w_mlpproj, H, g_ws.s1,
-                 g_ws.s2, side))
+  if (OB_PROF(OB_PF_GEMM_DW, side,
+              dw_bf16_ws(gact, 4 * H, dout, H, BS, g + bp.w_mlpproj, H,
+                         g_ws.s1, g_ws.s2, side)))
     return 1;
-  if (ob_colsum_bf16(dout, g + bp.b_mlpproj, BS, H, side)) return 1;
-  if (ob_gelu_bwd_bf16(u, DY4, DY4, BS * 4 * H, stream)) return 1;
+  if (OB_PROF(OB_PF_ELEM, side,
+              ob_colsum_bf16(dout, g + bp.b_mlpproj, BS, H, side)))
+    return 1;
+  if (OB_PROF(OB_PF_ELEM, stream,
+              ob_gelu_bwd_bf16(u, DY4, DY4, BS * 4 * H, stream)))
+    return 1;
   OB_HIP(hipEventRecord(g_side.e2, S(stream)));  // DY4 post-gelu
   OB_HIP(hipStreamWaitEvent(g_side.stream, g_side.e2, 0));
-  if (dw_bf16_ws(ln2, H, DY4, 4 * H, BS, g + bp.w_fc, 4 * H, g_ws.s1,
-                 g_ws.s2, side))
+  if (OB_PROF(OB_PF_GEMM_DW, side,
+              dw_bf16_ws(ln2, H, DY4, 4 * H, BS, g + bp.w_fc, 4 * H, g_ws.s1,
+                         g_ws.s2, side)))
     return 1;
-  if (ob_colsum_bf16(DY4, g + bp.b_fc, BS, 4 * H, side)) return 1;
-  if (gemm_bf(0, 1, BS, H, 4 * H, 1.f, DY4, 4 * H, 0, 0, sh + l->sh_fc,
-              4 * H, 0, 0, DLN, H, 0, 0, 1, 1, nullptr, nullptr, 0, 1,
-              stream))
+  if (OB_PROF(OB_PF_ELEM, side,
+              ob_colsum_bf16(DY4, g + bp.b_fc, BS, 4 * H, side)))
     return 1;
-  if (ob_layernorm_bwd_bf16(hmid, p + bp.ln2_w, st + l->o_mean2,
-                            st + l->o_rstd2, DLN, din, g + bp.ln2_w,
-                            g + bp.ln2_b, BS, H, 1, stream))
+  if (OB_PROF(OB_PF_GEMM_DX, stream,
+              gemm_bf(0, 1, BS, H, 4 * H, 1.f, DY4, 4 * H, 0, 0,
+                      sh + l->sh_fc, 4 * H, 0, 0, DLN, H, 0, 0, 1, 1,
+                      nullptr, nullptr, 0, 1, stream)))
+    return 1;
+  if (OB_PROF(OB_PF_LN, stream,
+              ob_layernorm_bwd_bf16(hmid, p + bp.ln2_w, st + l->o_mean2,
+                                    st + l->o_rstd2, DLN, din, g + bp.ln2_w,
+                                    g + bp.ln2_b, BS, H, 1, stream)))
     return 1;
   // ---- attention projection ----
   OB_HIP(hipEventRecord(g_side.e3, S(stream)));  // din post-ln2-bwd
   OB_HIP(hipStreamWaitEvent(g_side.stream, g_side.e3, 0));
-  if (gemm_bf(0, 1, BS, H, H, 1.f, din, H, 0, 0, sh + l->sh_ap, H, 0, 0,
-              DATT, H, 0, 0, 1, 1, nullptr, nullptr, 0, 1, stream))
+  if (OB_PROF(OB_PF_GEMM_DX, stream,
+              gemm_bf(0, 1, BS, H, H, 1.f, din, H, 0, 0, sh + l->sh_ap, H, 0,
+                      0, DATT, H, 0, 0, 1, 1, nullptr, nullptr, 0, 1,
+                      stream)))
     return 1;
-  if (dw_bf16_ws(am, H, din, H, BS, g + bp.w_attnproj, H, g_ws.s1, g_ws.s2,
-                 side))
+  if (OB_PROF(OB_PF_GEMM_DW, side,
+              dw_bf16_ws(am, H, din, H, BS, g + bp.w_attnproj, H, g_ws.s1,
+                         g_ws.s2, side)))
     return 1;
-  if (ob_colsum_bf16(din, g + bp.b_attnproj, BS, H, side)) return 1;
+  if (OB_PROF(OB_PF_ELEM, side,
+              ob_colsum_bf16(din, g + bp.b_attnproj, BS, H, side)))
+    return 1;
   // din is re-updated by the final ln1 backward: the main stream must
   // not reach it before the side finished reading din
   OB_HIP(hipEventRecord(g_side.s2, g_side.stream));
@@ -722,54 +833,74 @@ static int block_backward_bf16(ob_layer* l, int slot, const __bf16* dout,
     __bf16* dOTw = KTw + BSH;
     float* Dbuf = g_ws.t1 + (3 * BSH + 1) / 2;
     const float* lseP = st + l->o_p;
-    if (ob_transpose_bf16_b(qkv, QTw, Sq, 64, Sq * 3 * H, 64, 3 * H, B, nh,
-                            stream))
+    if (OB_PROF(OB_PF_FLASH_BWD, stream,
+                ob_transpose_bf16_b(qkv, QTw, Sq, 64, Sq * 3 * H, 64, 3 * H,
+                                    B, nh, stream)))
       return 1;
-    if (ob_transpose_bf16_b(qkv + H, KTw, Sq, 64, Sq * 3 * H, 64, 3 * H, B,
-                            nh, stream))
+    if (OB_PROF(OB_PF_FLASH_BWD, stream,
+                ob_transpose_bf16_b(qkv + H, KTw, Sq, 64, Sq * 3 * H, 64,
+                                    3 * H, B, nh, stream)))
       return 1;
-    if (ob_transpose_bf16_b(DATT, dOTw, Sq, 64, Sq * H, 64, H, B, nh,
-                            stream))
+    if (OB_PROF(OB_PF_FLASH_BWD, stream,
+                ob_transpose_bf16_b(DATT, dOTw, Sq, 64, Sq * H, 64, H, B, nh,
+                                    stream)))
       return 1;
-    if (ob_flash_dsum_bf16(am, DATT, Dbuf, B, Sq, H, nh, stream)) return 1;
-    if (ob_flash_bwd_bf16(qkv, QTw, KTw, dOTw, DATT, lseP, Dbuf, DQKV, B,
-                          Sq, H, nh, scale, stream))
+    if (OB_PROF(OB_PF_FLASH_BWD, stream,
+                ob_flash_dsum_bf16(am, DATT, Dbuf, B, Sq, H, nh, stream)))
+      return 1;
+    if (OB_PROF(OB_PF_FLASH_BWD, stream,
+                ob_flash_bwd_bf16(qkv, QTw, KTw, dOTw, DATT, lseP, Dbuf,
+                                  DQKV, B, Sq, H, nh, scale, stream)))
       return 1;
   } else {
-    if (gemm_bf(0, 1, Sq, Sq, hd, 1.f, DATT, H, Sq * H, hd, qkv + 2 * H,
-                3 * H, Sq * 3 * H, hd, DP, Sq, nh * Sq * Sq, Sq * Sq, B, nh,
-                nullptr, nullptr, 0, 1, stream))
+    if (OB_PROF(OB_PF_ATTN_MAT, stream,
+                gemm_bf(0, 1, Sq, Sq, hd, 1.f, DATT, H, Sq * H, hd,
+                        qkv + 2 * H, 3 * H, Sq * 3 * H, hd, DP, Sq,
+                        nh * Sq * Sq, Sq * Sq, B, nh, nullptr, nullptr, 0, 1,
+                        stream)))
       return 1;
-    if (ob_softmax_causal_bwd_bf16(P, DP, B * nh, Sq, stream)) return 1;
-    if (gemm_bf(0, 0, Sq, hd, Sq, scale, DP, Sq, nh * Sq * Sq, Sq * Sq,
-                qkv + H, 3 * H, Sq * 3 * H, hd, DQKV, 3 * H, Sq * 3 * H, hd,
-                B, nh, nullptr, nullptr, 0, 1, stream))
+    if (OB_PROF(OB_PF_ATTN_MAT, stream,
+                ob_softmax_causal_bwd_bf16(P, DP, B * nh, Sq, stream)))
       return 1;
-    if (gemm_bf(1, 0, Sq, hd, Sq, scale, DP, Sq, nh * Sq * Sq, Sq * Sq, qkv,
-                3 * H, Sq * 3 * H, hd, DQKV + H, 3 * H, Sq * 3 * H, hd, B,
-                nh, nullptr, nullptr, 0, 1, stream))
+    if (OB_PROF(OB_PF_ATTN_MAT, stream,
+                gemm_bf(0, 0, Sq, hd, Sq, scale, DP, Sq, nh * Sq * Sq,
+                        Sq * Sq, qkv + H, 3 * H, Sq * 3 * H, hd, DQKV, 3 * H,
+                        Sq * 3 * H, hd, B, nh, nullptr, nullptr, 0, 1,
+                        stream)))
       return 1;
-    if (gemm_bf(1, 0, Sq, hd, Sq, 1.f, P, Sq, nh * Sq * Sq, Sq * Sq, DATT,
-                H, Sq * H, hd, DQKV + 2 * H, 3 * H, Sq * 3 * H, hd, B, nh,
-                nullptr, nullptr, 0, 1, stream))
+    if (OB_PROF(OB_PF_ATTN_MAT, stream,
+                gemm_bf(1, 0, Sq, hd, Sq, scale, DP, Sq, nh * Sq * Sq,
+                        Sq * Sq, qkv, 3 * H, Sq * 3 * H, hd, DQKV + H, 3 * H,
+                        Sq * 3 * H, hd, B, nh, nullptr, nullptr, 0, 1,
+                        stream)))
+      return 1;
+    if (OB_PROF(OB_PF_ATTN_MAT, stream,
+                gemm_bf(1, 0, Sq, hd, Sq, 1.f, P, Sq, nh * Sq * Sq, Sq * Sq,
+                        DATT, H, Sq * H, hd, DQKV + 2 * H, 3 * H, Sq * 3 * H,
+                        hd, B, nh, nullptr, nullptr, 0, 1, stream)))
       return 1;
   }
   // ---- QKV projection ----
   OB_HIP(hipEventRecord(g_side.e1, S(stream)));  // DQKV ready
   OB_HIP(hipStreamWaitEvent(g_side.stream, g_side.e1, 0));
-  if (ob_colsum_bf16(DQKV, g + bp.b_qkv, BS, 3 * H, side)) return 1;
-  if (dw_bf16_ws(ln1, H, DQKV, 3 * H, BS, g + bp.w_qkv, 3 * H, g_ws.s1,
-                 g_ws.s2, side))
+  if (OB_PROF(OB_PF_ELEM, side,
+              ob_colsum_bf16(DQKV, g + bp.b_qkv, BS, 3 * H, side)))
     return 1;
-  if (gemm_bf(0, 1, BS, H, 3 * H, 1.f, DQKV, 3 * H, 0, 0, sh + l->sh_qkv,
-              3 * H, 0, 0, DLN, H, 0, 0, 1, 1, nullptr, nullptr, 0, 1,
-              stream))
+  if (OB_PROF(OB_PF_GEMM_DW, side,
+              dw_bf16_ws(ln1, H, DQKV, 3 * H, BS, g + bp.w_qkv, 3 * H,
+                         g_ws.s1, g_ws.s2, side)))
+    return 1;
+  if (OB_PROF(OB_PF_GEMM_DX, stream,
+              gemm_bf(0, 1, BS, H, 3 * H, 1.f, DQKV, 3 * H, 0, 0,
+                      sh + l->sh_qkv, 3 * H, 0, 0, DLN, H, 0, 0, 1, 1,
+                      nullptr, nullptr, 0, 1, stream)))
     return 1;
   // ln1 backward accumulates into din: wait for the side's din readers
   OB_HIP(hipStreamWaitEvent(S(stream), g_side.s2, 0));
-  if (ob_layernorm_bwd_bf16(x, p + bp.ln1_w, st + l->o_mean1,
-                            st + l->o_rstd1, DLN, din, g + bp.ln1_w,
-                            g + bp.ln1_b, BS, H, 1, stream))
+  if (OB_PROF(OB_PF_LN, stream,
+              ob_layernorm_bwd_bf16(x, p + bp.ln1_w, st + l->o_mean1,
+                                    st + l->o_rstd1, DLN, din, g + bp.ln1_w,
+                                    g + bp.ln1_b, BS, H, 1, stream)))
     return 1;
   // join: the next call reuses DY4/DQKV/din workspaces on the main stream
   OB_HIP(hipEventRecord(g_side.sf, g_side.stream));
@@ -790,8 +921,9 @@ static int final_backward_bf16(ob_layer* l, int slot, const float* dout,
   int64_t* labs = l->ids + (int64_t)slot * (int64_t)l->d.max_batch * Sq;
   __bf16* DLN = (__bf16*)g_ws.bsh1;
 
-  if (ob_ce_bwd_bf16(logits, labs, st + l->o_lse, dout, B, Sq, V, l->v_pad,
-                     stream))
+  if (OB_PROF(OB_PF_CE, stream,
+              ob_ce_bwd_bf16(logits, labs, st + l->o_lse, dout, B, Sq, V,
+                             l->v_pad, stream)))
     return 1;
   // dW_lm on the side stream (overlaps d_lnout + ln backward below):
   // transpose dlogits and lnf so the GEMM runs on the glds path; tile
@@ -809,27 +941,33 @@ static int final_backward_bf16(ob_layer* l, int slot, const float* dout,
     }();
     int r = -1;
     if (!no_lt)
-      r = ob_gemm_lt(1, 0, V, H, BS, 1.f, logits, l->v_pad, lnf, H, 1.f,
-                     g + 2 * H, H, 1, side);
+      r = OB_PROF(OB_PF_GEMM_DW, side,
+                  ob_gemm_lt(1, 0, V, H, BS, 1.f, logits, l->v_pad, lnf, H,
+                             1.f, g + 2 * H, H, 1, side));
     if (r > 0) return 1;
     if (r < 0) {
       __bf16* DLT = (__bf16*)g_ws.s1;
       __bf16* LNT = (__bf16*)g_ws.s2;
       if (ob_transpose_bf16(logits, DLT, BS, l->v_pad, side)) return 1;
       if (ob_transpose_bf16(lnf, LNT, BS, H, side)) return 1;
-      if (ob_gemm_bf16_nt_dispatch(DLT, LNT, g + 2 * H, nullptr, nullptr,
-                                   l->v_pad, H, BS, BS, BS, H, 0, 0, 0, 0,
-                                   0, 0, 1, 1, 1.f, 0.f, 2, 2, side, V))
+      if (OB_PROF(OB_PF_GEMM_DW, side,
+                  ob_gemm_bf16_nt_dispatch(DLT, LNT, g + 2 * H, nullptr,
+                                           nullptr, l->v_pad, H, BS, BS, BS,
+                                           H, 0, 0, 0, 0, 0, 0, 1, 1, 1.f,
+                                           0.f, 2, 2, side, V)))
         return 1;
     }
   }
   // d_lnout: K = v_pad (padded dlogits cols and shadow^T cols are zero)
-  if (gemm_bf(0, 1, BS, H, l->v_pad, 1.f, logits, l->v_pad, 0, 0,
-              l->shadows + l->sh_lm_t, l->v_pad, 0, 0, DLN, H, 0, 0, 1, 1,
-              nullptr, nullptr, 0, 1, stream))
+  if (OB_PROF(OB_PF_GEMM_DX, stream,
+              gemm_bf(0, 1, BS, H, l->v_pad, 1.f, logits, l->v_pad, 0, 0,
+                      l->shadows + l->sh_lm_t, l->v_pad, 0, 0, DLN, H, 0, 0,
+                      1, 1, nullptr, nullptr, 0, 1, stream)))
     return 1;
-  if (ob_layernorm_bwd_bf16(x, p + 0, st + l->o_mean1, st + l->o_rstd1, DLN,
-                            din, g + 0, g + H, BS, H, 0, stream))
+  if (OB_PROF(OB_PF_LN, stream,
+              ob_layernorm_bwd_bf16(x, p + 0, st + l->o_mean1,
+                                    st + l->o_rstd1, DLN, din, g + 0, g + H,
+                                    BS, H, 0, stream)))
     return 1;
   // join: the next backward reuses s1/s2 and reads g on the main stream
   OB_HIP(hipEventRecord(g_side.sf, g_side.stream));
@@ -851,10 +989,12 @@ extern "C" int ob_layer_forward(ob_layer_t l, int32_t slot, const void* in,
       OB_HIP(hipMemcpyAsync(ids, in, BS * sizeof(int64_t),
                             hipMemcpyDeviceToDevice, S(stream)));
       if (l->d.dtype == 1)
-        return ob_embed_fwd_bf16(ids, l->params, l->params + V * H, out, l->B,
-                                 Sq, H, stream);
-      return ob_embed_fwd_f32(ids, l->params, l->params + V * H, (float*)out,
-                              l->B, Sq, H, stream);
+        return OB_PROF(OB_PF_ELEM, stream,
+                       ob_embed_fwd_bf16(ids, l->params, l->params + V * H,
+                                         out, l->B, Sq, H, stream));
+      return OB_PROF(OB_PF_ELEM, stream,
+                     ob_embed_fwd_f32(ids, l->params, l->params + V * H,
+                                      (float*)out, l->B, Sq, H, stream));
     }
     case OB_KIND_BLOCK:
       if (l->d.dtype == 1)
@@ -1021,10 +1161,14 @@ extern "C" int ob_layer_backward(ob_layer_t l, int32_t slot, const void* dout,
       const int64_t Sq = l->d.seq_len, H = l->d.n_embd, V = l->d.vocab_size;
       int64_t* ids = l->ids + (int64_t)slot * (int64_t)l->d.max_batch * Sq;
       if (l->d.dtype == 1)
-        return ob_embed_bwd_bf16(ids, dout, l->grads, l->grads + V * H, l->B,
-                                 Sq, H, stream);
-      return ob_embed_bwd_f32(ids, (const float*)dout, l->grads,
-                              l->grads + V * H, l->B, Sq, H, stream);
+        return OB_PROF(OB_PF_ELEM, stream,
+                       ob_embed_bwd_bf16(ids, dout, l->grads,
+                                         l->grads + V * H, l->B, Sq, H,
+                                         stream));
+      return OB_PROF(OB_PF_ELEM, stream,
+                     ob_embed_bwd_f32(ids, (const float*)dout, l->grads,
+                                      l->grads + V * H, l->B, Sq, H,
+                                      stream));
     }
     case OB_KIND_BLOCK:
       if (!dout || !din) return ob_fail("block backward: dout/din required");

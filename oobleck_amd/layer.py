@@ -90,6 +90,7 @@ class Layer:
                   "layer_bind")
             self._param_handle = _FlatParamHandle(flat, grad)
         self._batch = max_batch
+        self._shadows_stale = False
         self.refresh_weights()
 
     # -- compute ------------------------------------------------------------
@@ -106,10 +107,21 @@ class Layer:
             check(get_ext().ob_layer_set_batch(self._h, batch), "set_batch")
             self._batch = batch
 
+    def _unshard(self) -> None:
+        """All-gather the shards into the bound full buffer, then (bf16)
+        re-cast the extension's bf16 shadows if the optimizer touched the
+        sharded masters since the last unshard — the shadows are cast from
+        the FULL buffer, which only becomes current here."""
+        self._sharded.unshard()
+        if self._shadows_stale:
+            check(get_ext().ob_layer_refresh_weights(self._h, _stream_ptr()),
+                  f"refresh_weights layer {self.layer_id}")
+            self._shadows_stale = False
+
     def forward_slot(self, slot: int, x: torch.Tensor, out: torch.Tensor,
                      labels: torch.Tensor | None = None) -> None:
         if self._sharded is not None:
-            self._sharded.unshard()  # pre_forward_hook (layer.py:147-153)
+            self._unshard()  # pre_forward_hook (layer.py:147-153)
         check(get_ext().ob_layer_forward(self._h, slot, _ptr(x), _ptr(out),
                                          _ptr(labels), _stream_ptr()),
               f"forward layer {self.layer_id}")
@@ -117,7 +129,7 @@ class Layer:
     def backward_slot(self, slot: int, dout: torch.Tensor | None,
                       din: torch.Tensor | None) -> None:
         if self._sharded is not None:
-            self._sharded.unshard()  # pre_backward_hook (layer.py:160-166)
+            self._unshard()  # pre_backward_hook (layer.py:160-166)
         check(get_ext().ob_layer_backward(self._h, slot, _ptr(dout), _ptr(din),
                                           _stream_ptr()),
               f"backward layer {self.layer_id}")
@@ -132,10 +144,18 @@ class Layer:
 
     def refresh_weights(self) -> None:
         """bf16 mode: re-cast the extension's bf16 weight shadows from the
-        fp32 master params (after init and after every optimizer step)."""
-        if self.dtype == "bf16":
-            check(get_ext().ob_layer_refresh_weights(self._h, _stream_ptr()),
-                  f"refresh_weights layer {self.layer_id}")
+        fp32 master params (after init and after every optimizer step).
+        FULL_SHARD: the optimizer updates the SHARDED masters, but the
+        shadows are cast from the bound full buffer, which is only
+        re-gathered at the next unshard — so defer the cast to _unshard()
+        instead of silently training on one-step-stale weights."""
+        if self.dtype != "bf16":
+            return
+        if self._sharded is not None:
+            self._shadows_stale = True
+            return
+        check(get_ext().ob_layer_refresh_weights(self._h, _stream_ptr()),
+              f"refresh_weights layer {self.layer_id}")
 
     # -- distributed surface (reference layer.py:272-291) --------------------
     def _shard_param(self, tensor: torch.Tensor, number: int) -> list[torch.Tensor]:
@@ -160,9 +180,26 @@ class Layer:
             grads = [self.flat_grad]
         for grad, (_idx, pg) in zip(grads, process_groups.items()):
             torch.distributed.all_reduce(tensor=grad, group=pg)
+        if len(process_groups) > 1:
+            # Deviation from the reference (layer.py:283-291): when the flat
+            # grad is not divisible by the group count, F.pad COPIES the tail
+            # chunk, so the reference's reduced tail values never land back
+            # in the grad (silently un-reduced).  Write the reduced tail
+            # back; every other chunk is a view and needs nothing.
+            flat = self.flat_grad
+            chunk = grads[0].numel()
+            tail_off = ((flat.numel() - 1) // chunk) * chunk
+            tail_len = flat.numel() - tail_off
+            if tail_len < chunk:
+                flat[tail_off:].copy_(grads[tail_off // chunk][:tail_len])
 
     def remove_tensors(self) -> None:
-        # reference layer.py:66-69 (reconfiguration discards a layer's state)
+        # reference layer.py:66-69 (reconfiguration discards a layer's state).
+        # Also unbind the extension from the raw device pointers so a stale
+        # forward/backward/adamw on the discarded layer fails loudly
+        # ("params not bound") instead of reading freed storage.
+        check(get_ext().ob_layer_bind(self._h, ctypes.c_void_p(0),
+                                      ctypes.c_void_p(0)), "unbind")
         self._param_handle.flat_param.grad = None
         self._param_handle.flat_param.data = torch.tensor([], device=self.device)
 
