@@ -247,11 +247,26 @@ class PipelineCommunication:
     def pipeline(self) -> "OobleckPipeline":
         return self._pipeline_ref
 
+    def _gloo_cuda(self, tensor: torch.Tensor) -> bool:
+        # gloo has no CUDA send/recv: stage through host memory.  Used by
+        # the single-GPU pp>1 composition tests (two ranks sharing one
+        # device); the production path is RCCL (backend "nccl").
+        return tensor.is_cuda and \
+            dist.get_backend(self._process_group) == "gloo"
+
     def _send(self, tensor: torch.Tensor, dest: int) -> None:
+        if self._gloo_cuda(tensor):
+            tensor = tensor.cpu()
         dist.send(tensor, dest, group=self._process_group)
 
     def _recv(self, tensor: torch.Tensor, src: int) -> None:
-        dist.recv(tensor, src, group=self._process_group)
+        if self._gloo_cuda(tensor):
+            host = torch.empty(tensor.shape, dtype=tensor.dtype, device="cpu")
+            dist.recv(host, src, group=self._process_group)
+            with torch.no_grad():
+                tensor.copy_(host)
+        else:
+            dist.recv(tensor, src, group=self._process_group)
 
     def _pack_meta(self, buffers: tuple[torch.Tensor, ...]) -> torch.Tensor:
         assert len(buffers) <= _META_MAX_TENSORS

@@ -1,0 +1,148 @@
+# @gpu pp>1 composition test: OobleckPipeline.train() driving REAL HIP
+# Layer objects at pp2 — schedule + p2p + HIP compute + optimizer, the
+# exact composition the multi-GPU bench depends on (reference
+# pipeline.py:288-427 + :458-487), previously covered only piecewise
+# (gloo tests used OracleLayer; GPU tests drove layers directly).
+#
+# Two processes SHARE cuda:0 (the round's lease is one GPU): rendezvous is
+# gloo, so the p2p path stages activations/grads through host memory
+# (PipelineCommunication._gloo_cuda) — same instruction order, same wire
+# content as the RCCL path.
+from __future__ import annotations
+
+import json
+import os
+import pathlib
+import sys
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(),
+                                  reason="needs MI355X")
+
+REPO_ROOT = pathlib.Path(__file__).resolve().parent.parent
+
+# head_dim 64 + S % 128 == 0 -> the bf16 leg runs the flash-attention path
+DIMS = dict(n_embd=128, n_head=2, n_layer=2, n_positions=128, vocab_size=307)
+B, S, MB = 2, 128, 4
+
+
+def _setup(rank: int, world: int, tmp: str):
+    if str(REPO_ROOT) not in sys.path:
+        sys.path.insert(0, str(REPO_ROOT))
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    dist.init_process_group(
+        "gloo", init_method=f"file://{tmp}/rdzv", rank=rank, world_size=world)
+    torch.cuda.set_device(0)
+
+
+def _batches(vocab, n, seed):
+    g = torch.Generator().manual_seed(seed)
+    out = []
+    for _ in range(n):
+        ids = torch.randint(0, vocab, (B, S), generator=g)
+        out.append((ids, ids.clone()))
+    return out
+
+
+def _run_pp2(rank: int, world: int, tmp: str, dtype: str):
+    _setup(rank, world, tmp)
+    from oracle.gpt2_oracle import OracleConfig, adamw_step, init_layer_params
+    from oracle.gpt2_oracle import stage_forward_backward
+
+    from oobleck_amd.config import ModelConfig, TrainingConfig
+    from oobleck_amd.engine import DataParallelEngine, make_rank_grid
+    from oobleck_amd.layer import Layer
+    from oobleck_amd.optimizer import FusedAdamW, WarmupLR
+    from oobleck_amd.pipeline import OobleckPipeline
+
+    dev = torch.device("cuda", 0)
+    mc = ModelConfig(**DIMS)
+    oc = OracleConfig(**DIMS)
+    tc = TrainingConfig(microbatch_size=B, global_microbatch_size=B * MB,
+                        seq_len=S, lr=1e-3)
+    L = oc.n_layers_total
+    flats = [init_layer_params(oc, oc.layer_kind(i), 42 * 100 + i)
+             for i in range(L)]
+    # stage 0: embedding + block 1; stage 1: block 2 + final
+    grid = make_rank_grid(L, [[0, 1], [2, 3]], [[0], [1]])
+
+    class Loader:
+        def __iter__(self):
+            return iter({"input_ids": i, "labels": l}
+                        for i, l in _batches(oc.vocab_size, MB, seed=7))
+
+    pipe = OobleckPipeline(0, grid, mc, tc, Loader(), MB, dev)
+    pipe.initialize_distributed_fsdp()
+    pipe.initialize_distributed_pipeline()
+
+    def layer_factory(lid, pg, n_slots):
+        layer = Layer(lid, mc, B, S, n_slots, dev, dtype=dtype)
+        layer.flat_param.copy_(flats[lid].to(dev))
+        layer.refresh_weights()
+        return layer
+
+    def optimizer_factory(layers):
+        opt = FusedAdamW(layers, lr=tc.lr)
+        return opt, WarmupLR(opt, 0)
+
+    pipe.initialize_execution(layer_factory, optimizer_factory)
+    dp = DataParallelEngine([pipe])
+
+    pipe.train()
+    dp.do_allreduce(pipe)
+    torch.cuda.synchronize()
+
+    # single-process oracle reference: grads summed over the microbatches
+    grads_ref = [torch.zeros_like(f) for f in flats]
+    losses_ref = []
+    for ids, labels in _batches(oc.vocab_size, MB, seed=7):
+        loss, _, gs = stage_forward_backward(oc, flats, list(range(L)), ids,
+                                             labels=labels)
+        losses_ref.append(loss)
+        for g, gi in zip(grads_ref, gs):
+            g += gi
+
+    if pipe.is_last_stage():
+        total_ref = sum(l.item() for l in losses_ref)
+        got = pipe.execution.total_loss.item()
+        tol = 1e-4 if dtype == "f32" else 2e-2
+        assert abs(got - total_ref) < tol * abs(total_ref), (got, total_ref)
+    for layer in pipe.execution._layers:
+        got = layer.flat_grad.cpu()
+        ref = grads_ref[layer.layer_id]
+        if dtype == "f32":
+            torch.testing.assert_close(got, ref, rtol=1e-3, atol=1e-3)
+        else:
+            rel = (got - ref).norm() / ref.norm().clamp_min(1e-12)
+            assert rel < 8e-2, (layer.layer_id, rel.item())
+
+    # optimizer step through the C-ABI fused AdamW; parity vs the oracle's
+    # adamw on the same accumulated grads
+    pipe.execution.optimizer_step()
+    torch.cuda.synchronize()
+    for layer in pipe.execution._layers:
+        lid = layer.layer_id
+        p = flats[lid].clone()
+        m = torch.zeros_like(p)
+        v = torch.zeros_like(p)
+        adamw_step(p, layer.flat_grad.cpu(), m, v, step=1, lr=tc.lr,
+                   beta1=tc.adam_beta1, beta2=tc.adam_beta2,
+                   eps=tc.adam_eps, weight_decay=tc.weight_decay)
+        torch.testing.assert_close(layer.flat_param.cpu(), p, rtol=1e-5,
+                                   atol=1e-6)
+
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", ["f32", "bf16"])
+def test_pp2_hip_end_to_end(dtype, tmp_path):
+    """pp2 with HIP compute: two ranks on one GPU, full 1F1B schedule."""
+    mp.spawn(_run_pp2, args=(2, str(tmp_path), dtype), nprocs=2, join=True)
