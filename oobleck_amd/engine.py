@@ -129,6 +129,12 @@ def copy_model_states(old_rank_grids: list[dict[int, list[int]]],
     # free on RCCL (stream-ordered).
     for w in works:
         w.wait()
+    # the copy wrote master params behind the layers' backs: refresh any
+    # derived state (bf16 shadows; FULL_SHARD resident-full re-gather)
+    for layer in my_layers_by_id.values():
+        refresh = getattr(layer, "refresh_weights", None)
+        if refresh is not None:
+            refresh()
     dist.barrier()
     if torch.cuda.is_available():
         torch.cuda.synchronize()

@@ -130,14 +130,20 @@ class Layer:
                       din: torch.Tensor | None) -> None:
         if self._sharded is not None:
             self._unshard()  # pre_backward_hook (layer.py:160-166)
+            # the post stream's previous reduce-scatter zeroes full_grad;
+            # this backward accumulates into it — order after it
+            self._sharded.wait_post()
         check(get_ext().ob_layer_backward(self._h, slot, _ptr(dout), _ptr(din),
                                           _stream_ptr()),
               f"backward layer {self.layer_id}")
         if self._sharded is not None:
-            # post_backward_hook: reduce-scatter + accumulate (layer.py:167-225)
+            # post_backward_hook: reduce-scatter + accumulate on the post
+            # stream (layer.py:167-225, post_backward_stream :183)
             self._sharded.reduce_scatter_grad()
 
     def zero_grads(self) -> None:
+        if self._sharded is not None:
+            self._sharded.wait_post()
         self.flat_grad.zero_()
         if self._sharded is not None:
             self._sharded.full_grad.zero_()
@@ -149,13 +155,16 @@ class Layer:
         shadows are cast from the bound full buffer, which is only
         re-gathered at the next unshard — so defer the cast to _unshard()
         instead of silently training on one-step-stale weights."""
-        if self.dtype != "bf16":
-            return
         if self._sharded is not None:
-            self._shadows_stale = True
+            # shard changed: the resident full buffer must be re-gathered
+            # at the next unshard (any dtype)
+            self._sharded.mark_dirty()
+            if self.dtype == "bf16":
+                self._shadows_stale = True
             return
-        check(get_ext().ob_layer_refresh_weights(self._h, _stream_ptr()),
-              f"refresh_weights layer {self.layer_id}")
+        if self.dtype == "bf16":
+            check(get_ext().ob_layer_refresh_weights(self._h, _stream_ptr()),
+                  f"refresh_weights layer {self.layer_id}")
 
     # -- distributed surface (reference layer.py:272-291) --------------------
     def _shard_param(self, tensor: torch.Tensor, number: int) -> list[torch.Tensor]:
@@ -174,6 +183,8 @@ class Layer:
         reference, which never divides by the DP degree)."""
         assert all(torch.distributed.get_rank(pg) >= 0
                    for pg in process_groups.values())
+        if self._sharded is not None:
+            self._sharded.wait_post()  # sharded grad fully accumulated
         if len(process_groups) > 1:
             grads = self._shard_param(self.flat_grad, len(process_groups))
         else:

@@ -5,17 +5,27 @@
 #   * the authoritative state is each rank's contiguous SHARD of the flat
 #     parameter (world-size-padded, rank-ordered chunks — FSDP layout);
 #   * forward/backward UNSHARD by all-gathering the shards into the full
-#     flat buffer the compute kernel is bound to (layer.py:117-131);
+#     flat buffer the compute kernel is bound to (layer.py:117-131).
+#     Deviation from the reference, by design: the reference reshards
+#     (frees) the full params after every fwd/bwd to save memory, so it
+#     must re-all-gather per microbatch; with 288 GB of HBM3E the full
+#     buffer stays RESIDENT here, so the gather runs only when the shard
+#     actually changed (after the optimizer step / a reconfig copy —
+#     mark_dirty()).  Numerically identical, (M-1) fewer all-gathers per
+#     step at M microbatches.
 #   * after each microbatch's backward, the full grad is reduce-scattered
-#     and ACCUMULATED into _saved_grad_shard, and the full grad cleared
-#     (layer.py:167-225 post_backward_hook semantics);
+#     and ACCUMULATED into the sharded grad, and the full grad cleared
+#     (layer.py:167-225 post_backward_hook semantics).  On CUDA this runs
+#     on a dedicated POST stream (the reference's post_backward_stream,
+#     layer.py:183), event-fenced: the next writer of full_grad / reader
+#     of the sharded grad calls wait_post() first, so the collective
+#     overlaps subsequent backward compute on the main stream.
 #   * the optimizer and DP all-reduce then operate on the shard
 #     (reduce_gradients' prepare_gradient_for_optim, layer.py:272-291).
 #
 # Used by both the HIP Layer (compute = C-ABI kernels) and the test-only
 # OracleLayer (compute = oracle), so the collective choreography is covered
-# by CPU gloo tests.  Round-1 note: collectives run on the compute stream
-# (the reference overlaps on pre/post side streams — a later optimization).
+# by CPU gloo tests (streams are CUDA-only and bypassed on CPU).
 from __future__ import annotations
 
 import torch
@@ -44,19 +54,55 @@ class ShardedFlatParam:
                                (self.rank + 1) * self.shard_size].clone()
         self.shard.grad = torch.zeros_like(self.shard)
         self._scatter_tmp = torch.empty_like(self.shard)
+        self._dirty = False  # full == gather(shards) right after init
+        self._use_streams = (isinstance(device, torch.device)
+                             and device.type == "cuda")
+        self._rs_pending = False
+        if self._use_streams:
+            self._post = torch.cuda.Stream()
+            self._post_ev = torch.cuda.Event()
 
     @property
     def saved_grad_shard(self) -> torch.Tensor:
         return self.shard.grad
 
+    def mark_dirty(self) -> None:
+        """The shard changed (optimizer step / reconfig copy): the next
+        unshard() must re-gather the full buffer."""
+        self._dirty = True
+
     def unshard(self) -> None:
         # all_gather_into_tensor: rank-ordered concat == the full flat param
+        if not self._dirty:
+            return
         dist.all_gather_into_tensor(self.full, self.shard, group=self.pg)
+        self._dirty = False
 
     def reduce_scatter_grad(self) -> None:
         """Per-microbatch: reduce-scatter the full grad, accumulate into the
-        sharded grad, clear the full grad for the next microbatch."""
-        dist.reduce_scatter_tensor(self._scatter_tmp, self.full_grad,
-                                   group=self.pg)
-        self.shard.grad += self._scatter_tmp
-        self.full_grad.zero_()
+        sharded grad, clear the full grad for the next microbatch.  On CUDA
+        this runs on the post stream (overlaps later backward compute);
+        wait_post() fences the next consumer."""
+        if not self._use_streams:
+            dist.reduce_scatter_tensor(self._scatter_tmp, self.full_grad,
+                                       group=self.pg)
+            self.shard.grad += self._scatter_tmp
+            self.full_grad.zero_()
+            return
+        cur = torch.cuda.current_stream()
+        self._post.wait_stream(cur)  # full_grad finished accumulating
+        with torch.cuda.stream(self._post):
+            dist.reduce_scatter_tensor(self._scatter_tmp, self.full_grad,
+                                       group=self.pg)
+            self.shard.grad += self._scatter_tmp
+            self.full_grad.zero_()
+            self._post_ev.record(self._post)
+        self._rs_pending = True
+
+    def wait_post(self) -> None:
+        """Order the main stream after the last posted reduce-scatter:
+        called before anything writes full_grad (the next microbatch's
+        backward) or reads the sharded grad (DP all-reduce, optimizer)."""
+        if self._rs_pending:
+            torch.cuda.current_stream().wait_event(self._post_ev)
+            self._rs_pending = False
