@@ -1,0 +1,252 @@
+# Planner rebuild (oobleck_amd/planning): behavior pinned by the
+# reference's OWN planner tests
+# (/root/reference/tests/planning/test_pipeline_template.py:15-93) plus
+# the rank-grid semantics of pipeline_template.h:57-84, and integrated
+# with the pipeline host logic: the world-8 4-stage x 2-replica drill is
+# driven off the REAL PipelineTemplate.get_rank_grid output instead of
+# the round-1 even_stage_split stand-in.
+from __future__ import annotations
+
+import json
+import os
+import pathlib
+import random
+import sys
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+REPO_ROOT = pathlib.Path(__file__).resolve().parent.parent
+
+
+@pytest.fixture(scope="module")
+def pt():
+    from oobleck_amd.planning import load
+    return load()
+
+
+@pytest.fixture(scope="module")
+def profile(pt):
+    # the reference's dummy profile shape (tests/conftest.py:119-142)
+    random.seed(3)
+    results = [pt.LayerExecutionResult(
+        layer_index=i, forward=random.random(), backward=random.random() * 3,
+        allreduce_in_node={j + 1: random.random() for j in range(8)},
+        allreduce_across_nodes={j + 1: random.random() * 4
+                                for j in range(64)},
+        mem_required=(1024, 1024)) for i in range(14)]
+    return pt.LayerExecutionResults(results)
+
+
+def test_create_templates_onegpu(pt, profile):
+    gen = pt.PipelineTemplateGenerator()
+    ts = gen.create_pipeline_templates(profile, (1, 1), 1)
+    assert len(ts) == 1
+    assert ts[0]._num_nodes == 1 and ts[0]._num_gpus_per_node == 1
+    assert len(ts[0].get_stages()) == 1
+    assert ts[0]._iteration_time > 0
+
+
+def test_create_templates_maxnode(pt, profile):
+    gen = pt.PipelineTemplateGenerator()
+    n = profile.size
+    ts = gen.create_pipeline_templates(profile, (n, n), 1)
+    assert len(ts) == 1
+    assert ts[0]._num_nodes == n and len(ts[0].get_stages()) == n
+    assert ts[0]._iteration_time > 0
+
+
+def test_create_templates_too_many_nodes(pt, profile):
+    gen = pt.PipelineTemplateGenerator()
+    n = profile.size + 1
+    assert gen.create_pipeline_templates(profile, (n, n), 1) == []
+
+
+def test_create_templates_node_range(pt, profile):
+    gen = pt.PipelineTemplateGenerator()
+    ts = gen.create_pipeline_templates(profile, (2, 8), 1)
+    assert 0 < len(ts) <= profile.size
+    for t in ts:
+        assert t._num_gpus_per_node == 1
+        assert 2 <= len(t.get_stages()) <= 8
+        assert t._iteration_time > 0
+
+
+def test_create_templates_multi_gpu_node(pt, profile):
+    gen = pt.PipelineTemplateGenerator()
+    ts = gen.create_pipeline_templates(profile, (1, 1), 4)
+    assert len(ts) >= 1
+    assert all(t._num_gpus_per_node == 4 for t in ts)
+
+
+def test_create_templates_multi_gpu_node_range(pt, profile):
+    gen = pt.PipelineTemplateGenerator()
+    ts = gen.create_pipeline_templates(profile, (1, 6), 4)
+    assert len(ts) >= 1
+    for i, t in enumerate(ts):
+        assert t._num_gpus_per_node == 4
+        assert t._num_nodes == i + 1
+
+
+def test_rank_grid_semantics(pt, profile):
+    """pipeline_template.h:57-84: stages consume ranks off the front; a
+    stage holding fewer GPUs than num_gpus_per_node repeats each rank so
+    every layer's list is num_gpus_per_node long (fsdp slots)."""
+    gen = pt.PipelineTemplateGenerator()
+    ts = gen.create_pipeline_templates(profile, (4, 4), 1)
+    grid = ts[0].get_rank_grid([10, 11, 12, 13])
+    assert sorted(grid) == list(range(14))
+    assert all(len(v) == 1 for v in grid.values())
+    # layers are covered contiguously, stage by stage
+    seen = [grid[i][0] for i in range(14)]
+    assert seen == sorted(seen) and set(seen) == {10, 11, 12, 13}
+
+    ts4 = gen.create_pipeline_templates(profile, (1, 1), 4)
+    stages4 = ts4[0].get_stages()
+    grid4 = ts4[0].get_rank_grid([0, 1, 2, 3])
+    for v in grid4.values():
+        assert len(v) == 4  # fsdp-slot lists padded by repetition
+    for s in stages4:
+        rep = 4 // s._num_gpus
+        for lid in s._layer_indices:
+            lst = grid4[lid]
+            assert all(lst[i * rep:(i + 1) * rep] == [lst[i * rep]] * rep
+                       for i in range(s._num_gpus))
+
+
+def test_get_profile_results_json(pt, tmp_path, monkeypatch):
+    """get_profile_results loads the profiler's JSON cache layout
+    (pipeline_template.cpp:29-80; profiler.py:290-319 writes it)."""
+    base = pathlib.Path("/tmp/oobleck/profiles/gpt2-test")
+    base.mkdir(parents=True, exist_ok=True)
+    n = 4
+    mb = [{"forward": 1.0 + i, "backward": 3.0 + i,
+           "mem_required": [128, 256]} for i in range(n)]
+    ar_in = [{str(j + 1): 0.1 * (j + 1) for j in range(8)} for _ in range(n)]
+    ar_across = [{str(j + 1): 0.2 * (j + 1) for j in range(8)}
+                 for _ in range(n)]
+    (base / "mb8.json").write_text(json.dumps(mb))
+    (base / "allreduce_in_node.json").write_text(json.dumps(ar_in))
+    (base / "allreduce_across_nodes.json").write_text(json.dumps(ar_across))
+    lers = pt.get_profile_results("gpt2", "test", 8)
+    assert lers.size == n
+    assert lers.at(1)._forward == 2.0
+    assert lers.at(2)._backward == 5.0
+    assert lers.at(0)._allreduce_in_node[2] == pytest.approx(0.2)
+    assert tuple(lers.at(3)._mem_required) == (128, 256)
+
+
+# ---------------------------------------------------------------------------
+# integration: the 4-stage x 2-replica world-8 drill off the REAL planner
+# rank grid (replacing round 1's even_stage_split stand-in)
+# ---------------------------------------------------------------------------
+
+TINY = dict(n_embd=96, n_head=4, n_layer=3, n_positions=64, vocab_size=211)
+B, S, MB = 2, 32, 2
+
+
+def _profile_for_oracle(pt, oc):
+    """A profile whose relative costs mirror the tiny oracle model:
+    embedding cheap, blocks equal, final (lm_head+CE) heavy."""
+    results = []
+    H, V = oc.n_embd, oc.vocab_size
+    for i in range(oc.n_layers_total):
+        kind = oc.layer_kind(i)
+        fwd = 0.01 if kind == 0 else (1.0 if kind == 1 else V * H / (12 * H * H))
+        results.append(pt.LayerExecutionResult(
+            layer_index=i, forward=fwd, backward=2 * fwd,
+            allreduce_in_node={j + 1: 0.01 for j in range(8)},
+            allreduce_across_nodes={j + 1: 0.04 for j in range(8)},
+            mem_required=(1024, 1024)))
+    return pt.LayerExecutionResults(results)
+
+
+def _run_4x2_planner(rank: int, world: int, tmp: str, grids_blob: str):
+    if str(REPO_ROOT) not in sys.path:
+        sys.path.insert(0, str(REPO_ROOT))
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    dist.init_process_group("gloo", init_method=f"file://{tmp}/rdzv",
+                            rank=rank, world_size=world)
+    from oobleck_amd.config import ModelConfig, TrainingConfig
+    from oobleck_amd.engine import DataParallelEngine
+    from oobleck_amd.pipeline import OobleckPipeline
+    from oracle.gpt2_oracle import OracleConfig, init_layer_params
+    from oracle.gpt2_oracle import stage_forward_backward
+    from tests.oracle_layer import NoOpOptimizer, OracleLayer
+
+    grids = [{int(k): v for k, v in g.items()}
+             for g in json.loads(grids_blob)]
+    mc, oc = ModelConfig(**TINY), OracleConfig(**TINY)
+    tc = TrainingConfig(microbatch_size=B,
+                        global_microbatch_size=B * MB * 2, seq_len=S)
+    flats = [init_layer_params(oc, oc.layer_kind(i), 42 * 1000 + i)
+             for i in range(oc.n_layers_total)]
+
+    def loader_for(pid):
+        class Loader:
+            def __iter__(self):
+                g = torch.Generator().manual_seed(7 + pid)
+                def gen():
+                    for _ in range(MB):
+                        ids = torch.randint(0, oc.vocab_size, (B, S),
+                                            generator=g)
+                        yield {"input_ids": ids, "labels": ids.clone()}
+                return gen()
+        return Loader()
+
+    pipelines, my_pipeline = [], None
+    for pid, grid in enumerate(grids):
+        p = OobleckPipeline(pid, grid, mc, tc, loader_for(pid), MB,
+                            torch.device("cpu"))
+        p.initialize_distributed_fsdp()
+        p.initialize_distributed_pipeline()
+        pipelines.append(p)
+    for p in pipelines:
+        if p.my_pipeline:
+            p.initialize_execution(
+                layer_factory=lambda lid, pg, n_slots: OracleLayer(
+                    lid, oc, flats[lid]),
+                optimizer_factory=lambda layers: (NoOpOptimizer(layers),
+                                                  None))
+            my_pipeline = p
+    dp = DataParallelEngine(pipelines)
+    my_pipeline.train()
+    dp.do_allreduce(my_pipeline)
+
+    def ref_grads(seed):
+        grads = [torch.zeros_like(f) for f in flats]
+        g = torch.Generator().manual_seed(seed)
+        for _ in range(MB):
+            ids = torch.randint(0, oc.vocab_size, (B, S), generator=g)
+            _, _, gs = stage_forward_backward(
+                oc, flats, list(range(oc.n_layers_total)), ids,
+                labels=ids.clone())
+            for acc, gi in zip(grads, gs):
+                acc += gi
+        return grads
+
+    g0, g1 = ref_grads(7), ref_grads(8)
+    for layer in my_pipeline.execution._layers:
+        ref = g0[layer.layer_id] + g1[layer.layer_id]
+        torch.testing.assert_close(layer.flat_grad, ref, rtol=1e-4,
+                                   atol=1e-5)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_4x2_drill_off_planner_rank_grid(pt, tmp_path):
+    """Build the 4-node template with the real divide-and-conquer, take
+    its get_rank_grid for both replicas' rank blocks (instantiator.py:
+    118-149's consumption pattern), and run the full world-8 drill."""
+    from oracle.gpt2_oracle import OracleConfig
+    oc = OracleConfig(**TINY)
+    gen = pt.PipelineTemplateGenerator()
+    ts = gen.create_pipeline_templates(_profile_for_oracle(pt, oc), (4, 4), 1)
+    assert len(ts) == 1 and len(ts[0].get_stages()) == 4
+    grids = [ts[0].get_rank_grid([0, 1, 2, 3]),
+             ts[0].get_rank_grid([4, 5, 6, 7])]
+    blob = json.dumps([{str(k): v for k, v in g.items()} for g in grids])
+    mp.spawn(_run_4x2_planner, args=(8, str(tmp_path), blob), nprocs=8,
+             join=True)
