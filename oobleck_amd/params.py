@@ -50,8 +50,14 @@ def init_layer_params(cfg: ModelConfig, layer_id: int, seed: int,
 
     style="gpt2": HF-like init (normal 0.02 weights, zero biases, unit LN
     weights).  style="reference": uniform [0,1) for every tensor, as the
-    reference's init_tensors does (layer.py:26-37, torch.rand)."""
+    reference's init_tensors does (layer.py:26-37, torch.rand).
+    style="zeros": all-zero buffer with NO host-side random generation —
+    for layers whose params arrive by reconfiguration broadcast
+    (engine.py:283-299), where a CPU random init is pure wasted wall-clock
+    on the <2 s recovery path."""
     kind = cfg.layer_kind(layer_id)
+    if style == "zeros":
+        return torch.zeros(layer_param_numel(cfg, kind))
     g = torch.Generator().manual_seed(seed * 1000 + layer_id)
     flats = []
     for name, shape in layer_param_spec(cfg, kind):

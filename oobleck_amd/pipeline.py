@@ -132,9 +132,19 @@ class PipelineExecution:
             import contextlib
             ctx = (torch.cuda.stream(self._s_fwd) if self._overlap
                    else contextlib.nullcontext())
+            ids_cpu, labels_cpu = batch["input_ids"], batch["labels"]
+            if self._overlap and dev.type == "cuda":
+                # pageable H2D on a non-default stream makes the HIP
+                # runtime stage the copy SYNCHRONOUSLY against the
+                # stream's backlog (the round-1 pp1-overlap stall);
+                # pin first so the copy is truly async (torch's caching
+                # host allocator keeps the pinned source alive until the
+                # stream consumes it).
+                ids_cpu = ids_cpu.pin_memory()
+                labels_cpu = labels_cpu.pin_memory()
             with ctx:
-                ids = batch["input_ids"].to(dev, non_blocking=True)
-                labels = batch["labels"].to(dev, non_blocking=True)
+                ids = ids_cpu.to(dev, non_blocking=True)
+                labels = labels_cpu.to(dev, non_blocking=True)
             self.pipeline.pipe_buffers["inputs"][buffer_id] = (ids, labels)
 
     def forward_pass(self, buffer_id: int) -> None:

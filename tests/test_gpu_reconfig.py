@@ -57,7 +57,13 @@ def _drill(rank: int, world: int, tmp: str, out_json: str):
                         global_microbatch_size=B * MB)
     L = mc.n_layers_total
 
-    def build_pipelines(ranks_lists, my_layers_prev):
+    def build_pipelines(ranks_lists, my_layers_prev, initial=False):
+        """my_layers_prev: Layer OBJECTS this rank already holds — reused
+        as-is (the reference's reconfiguration keeps existing layers and
+        only copies into / creates newly-acquired ones, engine.py:238-309).
+        Missing layers (non-initial): init_style='zeros' — their params
+        arrive by broadcast, so a CPU random init is wasted recovery
+        wall-clock."""
         pipelines, my_pipeline = [], None
         my_layers: dict[int, Layer] = {}
         for pid, ranks in enumerate(ranks_lists):
@@ -77,12 +83,13 @@ def _drill(rank: int, world: int, tmp: str, out_json: str):
             p.initialize_distributed_pipeline()
             if p.my_pipeline:
                 def factory(lid, pg, n_slots):
-                    layer = Layer(lid, mc, B, S, n_slots, dev, dtype="bf16",
-                                  seed=7 + lid)
                     prev = my_layers_prev.get(lid)
-                    if prev is not None:
-                        layer.flat_param.copy_(prev)
-                        layer.refresh_weights()
+                    if prev is not None and prev._desc.n_slots >= n_slots:
+                        my_layers[lid] = prev
+                        return prev
+                    layer = Layer(lid, mc, B, S, n_slots, dev, dtype="bf16",
+                                  seed=7 + lid,
+                                  init_style="gpt2" if initial else "zeros")
                     my_layers[lid] = layer
                     return layer
                 def opt_factory(layers):
@@ -98,7 +105,8 @@ def _drill(rank: int, world: int, tmp: str, out_json: str):
     dist.init_process_group("gloo", init_method=f"file://{tmp}/rdzv1",
                             rank=rank, world_size=world)
     old_ranks_lists = [[0, 1], [2]]
-    pipelines, my_pipeline, my_layers = build_pipelines(old_ranks_lists, {})
+    pipelines, my_pipeline, my_layers = build_pipelines(old_ranks_lists, {},
+                                                        initial=True)
     old_grids = [p.rank_grid for p in pipelines]
     # rank-dependent marker on each rank's own params, so the later
     # broadcast is verifiable: rank 0's acquired layers must carry RANK
@@ -130,10 +138,9 @@ def _drill(rank: int, world: int, tmp: str, out_json: str):
     new_ranks_re = [[remap[r] for r in rl] for rl in new_ranks]
     old_grids_re = [{lid: [remap.get(r, -1) for r in rs]
                      for lid, rs in g.items()} for g in old_grids]
-    # keep survivors' existing layer params across the rebuild
-    prev = {lid: layer.flat_param.detach().clone()
-            for lid, layer in my_layers.items()}
-    pipelines2, my_pipeline2, my_layers2 = build_pipelines(new_ranks_re, prev)
+    # survivors keep their existing Layer objects (stash + params intact)
+    pipelines2, my_pipeline2, my_layers2 = build_pipelines(new_ranks_re,
+                                                           my_layers)
     dp2 = DataParallelEngine(pipelines2)
     torch.cuda.synchronize()
     t_rebuild = time.perf_counter()
