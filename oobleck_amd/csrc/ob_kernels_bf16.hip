@@ -284,6 +284,45 @@ extern "C" int ob_gemm_bf16(int transA, int transB, int64_t M, int64_t N,
     return ob_fail("gemm_bf16: splitk needs atomic f32 out");
   // 16-byte staging requires 8-half-aligned leading dims and bases
   if ((lda | ldb) & 7) return ob_fail("gemm_bf16: lda/ldb must be 8-aligned");
+  // probe hook (tools/fc_probe.py): OB_BF16_FORCE picks one NT kernel
+  // unconditionally ("glds" | "n128" | "n256" | "8ph"); read per call so
+  // one process can sweep variants.
+  const char* force = getenv("OB_BF16_FORCE");
+  if (force && transA == 0 && transB == 1) {
+    const int64_t nbm = (M + 255) / 256;
+    if (force[0] == 'g')
+      return ob_gemm_bf16_nt_dispatch(A, B, C, bias, residual, M, N, K, lda,
+                                      ldb, ldc, strideA1, strideA2, strideB1,
+                                      strideB2, strideC1, strideC2, n1, n2,
+                                      alpha, beta, out_kind, splitk, stream,
+                                      M);
+    if (force[0] == 'n') {
+      extern int ob_gemm_bf16_nt256_dispatch(
+          const void* A, const void* B, void* C, const void* bias,
+          const void* residual, int64_t M, int64_t N, int64_t K, int64_t lda,
+          int64_t ldb, int64_t ldc, int64_t sA1, int64_t sA2, int64_t sB1,
+          int64_t sB2, int64_t sC1, int64_t sC2, int64_t n1, int64_t n2,
+          float alpha, float beta, int out_kind, int splitk, void* stream,
+          int64_t Mr, int BN);
+      return ob_gemm_bf16_nt256_dispatch(
+          A, B, C, bias, residual, M, N, K, lda, ldb, ldc, strideA1,
+          strideA2, strideB1, strideB2, strideC1, strideC2, n1, n2, alpha,
+          beta, out_kind, splitk, stream, M, force[1] == '2' ? 256 : 128);
+    }
+    if (force[0] == '8') {
+      extern int ob_gemm_bf16_nt_8ph(
+          const void* A, const void* B, void* C, const void* bias,
+          const void* residual, int64_t M, int64_t N, int64_t K, int64_t lda,
+          int64_t ldb, int64_t ldc, int64_t sA1, int64_t sA2, int64_t sB1,
+          int64_t sB2, int64_t sC1, int64_t sC2, int64_t n1, int64_t n2,
+          float alpha, float beta, int out_kind, int splitk, void* stream,
+          int64_t Mr);
+      return ob_gemm_bf16_nt_8ph(A, B, C, bias, residual, M, N, K, lda, ldb,
+                                 ldc, strideA1, strideA2, strideB1, strideB2,
+                                 strideC1, strideC2, n1, n2, alpha, beta,
+                                 out_kind, splitk, stream, M);
+    }
+  }
   // plain GEMMs (no bias/residual/beta, unbatched, bf16 out) go to
   // hipBLASLt — the dX family, the bias-free lm_head family (measured
   // 815-1166 TF vs 436-692 for the hand-written kernels on those
@@ -2191,6 +2230,130 @@ __global__ __launch_bounds__(256, MAXB) void k_flash_bwd_dkdv(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Paired-wave dKdV (the round-2 register diet): the
+// 2-wave-occupancy kernel above holds K+V residents (32 VGPR), BOTH the
+// S and dP accumulators (32) and BOTH d-halves of dV/dK (64) per wave —
+// 217 VGPRs, 2 waves/SIMD.  Here a WAVE PAIR shares one 32-kv-row tile:
+//   role 0: K resident, computes the S tile (Q K^T), owns d-half [0,32)
+//   role 1: V resident, computes the dP tile (dO V^T), owns d-half [32,64)
+// The S/dP tiles are exchanged through double-buffered LDS (one
+// __syncthreads per q-tile), so each wave holds HALF of everything:
+// ~140 VGPRs -> 3 waves/SIMD, same MFMA count per kv row, 2x the blocks
+// (64 kv rows per block).  The causal q-loop start is block-uniform; the
+// sub-diagonal tile pair-1 wastes is masked per element (p = 0).
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ int acc_row(int r, int kh) {
+  return (r & 3) + 8 * (r >> 2) + 4 * kh;
+}
+
+__global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_p(
+    const __bf16* __restrict__ qkv, const __bf16* __restrict__ QT,
+    const __bf16* __restrict__ dOT, const __bf16* __restrict__ dO,
+    const float* __restrict__ lse, const float* __restrict__ D,
+    __bf16* __restrict__ dqkv, int Sq, int H, int nh, float scale) {
+  // [buffer][pair][role][32 rows x 32 cols]
+  __shared__ float xch[2][2][2][32 * 32];
+  const int z = blockIdx.z;
+  const int b = z / nh, h = z % nh;
+  const int64_t base = (int64_t)b * Sq * 3 * H + h * 64;
+  const __bf16* Qp = qkv + base;
+  const __bf16* Kp = Qp + H;
+  const __bf16* Vp = Qp + 2 * H;
+  const __bf16* dOp = dO + (int64_t)b * Sq * H + h * 64;
+  const __bf16* QTp = QT + (int64_t)z * 64 * Sq;
+  const __bf16* dOTp = dOT + (int64_t)z * 64 * Sq;
+  const float* lsep = lse + (int64_t)z * Sq;
+  const float* Dp = D + (int64_t)z * Sq;
+
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int pair = w >> 1, role = w & 1;
+  const int il = lane & 31, kh = lane >> 5;
+  const int kv0 = blockIdx.x * 64 + pair * 32;
+  const int mykv = kv0 + il;
+
+  // one resident operand per role (K for S, V for dP)
+  const __bf16* KV = role ? Vp : Kp;
+  bf16x8 of[4];
+#pragma unroll
+  for (int s = 0; s < 4; ++s)
+    of[s] = *reinterpret_cast<const bf16x8*>(
+        KV + (int64_t)mykv * 3 * H + s * 16 + kh * 8);
+
+  f32x16 dvh = {}, dkh = {};
+  const int nqt = Sq / 32;
+  for (int qt = (blockIdx.x * 64) / 32; qt < nqt; ++qt) {
+    const int q0 = qt * 32;
+    const float lse_t = lsep[q0 + il];
+    const float d_t = Dp[q0 + il];
+    // own tile: S (role 0) or dP (role 1)
+    f32x16 own = {};
+    if (role == 0) {
+#pragma unroll
+      for (int s = 0; s < 4; ++s) {
+        const bf16x8 af = *reinterpret_cast<const bf16x8*>(
+            Qp + (int64_t)(q0 + il) * 3 * H + s * 16 + kh * 8);
+        own = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, of[s], own, 0, 0,
+                                                      0);
+      }
+    } else {
+#pragma unroll
+      for (int s = 0; s < 4; ++s) {
+        const bf16x8 df = *reinterpret_cast<const bf16x8*>(
+            dOp + (int64_t)(q0 + il) * H + s * 16 + kh * 8);
+        own = __builtin_amdgcn_mfma_f32_32x32x16_bf16(df, of[s], own, 0, 0,
+                                                      0);
+      }
+    }
+    // exchange: write own tile, barrier, read the partner's
+    float* mine = xch[qt & 1][pair][role];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) mine[acc_row(r, kh) * 32 + il] = own[r];
+    __syncthreads();
+    const float* theirs = xch[qt & 1][pair][role ^ 1];
+    float other[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) other[r] = theirs[acc_row(r, kh) * 32 + il];
+
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      float pv[8], dsv[8];
+#pragma unroll
+      for (int r8 = 0; r8 < 8; ++r8) {
+        const int r = t * 8 + r8;
+        const int qoff = acc_row(r, kh);
+        const bool ok = q0 + qoff >= mykv;
+        const float sv = role == 0 ? own[r] : other[r];
+        const float dpv = role == 0 ? other[r] : own[r];
+        const float lse_q = __shfl(lse_t, qoff, 64);
+        const float d_q = __shfl(d_t, qoff, 64);
+        const float p = ok ? __expf(sv * scale - lse_q) : 0.f;
+        pv[r8] = p;
+        dsv[r8] = ok ? p * (dpv - d_q) : 0.f;
+      }
+      const bf16x8 pa = bf_dance(pv);
+      const bf16x8 da = bf_dance(dsv);
+      const bf16x8 dof = *reinterpret_cast<const bf16x8*>(
+          dOTp + (int64_t)(role * 32 + il) * Sq + q0 + t * 16 + kh * 8);
+      const bf16x8 qtf = *reinterpret_cast<const bf16x8*>(
+          QTp + (int64_t)(role * 32 + il) * Sq + q0 + t * 16 + kh * 8);
+      dvh = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, dof, dvh, 0, 0, 0);
+      dkh = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, qtf, dkh, 0, 0, 0);
+    }
+  }
+
+  __bf16* dKp = dqkv + base + H;
+  __bf16* dVp = dqkv + base + 2 * H;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int kv = kv0 + acc_row(r, kh);
+    dVp[(int64_t)kv * 3 * H + role * 32 + il] = (__bf16)dvh[r];
+    dKp[(int64_t)kv * 3 * H + role * 32 + il] = (__bf16)(scale * dkh[r]);
+  }
+}
+
 __global__ __launch_bounds__(256, 2) void k_flash_bwd_dq(
     const __bf16* __restrict__ qkv, const __bf16* __restrict__ KT,
     const __bf16* __restrict__ dO, const float* __restrict__ lse,
@@ -2292,11 +2455,26 @@ extern "C" int ob_flash_bwd_bf16(const void* qkv, const void* QT,
     OB_LAUNCH_CHECK();
     return 0;
   }
-  k_flash_bwd_dkdv<2><<<grid, 256, 0, S(stream)>>>(
-      (const __bf16*)qkv, (const __bf16*)QT, (const __bf16*)dOT,
-      (const __bf16*)dO, (const float*)lse, (const float*)D, (__bf16*)dqkv,
-      (int)Sq, (int)H, (int)nh, scale);
-  OB_LAUNCH_CHECK();
+  // default: the paired-wave 3-occupancy kernel (64 kv rows/block);
+  // OB_FLASH_PAIR=0 reverts to the round-1 2-occupancy kernel
+  static const bool pair_off = [] {
+    const char* e = getenv("OB_FLASH_PAIR");
+    return e && e[0] == '0';
+  }();
+  if (!pair_off) {
+    dim3 gridp((unsigned)(Sq / 64), 1, (unsigned)(B * nh));
+    k_flash_bwd_dkdv_p<<<gridp, 256, 0, S(stream)>>>(
+        (const __bf16*)qkv, (const __bf16*)QT, (const __bf16*)dOT,
+        (const __bf16*)dO, (const float*)lse, (const float*)D,
+        (__bf16*)dqkv, (int)Sq, (int)H, (int)nh, scale);
+    OB_LAUNCH_CHECK();
+  } else {
+    k_flash_bwd_dkdv<2><<<grid, 256, 0, S(stream)>>>(
+        (const __bf16*)qkv, (const __bf16*)QT, (const __bf16*)dOT,
+        (const __bf16*)dO, (const float*)lse, (const float*)D, (__bf16*)dqkv,
+        (int)Sq, (int)H, (int)nh, scale);
+    OB_LAUNCH_CHECK();
+  }
   k_flash_bwd_dq<<<grid, 256, 0, S(stream)>>>(
       (const __bf16*)qkv, (const __bf16*)KT, (const __bf16*)dO,
       (const float*)lse, (const float*)D, (__bf16*)dqkv, (int)Sq, (int)H,
