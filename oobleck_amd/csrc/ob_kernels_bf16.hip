@@ -2455,12 +2455,11 @@ extern "C" int ob_flash_bwd_bf16(const void* qkv, const void* QT,
     OB_LAUNCH_CHECK();
     return 0;
   }
-  // default: the paired-wave 3-occupancy kernel (64 kv rows/block);
-  // OB_FLASH_PAIR=0 reverts to the round-1 2-occupancy kernel
-  static const bool pair_off = [] {
-    const char* e = getenv("OB_FLASH_PAIR");
-    return e && e[0] == '0';
-  }();
+  // default: the paired-wave 4-occupancy kernel (64 kv rows/block);
+  // OB_FLASH_PAIR=0 reverts to the round-1 2-occupancy kernel (read
+  // per call so tools/flash_probe.py can A/B in one process)
+  const char* pe = getenv("OB_FLASH_PAIR");
+  const bool pair_off = pe && pe[0] == '0';
   if (!pair_off) {
     dim3 gridp((unsigned)(Sq / 64), 1, (unsigned)(B * nh));
     k_flash_bwd_dkdv_p<<<gridp, 256, 0, S(stream)>>>(
