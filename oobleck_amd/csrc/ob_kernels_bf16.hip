@@ -1858,6 +1858,158 @@ __device__ __forceinline__ unsigned bf_pk2(float a, float b) {
   return bf_bits((__bf16)a) | (bf_bits((__bf16)b) << 16);
 }
 
+// pack-and-exchange: acc-held values (rows pattern) -> A-fragment k-runs
+__device__ __forceinline__ bf16x8 bf_dance(const float* pv8) {
+  unsigned x0 = bf_pk2(pv8[0], pv8[1]);
+  unsigned x1 = bf_pk2(pv8[2], pv8[3]);
+  unsigned y0 = bf_pk2(pv8[4], pv8[5]);
+  unsigned y1 = bf_pk2(pv8[6], pv8[7]);
+  {
+    auto r2 = __builtin_amdgcn_permlane32_swap(x0, y0, false, false);
+    x0 = r2[0];
+    y0 = r2[1];
+  }
+  {
+    auto r2 = __builtin_amdgcn_permlane32_swap(x1, y1, false, false);
+    x1 = r2[0];
+    y1 = r2[1];
+  }
+  uint4 fr;
+  fr.x = x0;
+  fr.y = x1;
+  fr.z = y0;
+  fr.w = y1;
+  return __builtin_bit_cast(bf16x8, fr);
+}
+
+// ---------------------------------------------------------------------------
+// Software-pipelined flash forward (round 2): the kernel above serializes
+// per tile (K loads -> S MFMAs -> softmax -> V^T loads -> PV MFMAs, nothing
+// in flight across tiles — measured 15x off the MFMA floor, and occupancy
+// alone did not move it: the stall is load latency on the critical path).
+// This variant double-buffers the K fragments (prefetch tile t+1 during
+// tile t's compute) and hoists the V^T fragment loads to the top of the
+// tile (consumed last, ~full-tile latency cover).  Per-wave diagonal loop
+// bound replaces the wave-uniform `continue`.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256, 2) void k_flash_fwd_bf16_pf(
+    const __bf16* __restrict__ qkv, const __bf16* __restrict__ VT,
+    __bf16* __restrict__ Obase, float* __restrict__ lse, int Sq, int H,
+    int nh, float scale) {
+  __shared__ float bcast[128];
+  const int z = blockIdx.z;
+  const int b = z / nh, h = z % nh;
+  const int64_t qoff = (int64_t)b * Sq * 3 * H + h * 64;
+  const __bf16* Qp = qkv + qoff;
+  const __bf16* Kp = Qp + H;
+  const __bf16* VTp = VT + (int64_t)z * 64 * Sq;
+  __bf16* Op = Obase + (int64_t)b * Sq * H + h * 64;
+  float* lsep = lse + (int64_t)z * Sq;
+
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int il = lane & 31, kh = lane >> 5;
+  const int q0 = blockIdx.x * 128 + w * 32;
+  const int myq = q0 + il;
+
+  bf16x8 qf[4];
+#pragma unroll
+  for (int s = 0; s < 4; ++s)
+    qf[s] = *reinterpret_cast<const bf16x8*>(
+        Qp + (int64_t)myq * 3 * H + s * 16 + kh * 8);
+
+  f32x16 o0 = {}, o1 = {};
+  float m = -INFINITY, l = 0.f;
+  const int my_ntiles = (q0 + 32) / 32;  // per-wave causal bound
+
+  bf16x8 kfA[4], kfB[4];
+#pragma unroll
+  for (int s = 0; s < 4; ++s)
+    kfA[s] = *reinterpret_cast<const bf16x8*>(
+        Kp + (int64_t)il * 3 * H + s * 16 + kh * 8);
+
+  auto tile = [&](int kvt, bf16x8 (&kcur)[4], bf16x8 (&knxt)[4]) {
+    const int kv0 = kvt * 32;
+    // V^T fragments for THIS tile first (consumed last)
+    const bf16x8 v00 = *reinterpret_cast<const bf16x8*>(
+        VTp + (int64_t)il * Sq + kv0 + kh * 8);
+    const bf16x8 v01 = *reinterpret_cast<const bf16x8*>(
+        VTp + (int64_t)(32 + il) * Sq + kv0 + kh * 8);
+    const bf16x8 v10 = *reinterpret_cast<const bf16x8*>(
+        VTp + (int64_t)il * Sq + kv0 + 16 + kh * 8);
+    const bf16x8 v11 = *reinterpret_cast<const bf16x8*>(
+        VTp + (int64_t)(32 + il) * Sq + kv0 + 16 + kh * 8);
+    // prefetch NEXT tile's K fragments
+    if (kvt + 1 < my_ntiles) {
+#pragma unroll
+      for (int s = 0; s < 4; ++s)
+        knxt[s] = *reinterpret_cast<const bf16x8*>(
+            Kp + (int64_t)(kv0 + 32 + il) * 3 * H + s * 16 + kh * 8);
+    }
+    f32x16 sacc = {};
+#pragma unroll
+    for (int s = 0; s < 4; ++s)
+      sacc =
+          __builtin_amdgcn_mfma_f32_32x32x16_bf16(kcur[s], qf[s], sacc, 0,
+                                                  0, 0);
+    float sv[16];
+    float mt = -INFINITY;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int kv = kv0 + (r & 3) + 8 * (r >> 2) + 4 * kh;
+      sv[r] = (kv <= myq) ? sacc[r] * scale : -INFINITY;
+      mt = fmaxf(mt, sv[r]);
+    }
+    mt = fmaxf(mt, __shfl_xor(mt, 32, 64));
+    const float mnew = fmaxf(m, mt);
+    const float af = __expf(m - mnew);
+    m = mnew;
+    float psum = 0.f;
+    float pv[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      pv[r] = (sv[r] == -INFINITY) ? 0.f : __expf(sv[r] - mnew);
+      psum += pv[r];
+    }
+    psum += __shfl_xor(psum, 32, 64);
+    l = l * af + psum;
+    if (kh == 0) bcast[w * 32 + il] = af;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const float a = bcast[w * 32 + (r & 3) + 8 * (r >> 2) + 4 * kh];
+      o0[r] *= a;
+      o1[r] *= a;
+    }
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      const bf16x8 pa = bf_dance(pv + t * 8);
+      const bf16x8 v0 = t ? v10 : v00;
+      const bf16x8 v1 = t ? v11 : v01;
+      o0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, v0, o0, 0, 0, 0);
+      o1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, v1, o1, 0, 0, 0);
+    }
+  };
+
+  for (int kvt = 0; kvt < my_ntiles; kvt += 2) {
+    tile(kvt, kfA, kfB);
+    if (kvt + 1 < my_ntiles) tile(kvt + 1, kfB, kfA);
+  }
+
+  // epilogue: O /= l ; lse = m + log(l)  (identical to the kernel below:
+  // acc rows = q offsets, cols = il = d)
+  if (kh == 0) {
+    bcast[w * 32 + il] = 1.f / l;
+    lsep[myq] = m + __logf(l);
+  }
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = (r & 3) + 8 * (r >> 2) + 4 * kh;
+    const float inv = bcast[w * 32 + row];
+    Op[(int64_t)(q0 + row) * H + il] = (__bf16)(o0[r] * inv);
+    Op[(int64_t)(q0 + row) * H + 32 + il] = (__bf16)(o1[r] * inv);
+  }
+}
+
 __global__ __launch_bounds__(256, 2) void k_flash_fwd_bf16(
     const __bf16* __restrict__ qkv, const __bf16* __restrict__ VT,
     __bf16* __restrict__ Obase, float* __restrict__ lse, int Sq, int H,
@@ -1988,9 +2140,16 @@ extern "C" int ob_flash_fwd_bf16(const void* qkv, const void* VT, void* O,
   if (H / nh != 64) return ob_fail("flash_fwd: head_dim must be 64");
   if (Sq % 128) return ob_fail("flash_fwd: S must be a multiple of 128");
   dim3 grid((unsigned)(Sq / 128), 1, (unsigned)(B * nh));
-  k_flash_fwd_bf16<<<grid, 256, 0, S(stream)>>>(
-      (const __bf16*)qkv, (const __bf16*)VT, (__bf16*)O, (float*)lse,
-      (int)Sq, (int)H, (int)nh, scale);
+  // default: software-pipelined variant; OB_FLASH_PF=0 reverts (A/B)
+  const char* pfe = getenv("OB_FLASH_PF");
+  if (pfe && pfe[0] == '0')
+    k_flash_fwd_bf16<<<grid, 256, 0, S(stream)>>>(
+        (const __bf16*)qkv, (const __bf16*)VT, (__bf16*)O, (float*)lse,
+        (int)Sq, (int)H, (int)nh, scale);
+  else
+    k_flash_fwd_bf16_pf<<<grid, 256, 0, S(stream)>>>(
+        (const __bf16*)qkv, (const __bf16*)VT, (__bf16*)O, (float*)lse,
+        (int)Sq, (int)H, (int)nh, scale);
   OB_LAUNCH_CHECK();
   return 0;
 }
@@ -2096,30 +2255,6 @@ extern "C" int ob_flash_dsum_bf16(const void* O, const void* dO, void* D,
       (int)nh);
   OB_LAUNCH_CHECK();
   return 0;
-}
-
-// pack-and-exchange: acc-held values (rows pattern) -> A-fragment k-runs
-__device__ __forceinline__ bf16x8 bf_dance(const float* pv8) {
-  unsigned x0 = bf_pk2(pv8[0], pv8[1]);
-  unsigned x1 = bf_pk2(pv8[2], pv8[3]);
-  unsigned y0 = bf_pk2(pv8[4], pv8[5]);
-  unsigned y1 = bf_pk2(pv8[6], pv8[7]);
-  {
-    auto r2 = __builtin_amdgcn_permlane32_swap(x0, y0, false, false);
-    x0 = r2[0];
-    y0 = r2[1];
-  }
-  {
-    auto r2 = __builtin_amdgcn_permlane32_swap(x1, y1, false, false);
-    x1 = r2[0];
-    y1 = r2[1];
-  }
-  uint4 fr;
-  fr.x = x0;
-  fr.y = x1;
-  fr.z = y0;
-  fr.w = y1;
-  return __builtin_bit_cast(bf16x8, fr);
 }
 
 template <int MAXB>
@@ -2284,29 +2419,43 @@ __global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_p(
 
   f32x16 dvh = {}, dkh = {};
   const int nqt = Sq / 32;
-  for (int qt = (blockIdx.x * 64) / 32; qt < nqt; ++qt) {
+  const int qt0 = (blockIdx.x * 64) / 32;
+  // A-operand rows for this wave's role: Q rows (S tile) or dO rows (dP)
+  const __bf16* Arow = role ? dOp : Qp;
+  const int64_t Astride = role ? H : 3 * H;
+  // software pipeline: fragments + lse/D for tile t+1 prefetched during
+  // tile t's compute (the barrier kept the compiler from doing this)
+  bf16x8 fA[4], fB[4];
+#pragma unroll
+  for (int s = 0; s < 4; ++s)
+    fA[s] = *reinterpret_cast<const bf16x8*>(
+        Arow + (int64_t)(qt0 * 32 + il) * Astride + s * 16 + kh * 8);
+  float lseA = lsep[qt0 * 32 + il], dA = Dp[qt0 * 32 + il];
+  float lseB, dB;
+
+  auto tile = [&](int qt, bf16x8 (&fcur)[4], bf16x8 (&fnxt)[4],
+                  float lse_t, float d_t, float& lse_n, float& d_n) {
     const int q0 = qt * 32;
-    const float lse_t = lsep[q0 + il];
-    const float d_t = Dp[q0 + il];
+    // early: B-operands for this tile's dv/dk (consumed after the barrier)
+    const bf16x8 dof0 = *reinterpret_cast<const bf16x8*>(
+        dOTp + (int64_t)(role * 32 + il) * Sq + q0 + kh * 8);
+    const bf16x8 qtf0 = *reinterpret_cast<const bf16x8*>(
+        QTp + (int64_t)(role * 32 + il) * Sq + q0 + kh * 8);
+    // prefetch tile t+1's A-fragments + lse/D
+    if (qt + 1 < nqt) {
+#pragma unroll
+      for (int s = 0; s < 4; ++s)
+        fnxt[s] = *reinterpret_cast<const bf16x8*>(
+            Arow + (int64_t)(q0 + 32 + il) * Astride + s * 16 + kh * 8);
+      lse_n = lsep[q0 + 32 + il];
+      d_n = Dp[q0 + 32 + il];
+    }
     // own tile: S (role 0) or dP (role 1)
     f32x16 own = {};
-    if (role == 0) {
 #pragma unroll
-      for (int s = 0; s < 4; ++s) {
-        const bf16x8 af = *reinterpret_cast<const bf16x8*>(
-            Qp + (int64_t)(q0 + il) * 3 * H + s * 16 + kh * 8);
-        own = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, of[s], own, 0, 0,
-                                                      0);
-      }
-    } else {
-#pragma unroll
-      for (int s = 0; s < 4; ++s) {
-        const bf16x8 df = *reinterpret_cast<const bf16x8*>(
-            dOp + (int64_t)(q0 + il) * H + s * 16 + kh * 8);
-        own = __builtin_amdgcn_mfma_f32_32x32x16_bf16(df, of[s], own, 0, 0,
-                                                      0);
-      }
-    }
+    for (int s = 0; s < 4; ++s)
+      own = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fcur[s], of[s], own, 0,
+                                                    0, 0);
     // exchange: write own tile, barrier, read the partner's
     float* mine = xch[qt & 1][pair][role];
 #pragma unroll
@@ -2335,13 +2484,22 @@ __global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_p(
       }
       const bf16x8 pa = bf_dance(pv);
       const bf16x8 da = bf_dance(dsv);
-      const bf16x8 dof = *reinterpret_cast<const bf16x8*>(
-          dOTp + (int64_t)(role * 32 + il) * Sq + q0 + t * 16 + kh * 8);
-      const bf16x8 qtf = *reinterpret_cast<const bf16x8*>(
-          QTp + (int64_t)(role * 32 + il) * Sq + q0 + t * 16 + kh * 8);
+      const bf16x8 dof =
+          t ? *reinterpret_cast<const bf16x8*>(
+                  dOTp + (int64_t)(role * 32 + il) * Sq + q0 + 16 + kh * 8)
+            : dof0;
+      const bf16x8 qtf =
+          t ? *reinterpret_cast<const bf16x8*>(
+                  QTp + (int64_t)(role * 32 + il) * Sq + q0 + 16 + kh * 8)
+            : qtf0;
       dvh = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, dof, dvh, 0, 0, 0);
       dkh = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, qtf, dkh, 0, 0, 0);
     }
+  };
+
+  for (int qt = qt0; qt < nqt; qt += 2) {
+    tile(qt, fA, fB, lseA, dA, lseB, dB);
+    if (qt + 1 < nqt) tile(qt + 1, fB, fA, lseB, dB, lseA, dA);
   }
 
   __bf16* dKp = dqkv + base + H;
@@ -2388,20 +2546,45 @@ __global__ __launch_bounds__(256, 2) void k_flash_bwd_dq(
   }
 
   f32x16 dq0 = {}, dq1 = {};
-  const int ntiles = (blockIdx.x * 128 + 128) / 32;
-  for (int kvt = 0; kvt < ntiles; ++kvt) {
+  const int my_nt = (q0 + 32) / 32;  // per-wave causal bound (was continue)
+  // software pipeline: K/V fragments double-buffered across kv-tiles;
+  // K^T fragments hoisted to the tile top (consumed last)
+  bf16x8 kfA[4], vfA[4], kfB[4], vfB[4];
+#pragma unroll
+  for (int s = 0; s < 4; ++s) {
+    kfA[s] = *reinterpret_cast<const bf16x8*>(
+        Kp + (int64_t)il * 3 * H + s * 16 + kh * 8);
+    vfA[s] = *reinterpret_cast<const bf16x8*>(
+        Vp + (int64_t)il * 3 * H + s * 16 + kh * 8);
+  }
+
+  auto tile = [&](int kvt, bf16x8 (&kc)[4], bf16x8 (&vc)[4],
+                  bf16x8 (&kn)[4], bf16x8 (&vn)[4]) {
     const int kv0 = kvt * 32;
-    if (kv0 > q0 + 31) continue;
+    const bf16x8 kt00 = *reinterpret_cast<const bf16x8*>(
+        KTp + (int64_t)il * Sq + kv0 + kh * 8);
+    const bf16x8 kt01 = *reinterpret_cast<const bf16x8*>(
+        KTp + (int64_t)(32 + il) * Sq + kv0 + kh * 8);
+    const bf16x8 kt10 = *reinterpret_cast<const bf16x8*>(
+        KTp + (int64_t)il * Sq + kv0 + 16 + kh * 8);
+    const bf16x8 kt11 = *reinterpret_cast<const bf16x8*>(
+        KTp + (int64_t)(32 + il) * Sq + kv0 + 16 + kh * 8);
+    if (kvt + 1 < my_nt) {
+#pragma unroll
+      for (int s = 0; s < 4; ++s) {
+        kn[s] = *reinterpret_cast<const bf16x8*>(
+            Kp + (int64_t)(kv0 + 32 + il) * 3 * H + s * 16 + kh * 8);
+        vn[s] = *reinterpret_cast<const bf16x8*>(
+            Vp + (int64_t)(kv0 + 32 + il) * 3 * H + s * 16 + kh * 8);
+      }
+    }
     f32x16 sacc = {}, dpacc = {};
 #pragma unroll
     for (int s = 0; s < 4; ++s) {
-      const bf16x8 kf = *reinterpret_cast<const bf16x8*>(
-          Kp + (int64_t)(kv0 + il) * 3 * H + s * 16 + kh * 8);
-      const bf16x8 vf = *reinterpret_cast<const bf16x8*>(
-          Vp + (int64_t)(kv0 + il) * 3 * H + s * 16 + kh * 8);
-      sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[s], sacc, 0, 0, 0);
-      dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, df[s], dpacc, 0, 0,
-                                                      0);
+      sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kc[s], qf[s], sacc, 0,
+                                                     0, 0);
+      dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vc[s], df[s], dpacc, 0,
+                                                      0, 0);
     }
     float dsv[16];
 #pragma unroll
@@ -2414,13 +2597,16 @@ __global__ __launch_bounds__(256, 2) void k_flash_bwd_dq(
 #pragma unroll
     for (int t = 0; t < 2; ++t) {
       const bf16x8 da = bf_dance(dsv + t * 8);
-      const bf16x8 kt0 = *reinterpret_cast<const bf16x8*>(
-          KTp + (int64_t)il * Sq + kv0 + t * 16 + kh * 8);
-      const bf16x8 kt1 = *reinterpret_cast<const bf16x8*>(
-          KTp + (int64_t)(32 + il) * Sq + kv0 + t * 16 + kh * 8);
+      const bf16x8 kt0 = t ? kt10 : kt00;
+      const bf16x8 kt1 = t ? kt11 : kt01;
       dq0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, kt0, dq0, 0, 0, 0);
       dq1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, kt1, dq1, 0, 0, 0);
     }
+  };
+
+  for (int kvt = 0; kvt < my_nt; kvt += 2) {
+    tile(kvt, kfA, vfA, kfB, vfB);
+    if (kvt + 1 < my_nt) tile(kvt + 1, kfB, vfB, kfA, vfA);
   }
   __bf16* dQp = dqkv + base;
 #pragma unroll

@@ -22,7 +22,15 @@ def stream():
     return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
 
 
-def timeit(fn, reps=50):
+_SEQ = ["ln_fwd", "ln_bwd", "gelu_fwd", "gelu_bwd", "colsum_3H", "ce_fwd",
+        "ce_bwd", "flash_dsum", "transpose_b"]
+_IDX = [0]
+
+
+def timeit(fn, reps=50, name=""):
+    label = _SEQ[_IDX[0]] if _IDX[0] < len(_SEQ) else str(_IDX[0])
+    _IDX[0] += 1
+    print(f"-> {label}", flush=True)
     for _ in range(5):
         fn()
     torch.cuda.synchronize()
@@ -39,7 +47,8 @@ def timeit(fn, reps=50):
 
 
 def report(name, ms, bytes_moved):
-    print(f"{name:18s} {ms*1e3:8.1f} us  {bytes_moved/ms/1e6:7.0f} GB/s")
+    print(f"{name:18s} {ms*1e3:8.1f} us  {bytes_moved/ms/1e6:7.0f} GB/s",
+          flush=True)
 
 
 ext = get_ext()

@@ -58,11 +58,14 @@ check(ext.ob_transpose_bf16_b(ptr(qkv) if False else
                               ctypes.c_void_p(qkv.data_ptr() + 2 * H * 2),
                               ptr(VT), S, hd, S * 3 * H, hd, 3 * H, B, nh,
                               stream()))
-ms = timeit(lambda: check(ext.ob_flash_fwd_bf16(
-    ptr(qkv), ptr(VT), ptr(O), ptr(lse), B, S, H, nh, scale, stream())))
 # causal attention fwd flops: 2 matmuls, ~half masked
 fl_fwd = 2 * 2 * B * nh * S * S * hd / 2
-print(f"flash_fwd        {ms*1e3:8.1f} us  {fl_fwd/ms/1e9:6.0f} TF")
+for mode, name in [("1", "pipelined"), ("0", "round1   ")]:
+    os.environ["OB_FLASH_PF"] = mode
+    ms = timeit(lambda: check(ext.ob_flash_fwd_bf16(
+        ptr(qkv), ptr(VT), ptr(O), ptr(lse), B, S, H, nh, scale, stream())))
+    print(f"flash_fwd {name} {ms*1e3:8.1f} us  {fl_fwd/ms/1e9:6.0f} TF")
+os.environ.pop("OB_FLASH_PF", None)
 
 check(ext.ob_transpose_bf16_b(ptr(qkv), ptr(QT), S, hd, S * 3 * H, hd,
                               3 * H, B, nh, stream()))
