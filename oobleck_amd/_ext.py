@@ -60,7 +60,24 @@ def _configure(lib: ctypes.CDLL) -> ctypes.CDLL:
                                  i64, i64, i64, vp, vp, i32, i32, vp]
     lib.ob_f32_to_bf16.argtypes = [vp, vp, i64, vp]
     lib.ob_f32_to_bf16_t.argtypes = [vp, vp, i64, i64, vp]
+    lib.ob_f32_to_bf16_t_ld.argtypes = [vp, vp, i64, i64, i64, vp]
     lib.ob_bf16_to_f32.argtypes = [vp, vp, i64, vp]
+    # every remaining kernel entry point: ctypes without argtypes passes
+    # Python ints as 32-bit c_int, leaving int64 params with undefined
+    # upper halves (the elem_probe/ce_probe GPU faults) — declare them all
+    lib.ob_gelu_fwd_bf16.argtypes = [vp, vp, i64, vp]
+    lib.ob_gelu_bwd_bf16.argtypes = [vp, vp, vp, i64, vp]
+    lib.ob_ce_fwd_bf16.argtypes = [vp, vp, vp, vp, i64, i64, i64, i64, vp]
+    lib.ob_ce_bwd_bf16.argtypes = [vp, vp, vp, vp, i64, i64, i64, i64, vp]
+    lib.ob_embed_fwd_bf16.argtypes = [vp, vp, vp, vp, i64, i64, i64, vp]
+    lib.ob_embed_bwd_bf16.argtypes = [vp, vp, vp, vp, i64, i64, i64, vp]
+    lib.ob_softmax_causal_fwd_bf16.argtypes = [vp, i64, i64, f32, vp]
+    lib.ob_softmax_causal_bwd_bf16.argtypes = [vp, vp, i64, i64, vp]
+    lib.ob_transpose_bf16.argtypes = [vp, vp, i64, i64, vp]
+    lib.ob_gemm_lt_bias.argtypes = [i32, i32, i64, i64, i64, f32, vp, i64,
+                                    vp, i64, f32, vp, i64, i32, vp, vp]
+    lib.ob_gemm_lt_f32.argtypes = [i32, i32, i64, i64, i64, f32, vp, i64,
+                                   vp, i64, f32, vp, i64, vp]
     lib.ob_flash_fwd_bf16.argtypes = [vp, vp, vp, vp, i64, i64, i64, i64,
                                       f32, vp]
     lib.ob_transpose_bf16_b.argtypes = [vp, vp, i64, i64, i64, i64, i64,

@@ -444,12 +444,15 @@ class OobleckPipeline:
                 if handler is None:
                     raise RuntimeError(f"unknown instruction {cmd!r}")
                 if trace:
-                    print(f"[sched] {type(cmd).__name__} buf={cmd.buffer_id}",
+                    import time as _t
+                    print(f"[sched {_t.monotonic():.3f}] "
+                          f"{type(cmd).__name__} buf={cmd.buffer_id}",
                           file=sys.stderr, flush=True)
                 handler(**cmd.kwargs)
                 if trace:
-                    print(f"[sched] done {type(cmd).__name__}",
-                          file=sys.stderr, flush=True)
+                    print(f"[sched {_t.monotonic():.3f}] done "
+                          f"{type(cmd).__name__}", file=sys.stderr,
+                          flush=True)
         if getattr(self.execution, "_overlap", False):
             cur = torch.cuda.current_stream()
             cur.wait_stream(self.execution._s_fwd)
