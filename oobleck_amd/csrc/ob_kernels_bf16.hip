@@ -620,55 +620,88 @@ __global__ __launch_bounds__(256) void k_ln_fwd_bf16_w(
     bv1[j] = has1 ? b[c1 + j] : 0.f;
   }
   const int64_t r0 = (int64_t)blockIdx.x * (4 * wrows) + wid;
-  for (int i = 0; i < wrows; ++i) {
-    const int64_t row = r0 + (int64_t)i * 4;
-    if (row >= rows) return;
-    const __bf16* xr = x + row * H;
-    const uint4 xv0 = *reinterpret_cast<const uint4*>(xr + c0);
-    uint4 xv1 = {};
-    if (has1) xv1 = *reinterpret_cast<const uint4*>(xr + c1);
-    float v0[8], v1[8];
-    float s = 0.f, sq = 0.f;
+  // ROW PAIRS per iteration: the wave shfl reduction ladders of the two
+  // rows interleave (ILP 4 on the shuffle latency) — the single-row
+  // version measured 1.9 TB/s, shuffle-latency-bound, ~3x off roofline
+  for (int i = 0; i < wrows; i += 2) {
+    const int64_t rowa = r0 + (int64_t)i * 4;
+    const int64_t rowb = r0 + (int64_t)(i + 1) * 4;
+    if (rowa >= rows) return;
+    const bool hb = rowb < rows;
+    const __bf16* xa = x + rowa * H;
+    const __bf16* xb = x + (hb ? rowb : rowa) * H;
+    const uint4 xa0 = *reinterpret_cast<const uint4*>(xa + c0);
+    const uint4 xb0 = *reinterpret_cast<const uint4*>(xb + c0);
+    uint4 xa1 = {}, xb1 = {};
+    if (has1) {
+      xa1 = *reinterpret_cast<const uint4*>(xa + c1);
+      xb1 = *reinterpret_cast<const uint4*>(xb + c1);
+    }
+    float va0[8], va1[8], vb0[8], vb1[8];
+    float sa = 0.f, sqa = 0.f, sb = 0.f, sqb = 0.f;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      v0[j] = bf2f(bf_extract(xv0, j));
-      s += v0[j];
-      sq += v0[j] * v0[j];
+      va0[j] = bf2f(bf_extract(xa0, j));
+      vb0[j] = bf2f(bf_extract(xb0, j));
+      sa += va0[j];
+      sqa += va0[j] * va0[j];
+      sb += vb0[j];
+      sqb += vb0[j] * vb0[j];
     }
     if (has1) {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        v1[j] = bf2f(bf_extract(xv1, j));
-        s += v1[j];
-        sq += v1[j] * v1[j];
+        va1[j] = bf2f(bf_extract(xa1, j));
+        vb1[j] = bf2f(bf_extract(xb1, j));
+        sa += va1[j];
+        sqa += va1[j] * va1[j];
+        sb += vb1[j];
+        sqb += vb1[j] * vb1[j];
       }
     }
 #pragma unroll
     for (int o = 32; o > 0; o >>= 1) {
-      s += __shfl_xor(s, o, 64);
-      sq += __shfl_xor(sq, o, 64);
+      sa += __shfl_xor(sa, o, 64);
+      sqa += __shfl_xor(sqa, o, 64);
+      sb += __shfl_xor(sb, o, 64);
+      sqb += __shfl_xor(sqb, o, 64);
     }
-    const float mu = s / H;
-    const float var = sq / H - mu * mu;
-    const float rs = rsqrtf(var + eps);
+    const float mua = sa / H, mub = sb / H;
+    const float rsa = rsqrtf(sqa / H - mua * mua + eps);
+    const float rsb = rsqrtf(sqb / H - mub * mub + eps);
     if (lane == 0) {
-      mean[row] = mu;
-      rstd[row] = rs;
+      mean[rowa] = mua;
+      rstd[rowa] = rsa;
+      if (hb) {
+        mean[rowb] = mub;
+        rstd[rowb] = rsb;
+      }
     }
-    __bf16 oy[16];
+    __bf16 oa[16], ob[16];
 #pragma unroll
-    for (int j = 0; j < 8; ++j)
-      oy[j] = (__bf16)((v0[j] - mu) * rs * wv0[j] + bv0[j]);
+    for (int j = 0; j < 8; ++j) {
+      oa[j] = (__bf16)((va0[j] - mua) * rsa * wv0[j] + bv0[j]);
+      ob[j] = (__bf16)((vb0[j] - mub) * rsb * wv0[j] + bv0[j]);
+    }
     if (has1) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        oy[8 + j] = (__bf16)((v1[j] - mu) * rs * wv1[j] + bv1[j]);
+      for (int j = 0; j < 8; ++j) {
+        oa[8 + j] = (__bf16)((va1[j] - mua) * rsa * wv1[j] + bv1[j]);
+        ob[8 + j] = (__bf16)((vb1[j] - mub) * rsb * wv1[j] + bv1[j]);
+      }
     }
-    *reinterpret_cast<uint4*>(y + row * H + c0) =
-        *reinterpret_cast<const uint4*>(&oy[0]);
+    *reinterpret_cast<uint4*>(y + rowa * H + c0) =
+        *reinterpret_cast<const uint4*>(&oa[0]);
     if (has1)
-      *reinterpret_cast<uint4*>(y + row * H + c1) =
-          *reinterpret_cast<const uint4*>(&oy[8]);
+      *reinterpret_cast<uint4*>(y + rowa * H + c1) =
+          *reinterpret_cast<const uint4*>(&oa[8]);
+    if (hb) {
+      *reinterpret_cast<uint4*>(y + rowb * H + c0) =
+          *reinterpret_cast<const uint4*>(&ob[0]);
+      if (has1)
+        *reinterpret_cast<uint4*>(y + rowb * H + c1) =
+            *reinterpret_cast<const uint4*>(&ob[8]);
+    }
   }
 }
 
@@ -720,67 +753,110 @@ __global__ __launch_bounds__(256) void k_ln_bwd_bf16_w(
     wv1[j] = has1 ? w[c1 + j] : 0.f;
   }
   const int64_t r0 = (int64_t)blockIdx.x * (4 * BLN_WROWS) + wid;
-  for (int i = 0; i < BLN_WROWS; ++i) {
-    const int64_t row = r0 + (int64_t)i * 4;
-    if (row >= rows) break;
-    const __bf16* xr = x + row * H;
-    const __bf16* dyr = dy + row * H;
-    const float mu = mean[row], rs = rstd[row];
-    const uint4 xv0 = *reinterpret_cast<const uint4*>(xr + c0);
-    const uint4 yv0 = *reinterpret_cast<const uint4*>(dyr + c0);
-    uint4 xv1 = {}, yv1 = {};
+  // ROW PAIRS: interleave the two rows' shfl reduction ladders (ILP 4) —
+  // single-row version measured 2.0 TB/s, shuffle-latency-bound
+  for (int i = 0; i < BLN_WROWS; i += 2) {
+    const int64_t rowa = r0 + (int64_t)i * 4;
+    const int64_t rowb = r0 + (int64_t)(i + 1) * 4;
+    if (rowa >= rows) break;
+    const bool hb = rowb < rows;
+    const __bf16* xra = x + rowa * H;
+    const __bf16* dyra = dy + rowa * H;
+    const __bf16* xrb = x + (hb ? rowb : rowa) * H;
+    const __bf16* dyrb = dy + (hb ? rowb : rowa) * H;
+    const float mua = mean[rowa], rsa = rstd[rowa];
+    const float mub = hb ? mean[rowb] : 0.f, rsb = hb ? rstd[rowb] : 0.f;
+    const uint4 xa0 = *reinterpret_cast<const uint4*>(xra + c0);
+    const uint4 ya0 = *reinterpret_cast<const uint4*>(dyra + c0);
+    const uint4 xb0 = *reinterpret_cast<const uint4*>(xrb + c0);
+    const uint4 yb0 = *reinterpret_cast<const uint4*>(dyrb + c0);
+    uint4 xa1 = {}, ya1 = {}, xb1 = {}, yb1 = {};
     if (has1) {
-      xv1 = *reinterpret_cast<const uint4*>(xr + c1);
-      yv1 = *reinterpret_cast<const uint4*>(dyr + c1);
+      xa1 = *reinterpret_cast<const uint4*>(xra + c1);
+      ya1 = *reinterpret_cast<const uint4*>(dyra + c1);
+      xb1 = *reinterpret_cast<const uint4*>(xrb + c1);
+      yb1 = *reinterpret_cast<const uint4*>(dyrb + c1);
     }
-    float s1 = 0.f, s2 = 0.f;
-    float xh0[8], dy0[8], xh1[8], dy1[8];
+    float s1a = 0.f, s2a = 0.f, s1b = 0.f, s2b = 0.f;
+    float xha0[8], dya0[8], xha1[8], dya1[8];
+    float xhb0[8], dyb0[8], xhb1[8], dyb1[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      xh0[j] = (bf2f(bf_extract(xv0, j)) - mu) * rs;
-      dy0[j] = bf2f(bf_extract(yv0, j));
-      const float dyw = dy0[j] * wv0[j];
-      s1 += dyw * xh0[j];
-      s2 += dyw;
-      accw[0][j] += dy0[j] * xh0[j];
-      accb[0][j] += dy0[j];
+      xha0[j] = (bf2f(bf_extract(xa0, j)) - mua) * rsa;
+      dya0[j] = bf2f(bf_extract(ya0, j));
+      xhb0[j] = (bf2f(bf_extract(xb0, j)) - mub) * rsb;
+      dyb0[j] = bf2f(bf_extract(yb0, j));
+      s1a += dya0[j] * wv0[j] * xha0[j];
+      s2a += dya0[j] * wv0[j];
+      s1b += dyb0[j] * wv0[j] * xhb0[j];
+      s2b += dyb0[j] * wv0[j];
+      accw[0][j] += dya0[j] * xha0[j];
+      accb[0][j] += dya0[j];
+      if (hb) {
+        accw[0][j] += dyb0[j] * xhb0[j];
+        accb[0][j] += dyb0[j];
+      }
     }
     if (has1) {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        xh1[j] = (bf2f(bf_extract(xv1, j)) - mu) * rs;
-        dy1[j] = bf2f(bf_extract(yv1, j));
-        const float dyw = dy1[j] * wv1[j];
-        s1 += dyw * xh1[j];
-        s2 += dyw;
-        accw[1][j] += dy1[j] * xh1[j];
-        accb[1][j] += dy1[j];
+        xha1[j] = (bf2f(bf_extract(xa1, j)) - mua) * rsa;
+        dya1[j] = bf2f(bf_extract(ya1, j));
+        xhb1[j] = (bf2f(bf_extract(xb1, j)) - mub) * rsb;
+        dyb1[j] = bf2f(bf_extract(yb1, j));
+        s1a += dya1[j] * wv1[j] * xha1[j];
+        s2a += dya1[j] * wv1[j];
+        s1b += dyb1[j] * wv1[j] * xhb1[j];
+        s2b += dyb1[j] * wv1[j];
+        accw[1][j] += dya1[j] * xha1[j];
+        accb[1][j] += dya1[j];
+        if (hb) {
+          accw[1][j] += dyb1[j] * xhb1[j];
+          accb[1][j] += dyb1[j];
+        }
       }
     }
 #pragma unroll
     for (int o = 32; o > 0; o >>= 1) {
-      s1 += __shfl_xor(s1, o, 64);
-      s2 += __shfl_xor(s2, o, 64);
+      s1a += __shfl_xor(s1a, o, 64);
+      s2a += __shfl_xor(s2a, o, 64);
+      s1b += __shfl_xor(s1b, o, 64);
+      s2b += __shfl_xor(s2b, o, 64);
     }
-    const float m1 = s1 / H, m2 = s2 / H;
-    __bf16 ox[16];
+    const float m1a = s1a / H, m2a = s2a / H;
+    const float m1b = s1b / H, m2b = s2b / H;
+    __bf16 oxa[16], oxb[16];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      const float v = rs * (dy0[j] * wv0[j] - m2 - xh0[j] * m1);
-      ox[j] = (__bf16)(DX_ACCUM ? bf2f(dx[row * H + c0 + j]) + v : v);
+      const float va = rsa * (dya0[j] * wv0[j] - m2a - xha0[j] * m1a);
+      oxa[j] = (__bf16)(DX_ACCUM ? bf2f(dx[rowa * H + c0 + j]) + va : va);
+      const float vb = rsb * (dyb0[j] * wv0[j] - m2b - xhb0[j] * m1b);
+      oxb[j] =
+          (__bf16)(DX_ACCUM && hb ? bf2f(dx[rowb * H + c0 + j]) + vb : vb);
     }
     if (has1) {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        const float v = rs * (dy1[j] * wv1[j] - m2 - xh1[j] * m1);
-        ox[8 + j] = (__bf16)(DX_ACCUM ? bf2f(dx[row * H + c1 + j]) + v : v);
+        const float va = rsa * (dya1[j] * wv1[j] - m2a - xha1[j] * m1a);
+        oxa[8 + j] =
+            (__bf16)(DX_ACCUM ? bf2f(dx[rowa * H + c1 + j]) + va : va);
+        const float vb = rsb * (dyb1[j] * wv1[j] - m2b - xhb1[j] * m1b);
+        oxb[8 + j] =
+            (__bf16)(DX_ACCUM && hb ? bf2f(dx[rowb * H + c1 + j]) + vb : vb);
       }
     }
-    *reinterpret_cast<uint4*>(dx + row * H + c0) =
-        *reinterpret_cast<const uint4*>(&ox[0]);
+    *reinterpret_cast<uint4*>(dx + rowa * H + c0) =
+        *reinterpret_cast<const uint4*>(&oxa[0]);
     if (has1)
-      *reinterpret_cast<uint4*>(dx + row * H + c1) =
-          *reinterpret_cast<const uint4*>(&ox[8]);
+      *reinterpret_cast<uint4*>(dx + rowa * H + c1) =
+          *reinterpret_cast<const uint4*>(&oxa[8]);
+    if (hb) {
+      *reinterpret_cast<uint4*>(dx + rowb * H + c0) =
+          *reinterpret_cast<const uint4*>(&oxb[0]);
+      if (has1)
+        *reinterpret_cast<uint4*>(dx + rowb * H + c1) =
+            *reinterpret_cast<const uint4*>(&oxb[8]);
+    }
   }
   // dw/db: lane accs -> LDS (per wave) -> wave 0 folds -> one atomic/col
 #pragma unroll
@@ -2010,6 +2086,175 @@ __global__ __launch_bounds__(256, 2) void k_flash_fwd_bf16_pf(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Flash forward v3 — the guide's 8-wave ladder structure
+// (cdna_hip_programming.md §"Fused attention prefill", plain-HIP ladder):
+//   * 8 waves x QBLK=32 q-rows = 256 q-rows per block; KVBLK=64.
+//   * K tiles STAGED IN LDS, double-buffered, shared by all 8 waves (the
+//     ladder's single biggest lever: the per-wave redundant global K
+//     fragment loads become one coalesced 16 B/thread load + ds_write),
+//     XOR-swizzled image (byte ^= (row&7)<<4) for ~conflict-free
+//     ds_read_b128 A-fragments.
+//   * V^T fragments stay GLOBAL (the VT materialization already exists;
+//     8-wave + all q-blocks of a (b,h) reuse them through L2), hoisted to
+//     the tile top.
+//   * two independent S chains (kv sub-tiles) per tile + tree reductions
+//     replace the serial fmax/sum chains (SQ_WAIT_INST_ANY was 51% after
+//     register prefetch alone — dependency-bound).
+//   * one barrier per tile (stage t+1 into the buffer whose readers
+//     finished at the previous barrier).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(512, 2) void k_flash_fwd_bf16_v3(
+    const __bf16* __restrict__ qkv, const __bf16* __restrict__ VT,
+    __bf16* __restrict__ Obase, float* __restrict__ lse, int Sq, int H,
+    int nh, float scale) {
+  __shared__ __bf16 kbuf[2][64 * 64];  // [buffer][row 64][col 64], swizzled
+  __shared__ float bcast[256];
+  const int z = blockIdx.z;
+  const int b = z / nh, h = z % nh;
+  const int64_t qoff = (int64_t)b * Sq * 3 * H + h * 64;
+  const __bf16* Qp = qkv + qoff;
+  const __bf16* Kp = Qp + H;
+  const __bf16* VTp = VT + (int64_t)z * 64 * Sq;
+  __bf16* Op = Obase + (int64_t)b * Sq * H + h * 64;
+  float* lsep = lse + (int64_t)z * Sq;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int il = lane & 31, kh = lane >> 5;
+  const int q0b = blockIdx.x * 256;
+  const int q0 = q0b + w * 32;
+  const int myq = q0 + il;
+
+  // staging map: thread t loads K row (t>>3), 16 B chunk (t&7) of the
+  // tile; LDS write at chunk ^ (row&7) (read side XORs the same key)
+  const int srow = tid >> 3, schunk = tid & 7;
+  const int swz_chunk = schunk ^ (srow & 7);
+  __bf16* swr = &kbuf[0][srow * 64 + swz_chunk * 8];
+  __bf16* swr1 = &kbuf[1][srow * 64 + swz_chunk * 8];
+
+  bf16x8 qf[4];
+#pragma unroll
+  for (int s = 0; s < 4; ++s)
+    qf[s] = *reinterpret_cast<const bf16x8*>(
+        Qp + (int64_t)myq * 3 * H + s * 16 + kh * 8);
+
+  const int ntiles = (q0b + 256) / 64;  // block-uniform (barrier safety)
+  // prologue: stage tile 0, load tile 1 into registers
+  bf16x8 g1;
+  {
+    const bf16x8 g0 = *reinterpret_cast<const bf16x8*>(
+        Kp + (int64_t)srow * 3 * H + schunk * 8);
+    *reinterpret_cast<bf16x8*>(swr) = g0;
+    if (1 < ntiles)
+      g1 = *reinterpret_cast<const bf16x8*>(
+          Kp + (int64_t)(64 + srow) * 3 * H + schunk * 8);
+    __syncthreads();
+  }
+
+  f32x16 o0 = {}, o1 = {};
+  float m = -INFINITY, l = 0.f;
+
+  for (int kvt = 0; kvt < ntiles; ++kvt) {
+    const int kv0 = kvt * 64;
+    // stage tile t+1 (register -> LDS; its readers finished at the
+    // previous barrier), then load tile t+2 into the register
+    if (kvt + 1 < ntiles) {
+      *reinterpret_cast<bf16x8*>((kvt & 1) ? swr : swr1) = g1;
+      if (kvt + 2 < ntiles)
+        g1 = *reinterpret_cast<const bf16x8*>(
+            Kp + (int64_t)(kv0 + 128 + srow) * 3 * H + schunk * 8);
+    }
+    const bool active = kv0 <= q0 + 31;  // wave-uniform causal skip
+    if (active) {
+      // V^T fragments for this tile (global, L2-shared), issued early
+      bf16x8 vt[8];
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        vt[t] = *reinterpret_cast<const bf16x8*>(
+            VTp + (int64_t)il * Sq + kv0 + t * 16 + kh * 8);
+        vt[4 + t] = *reinterpret_cast<const bf16x8*>(
+            VTp + (int64_t)(32 + il) * Sq + kv0 + t * 16 + kh * 8);
+      }
+      // two independent S chains from the swizzled LDS image
+      const __bf16* kb = kbuf[kvt & 1];
+      f32x16 sacc0 = {}, sacc1 = {};
+#pragma unroll
+      for (int s = 0; s < 4; ++s) {
+        const int ch = (s * 2 + kh);
+        const bf16x8 ka = *reinterpret_cast<const bf16x8*>(
+            &kb[il * 64 + (ch ^ (il & 7)) * 8]);
+        const bf16x8 kb2 = *reinterpret_cast<const bf16x8*>(
+            &kb[(32 + il) * 64 + (ch ^ ((32 + il) & 7)) * 8]);
+        sacc0 =
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qf[s], sacc0, 0, 0,
+                                                    0);
+        sacc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kb2, qf[s], sacc1,
+                                                        0, 0, 0);
+      }
+      // masked scale + tree max over the 32 kv entries of this lane
+      float sv[32];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kvr = kv0 + (r & 3) + 8 * (r >> 2) + 4 * kh;
+        sv[r] = (kvr <= myq) ? sacc0[r] * scale : -INFINITY;
+        sv[16 + r] = (kvr + 32 <= myq) ? sacc1[r] * scale : -INFINITY;
+      }
+      float mx[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        mx[j] = fmaxf(fmaxf(sv[j], sv[8 + j]),
+                      fmaxf(sv[16 + j], sv[24 + j]));
+      float mt = fmaxf(fmaxf(fmaxf(mx[0], mx[1]), fmaxf(mx[2], mx[3])),
+                       fmaxf(fmaxf(mx[4], mx[5]), fmaxf(mx[6], mx[7])));
+      mt = fmaxf(mt, __shfl_xor(mt, 32, 64));
+      const float mnew = fmaxf(m, mt);
+      const float af = __expf(m - mnew);
+      m = mnew;
+      float pv[32];
+      float ps[8] = {};
+#pragma unroll
+      for (int r = 0; r < 32; ++r) {
+        pv[r] = (sv[r] == -INFINITY) ? 0.f : __expf(sv[r] - mnew);
+        ps[r & 7] += pv[r];
+      }
+      float psum = ((ps[0] + ps[1]) + (ps[2] + ps[3])) +
+                   ((ps[4] + ps[5]) + (ps[6] + ps[7]));
+      psum += __shfl_xor(psum, 32, 64);
+      l = l * af + psum;
+      if (kh == 0) bcast[w * 32 + il] = af;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const float a = bcast[w * 32 + (r & 3) + 8 * (r >> 2) + 4 * kh];
+        o0[r] *= a;
+        o1[r] *= a;
+      }
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        const bf16x8 pa = bf_dance(pv + t * 8);
+        o0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, vt[t], o0, 0, 0, 0);
+        o1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, vt[4 + t], o1, 0,
+                                                     0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // epilogue (acc rows = q offsets, cols = il = d)
+  if (kh == 0) {
+    bcast[w * 32 + il] = 1.f / l;
+    lsep[myq] = m + __logf(l);
+  }
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = (r & 3) + 8 * (r >> 2) + 4 * kh;
+    const float inv = bcast[w * 32 + row];
+    Op[(int64_t)(q0 + row) * H + il] = (__bf16)(o0[r] * inv);
+    Op[(int64_t)(q0 + row) * H + 32 + il] = (__bf16)(o1[r] * inv);
+  }
+}
+
 __global__ __launch_bounds__(256, 2) void k_flash_fwd_bf16(
     const __bf16* __restrict__ qkv, const __bf16* __restrict__ VT,
     __bf16* __restrict__ Obase, float* __restrict__ lse, int Sq, int H,
@@ -2140,14 +2385,25 @@ extern "C" int ob_flash_fwd_bf16(const void* qkv, const void* VT, void* O,
   if (H / nh != 64) return ob_fail("flash_fwd: head_dim must be 64");
   if (Sq % 128) return ob_fail("flash_fwd: S must be a multiple of 128");
   dim3 grid((unsigned)(Sq / 128), 1, (unsigned)(B * nh));
-  // default: software-pipelined variant; OB_FLASH_PF=0 reverts (A/B)
-  const char* pfe = getenv("OB_FLASH_PF");
-  if (pfe && pfe[0] == '0')
-    k_flash_fwd_bf16<<<grid, 256, 0, S(stream)>>>(
+  // kernel choice via OB_FLASH_FWD: "v3" (default; the 8-wave LDS-staged
+  // ladder, needs Sq % 256 == 0 — other shapes fall back to v1), "pf"
+  // (register-pipelined), "v1" (round 1)
+  const char* fsel = getenv("OB_FLASH_FWD");
+  const char sel = fsel ? fsel[0] : '3';  // "3" v3 / "pf" / anything else v1
+  if (sel == '3' && Sq % 256 == 0) {
+    dim3 grid3((unsigned)(Sq / 256), 1, (unsigned)(B * nh));
+    k_flash_fwd_bf16_v3<<<grid3, 512, 0, S(stream)>>>(
+        (const __bf16*)qkv, (const __bf16*)VT, (__bf16*)O, (float*)lse,
+        (int)Sq, (int)H, (int)nh, scale);
+    OB_LAUNCH_CHECK();
+    return 0;
+  }
+  if (sel == 'p')
+    k_flash_fwd_bf16_pf<<<grid, 256, 0, S(stream)>>>(
         (const __bf16*)qkv, (const __bf16*)VT, (__bf16*)O, (float*)lse,
         (int)Sq, (int)H, (int)nh, scale);
   else
-    k_flash_fwd_bf16_pf<<<grid, 256, 0, S(stream)>>>(
+    k_flash_fwd_bf16<<<grid, 256, 0, S(stream)>>>(
         (const __bf16*)qkv, (const __bf16*)VT, (__bf16*)O, (float*)lse,
         (int)Sq, (int)H, (int)nh, scale);
   OB_LAUNCH_CHECK();

@@ -60,12 +60,13 @@ check(ext.ob_transpose_bf16_b(ptr(qkv) if False else
                               stream()))
 # causal attention fwd flops: 2 matmuls, ~half masked
 fl_fwd = 2 * 2 * B * nh * S * S * hd / 2
-for mode, name in [("1", "pipelined"), ("0", "round1   ")]:
-    os.environ["OB_FLASH_PF"] = mode
+for mode, name in [("3", "v3-ladder"), ("pf", "pipelined"),
+                   ("v1", "round1   ")]:
+    os.environ["OB_FLASH_FWD"] = mode
     ms = timeit(lambda: check(ext.ob_flash_fwd_bf16(
         ptr(qkv), ptr(VT), ptr(O), ptr(lse), B, S, H, nh, scale, stream())))
     print(f"flash_fwd {name} {ms*1e3:8.1f} us  {fl_fwd/ms/1e9:6.0f} TF")
-os.environ.pop("OB_FLASH_PF", None)
+os.environ.pop("OB_FLASH_FWD", None)
 
 check(ext.ob_transpose_bf16_b(ptr(qkv), ptr(QT), S, hd, S * 3 * H, hd,
                               3 * H, B, nh, stream()))
