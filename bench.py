@@ -81,6 +81,8 @@ def train_step(my_pipeline, dp):
     # the reference's _train_step (engine.py:645-649): pipeline.train() ->
     # dp all-reduce -> optimizer step
     my_pipeline.train()
+    if os.environ.get("OB_SKIP_OPT", "0") == "1":  # overlap-debug bisect
+        return
     dp.do_allreduce(my_pipeline)
     my_pipeline.execution.optimizer_step()
     # grads accumulate across microbatches within a step; clear for the next

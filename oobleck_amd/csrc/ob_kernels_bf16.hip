@@ -2109,6 +2109,7 @@ __global__ __launch_bounds__(512, 2) void k_flash_fwd_bf16_v3(
     __bf16* __restrict__ Obase, float* __restrict__ lse, int Sq, int H,
     int nh, float scale) {
   __shared__ __bf16 kbuf[2][64 * 64];  // [buffer][row 64][col 64], swizzled
+  __shared__ __bf16 vbuf[2][64 * 64];  // V^T tile ([d 64][kv 64]), swizzled
   __shared__ float bcast[256];
   const int z = blockIdx.z;
   const int b = z / nh, h = z % nh;
@@ -2133,6 +2134,8 @@ __global__ __launch_bounds__(512, 2) void k_flash_fwd_bf16_v3(
   const int swz_chunk = schunk ^ (srow & 7);
   __bf16* swr = &kbuf[0][srow * 64 + swz_chunk * 8];
   __bf16* swr1 = &kbuf[1][srow * 64 + swz_chunk * 8];
+  __bf16* svr = &vbuf[0][srow * 64 + swz_chunk * 8];
+  __bf16* svr1 = &vbuf[1][srow * 64 + swz_chunk * 8];
 
   bf16x8 qf[4];
 #pragma unroll
@@ -2141,15 +2144,21 @@ __global__ __launch_bounds__(512, 2) void k_flash_fwd_bf16_v3(
         Qp + (int64_t)myq * 3 * H + s * 16 + kh * 8);
 
   const int ntiles = (q0b + 256) / 64;  // block-uniform (barrier safety)
-  // prologue: stage tile 0, load tile 1 into registers
-  bf16x8 g1;
+  // prologue: stage tile 0 (K + V^T), load tile 1 into registers
+  bf16x8 g1, g1v;
   {
     const bf16x8 g0 = *reinterpret_cast<const bf16x8*>(
         Kp + (int64_t)srow * 3 * H + schunk * 8);
+    const bf16x8 g0v = *reinterpret_cast<const bf16x8*>(
+        VTp + (int64_t)srow * Sq + schunk * 8);
     *reinterpret_cast<bf16x8*>(swr) = g0;
-    if (1 < ntiles)
+    *reinterpret_cast<bf16x8*>(svr) = g0v;
+    if (1 < ntiles) {
       g1 = *reinterpret_cast<const bf16x8*>(
           Kp + (int64_t)(64 + srow) * 3 * H + schunk * 8);
+      g1v = *reinterpret_cast<const bf16x8*>(
+          VTp + (int64_t)srow * Sq + 64 + schunk * 8);
+    }
     __syncthreads();
   }
 
@@ -2162,23 +2171,19 @@ __global__ __launch_bounds__(512, 2) void k_flash_fwd_bf16_v3(
     // previous barrier), then load tile t+2 into the register
     if (kvt + 1 < ntiles) {
       *reinterpret_cast<bf16x8*>((kvt & 1) ? swr : swr1) = g1;
-      if (kvt + 2 < ntiles)
+      *reinterpret_cast<bf16x8*>((kvt & 1) ? svr : svr1) = g1v;
+      if (kvt + 2 < ntiles) {
         g1 = *reinterpret_cast<const bf16x8*>(
             Kp + (int64_t)(kv0 + 128 + srow) * 3 * H + schunk * 8);
+        g1v = *reinterpret_cast<const bf16x8*>(
+            VTp + (int64_t)srow * Sq + kv0 + 128 + schunk * 8);
+      }
     }
     const bool active = kv0 <= q0 + 31;  // wave-uniform causal skip
     if (active) {
-      // V^T fragments for this tile (global, L2-shared), issued early
-      bf16x8 vt[8];
-#pragma unroll
-      for (int t = 0; t < 4; ++t) {
-        vt[t] = *reinterpret_cast<const bf16x8*>(
-            VTp + (int64_t)il * Sq + kv0 + t * 16 + kh * 8);
-        vt[4 + t] = *reinterpret_cast<const bf16x8*>(
-            VTp + (int64_t)(32 + il) * Sq + kv0 + t * 16 + kh * 8);
-      }
       // two independent S chains from the swizzled LDS image
       const __bf16* kb = kbuf[kvt & 1];
+      const __bf16* vb = vbuf[kvt & 1];
       f32x16 sacc0 = {}, sacc1 = {};
 #pragma unroll
       for (int s = 0; s < 4; ++s) {
@@ -2212,12 +2217,12 @@ __global__ __launch_bounds__(512, 2) void k_flash_fwd_bf16_v3(
       const float mnew = fmaxf(m, mt);
       const float af = __expf(m - mnew);
       m = mnew;
-      float pv[32];
+      // exponentiate IN PLACE (sv becomes P: -32 VGPR vs a second array)
       float ps[8] = {};
 #pragma unroll
       for (int r = 0; r < 32; ++r) {
-        pv[r] = (sv[r] == -INFINITY) ? 0.f : __expf(sv[r] - mnew);
-        ps[r & 7] += pv[r];
+        sv[r] = (sv[r] == -INFINITY) ? 0.f : __expf(sv[r] - mnew);
+        ps[r & 7] += sv[r];
       }
       float psum = ((ps[0] + ps[1]) + (ps[2] + ps[3])) +
                    ((ps[4] + ps[5]) + (ps[6] + ps[7]));
@@ -2232,10 +2237,14 @@ __global__ __launch_bounds__(512, 2) void k_flash_fwd_bf16_v3(
       }
 #pragma unroll
       for (int t = 0; t < 4; ++t) {
-        const bf16x8 pa = bf_dance(pv + t * 8);
-        o0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, vt[t], o0, 0, 0, 0);
-        o1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, vt[4 + t], o1, 0,
-                                                     0, 0);
+        const bf16x8 pa = bf_dance(sv + t * 8);
+        const int ch = t * 2 + kh;
+        const bf16x8 v0f = *reinterpret_cast<const bf16x8*>(
+            &vb[il * 64 + (ch ^ (il & 7)) * 8]);
+        const bf16x8 v1f = *reinterpret_cast<const bf16x8*>(
+            &vb[(32 + il) * 64 + (ch ^ ((32 + il) & 7)) * 8]);
+        o0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, v0f, o0, 0, 0, 0);
+        o1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, v1f, o1, 0, 0, 0);
       }
     }
     __syncthreads();

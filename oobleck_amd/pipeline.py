@@ -438,6 +438,14 @@ class OobleckPipeline:
         import os
         import sys
         trace = os.environ.get("OB_TRACE_SCHED", "0") == "1"
+        if getattr(self.execution, "_overlap", False):
+            # the optimizer / DP all-reduce / zero_grads of the previous
+            # step ran on the default stream AFTER the epoch join; the
+            # overlap streams must not start this step's compute before
+            # those writes land (param update, grad zero, shadow refresh)
+            cur = torch.cuda.current_stream()
+            self.execution._s_fwd.wait_stream(cur)
+            self.execution._s_bwd.wait_stream(cur)
         for step_cmds in self.train_schedule:
             for cmd in step_cmds:
                 handler = instruction_map.get(type(cmd))
