@@ -73,6 +73,7 @@ class PipelineExecution:
         # inter-layer temps, 2/3: backward-pass temps — disjoint so the
         # dual-stream overlap below can run a forward beside a backward)
         self._tmp: list[torch.Tensor | None] = [None, None, None, None]
+        self._inbufs: dict[int, tuple] = {}  # per-slot device id/label bufs
         self._out_bufs: dict[int, torch.Tensor] = {}
         self._loss_bufs: dict[int, torch.Tensor] = {}
         self._din_bufs: dict[int, torch.Tensor] = {}
@@ -134,14 +135,28 @@ class PipelineExecution:
                    else contextlib.nullcontext())
             ids_cpu, labels_cpu = batch["input_ids"], batch["labels"]
             if self._overlap and dev.type == "cuda":
-                # pageable H2D on a non-default stream makes the HIP
-                # runtime stage the copy SYNCHRONOUSLY against the
-                # stream's backlog (the round-1 pp1-overlap stall);
-                # pin first so the copy is truly async (torch's caching
-                # host allocator keeps the pinned source alive until the
-                # stream consumes it).
+                # (a) pin the host side: a pageable H2D on a non-default
+                # stream is staged SYNCHRONOUSLY against the stream's
+                # backlog; (b) allocate the DEVICE buffers once per slot
+                # and copy_ into them — fresh per-microbatch allocations
+                # on a side stream push the caching allocator into
+                # hipMalloc (device-wide sync) once the queues run deep:
+                # the round-1 "overlap hangs at 16 microbatches" grew
+                # ~quadratically with depth from exactly this.
                 ids_cpu = ids_cpu.pin_memory()
                 labels_cpu = labels_cpu.pin_memory()
+                bufs = self._inbufs.get(buffer_id)
+                if bufs is None:
+                    bufs = (torch.empty(ids_cpu.shape, dtype=ids_cpu.dtype,
+                                        device=dev),
+                            torch.empty(labels_cpu.shape,
+                                        dtype=labels_cpu.dtype, device=dev))
+                    self._inbufs[buffer_id] = bufs
+                with ctx:
+                    bufs[0].copy_(ids_cpu, non_blocking=True)
+                    bufs[1].copy_(labels_cpu, non_blocking=True)
+                self.pipeline.pipe_buffers["inputs"][buffer_id] = bufs
+                return
             with ctx:
                 ids = ids_cpu.to(dev, non_blocking=True)
                 labels = labels_cpu.to(dev, non_blocking=True)
