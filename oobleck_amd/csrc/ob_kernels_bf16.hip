@@ -2214,27 +2214,36 @@ __global__ __launch_bounds__(512, 2) void k_flash_fwd_bf16_v3(
       float mt = fmaxf(fmaxf(fmaxf(mx[0], mx[1]), fmaxf(mx[2], mx[3])),
                        fmaxf(fmaxf(mx[4], mx[5]), fmaxf(mx[6], mx[7])));
       mt = fmaxf(mt, __shfl_xor(mt, 32, 64));
-      const float mnew = fmaxf(m, mt);
-      const float af = __expf(m - mnew);
-      m = mnew;
+      // defer-max (guide T13): skip the O-rescale (and its LDS
+      // round-trip) when no row's max grew by more than 8 — P is then
+      // bounded by e^8, which the f32 accumulators absorb; ~3x max-abs
+      // error vs always-rescaling (documented, within the stated bf16
+      // test tolerances).  Decision BEFORE exponentiation; wave-uniform.
+      const bool rescale = !__all(mt - m <= 8.f);
+      if (rescale) {
+        const float mnew = fmaxf(m, mt);
+        const float af = __expf(m - mnew);
+        m = mnew;
+        if (kh == 0) bcast[w * 32 + il] = af;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const float a = bcast[w * 32 + (r & 3) + 8 * (r >> 2) + 4 * kh];
+          o0[r] *= a;
+          o1[r] *= a;
+        }
+        l *= af;
+      }
       // exponentiate IN PLACE (sv becomes P: -32 VGPR vs a second array)
       float ps[8] = {};
 #pragma unroll
       for (int r = 0; r < 32; ++r) {
-        sv[r] = (sv[r] == -INFINITY) ? 0.f : __expf(sv[r] - mnew);
+        sv[r] = (sv[r] == -INFINITY) ? 0.f : __expf(sv[r] - m);
         ps[r & 7] += sv[r];
       }
       float psum = ((ps[0] + ps[1]) + (ps[2] + ps[3])) +
                    ((ps[4] + ps[5]) + (ps[6] + ps[7]));
       psum += __shfl_xor(psum, 32, 64);
-      l = l * af + psum;
-      if (kh == 0) bcast[w * 32 + il] = af;
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const float a = bcast[w * 32 + (r & 3) + 8 * (r >> 2) + 4 * kh];
-        o0[r] *= a;
-        o1[r] *= a;
-      }
+      l += psum;
 #pragma unroll
       for (int t = 0; t < 4; ++t) {
         const bf16x8 pa = bf_dance(sv + t * 8);

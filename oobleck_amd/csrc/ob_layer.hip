@@ -44,23 +44,38 @@ static Workspace g_ws;
 // enqueued host-side in order, so event reuse across calls is safe.
 struct SideSync {
   hipStream_t stream = nullptr;
-  hipEvent_t e1 = nullptr, e2 = nullptr, e3 = nullptr;  // main -> side
-  hipEvent_t s2 = nullptr, sf = nullptr;                // side -> main
+  // FRESH event per record (ring): re-recording an event that still has
+  // queued waiters serializes those waiters against the future record,
+  // which compounds with queue depth (the pp1-overlap stall); a 128-deep
+  // ring keeps every in-flight record a distinct object.
+  hipEvent_t ring[128] = {};
+  int ri = 0;
   bool ready = false;
 };
 static SideSync g_side;
 
+static hipEvent_t side_evt() {
+  hipEvent_t e = g_side.ring[g_side.ri & 127];
+  g_side.ri++;
+  return e;
+}
+
 static int side_init() {
   if (g_side.ready) return 0;
   OB_HIP(hipStreamCreateWithFlags(&g_side.stream, hipStreamNonBlocking));
-  OB_HIP(hipEventCreateWithFlags(&g_side.e1, hipEventDisableTiming));
-  OB_HIP(hipEventCreateWithFlags(&g_side.e2, hipEventDisableTiming));
-  OB_HIP(hipEventCreateWithFlags(&g_side.e3, hipEventDisableTiming));
-  OB_HIP(hipEventCreateWithFlags(&g_side.s2, hipEventDisableTiming));
-  OB_HIP(hipEventCreateWithFlags(&g_side.sf, hipEventDisableTiming));
+  for (int i = 0; i < 128; i++)
+    OB_HIP(hipEventCreateWithFlags(&g_side.ring[i], hipEventDisableTiming));
   g_side.ready = true;
   return 0;
 }
+
+// record a fresh event on `from` and make `to` wait it
+#define OB_SIDE_FENCE(from, to)                      \
+  do {                                               \
+    hipEvent_t ev_ = side_evt();                     \
+    OB_HIP(hipEventRecord(ev_, (from)));             \
+    OB_HIP(hipStreamWaitEvent((to), ev_, 0));        \
+  } while (0)
 
 // ---------------------------------------------------------------------------
 // in-step profiler (ob_internal.h declares the family ids + OB_PROF macro).
@@ -767,8 +782,7 @@ static int block_backward_bf16(ob_layer* l, int slot, const __bf16* dout,
   // the producer edges and JOINED at the end of the call (the g_ws
   // buffers it reads are reused by the next microbatch's backward).
   void* const side = (void*)g_side.stream;
-  OB_HIP(hipEventRecord(g_side.e1, S(stream)));  // entry state (dout ready)
-  OB_HIP(hipStreamWaitEvent(g_side.stream, g_side.e1, 0));
+  OB_SIDE_FENCE(S(stream), g_side.stream);  // entry state (dout ready)
 
   OB_HIP(hipMemcpyAsync(din, dout, BS * H * sizeof(__bf16),
                         hipMemcpyDeviceToDevice, S(stream)));
@@ -788,8 +802,7 @@ static int block_backward_bf16(ob_layer* l, int slot, const __bf16* dout,
   if (OB_PROF(OB_PF_ELEM, stream,
               ob_gelu_bwd_bf16(u, DY4, DY4, BS * 4 * H, stream)))
     return 1;
-  OB_HIP(hipEventRecord(g_side.e2, S(stream)));  // DY4 post-gelu
-  OB_HIP(hipStreamWaitEvent(g_side.stream, g_side.e2, 0));
+  OB_SIDE_FENCE(S(stream), g_side.stream);  // DY4 post-gelu
   if (OB_PROF(OB_PF_GEMM_DW, side,
               dw_bf16_ws(ln2, H, DY4, 4 * H, BS, g + bp.w_fc, 4 * H, g_ws.s1,
                          g_ws.s2, side)))
@@ -808,8 +821,7 @@ static int block_backward_bf16(ob_layer* l, int slot, const __bf16* dout,
                                     g + bp.ln2_b, BS, H, 1, stream)))
     return 1;
   // ---- attention projection ----
-  OB_HIP(hipEventRecord(g_side.e3, S(stream)));  // din post-ln2-bwd
-  OB_HIP(hipStreamWaitEvent(g_side.stream, g_side.e3, 0));
+  OB_SIDE_FENCE(S(stream), g_side.stream);  // din post-ln2-bwd
   if (OB_PROF(OB_PF_GEMM_DX, stream,
               gemm_bf(0, 1, BS, H, H, 1.f, din, H, 0, 0, sh + l->sh_ap, H, 0,
                       0, DATT, H, 0, 0, 1, 1, nullptr, nullptr, 0, 1,
@@ -824,7 +836,8 @@ static int block_backward_bf16(ob_layer* l, int slot, const __bf16* dout,
     return 1;
   // din is re-updated by the final ln1 backward: the main stream must
   // not reach it before the side finished reading din
-  OB_HIP(hipEventRecord(g_side.s2, g_side.stream));
+  hipEvent_t ev_din_done = side_evt();
+  OB_HIP(hipEventRecord(ev_din_done, g_side.stream));
   // ---- attention core ----
   if (use_flash(l)) {
     const int64_t BSH = BS * H;
@@ -881,8 +894,7 @@ static int block_backward_bf16(ob_layer* l, int slot, const __bf16* dout,
       return 1;
   }
   // ---- QKV projection ----
-  OB_HIP(hipEventRecord(g_side.e1, S(stream)));  // DQKV ready
-  OB_HIP(hipStreamWaitEvent(g_side.stream, g_side.e1, 0));
+  OB_SIDE_FENCE(S(stream), g_side.stream);  // DQKV ready
   if (OB_PROF(OB_PF_ELEM, side,
               ob_colsum_bf16(DQKV, g + bp.b_qkv, BS, 3 * H, side)))
     return 1;
@@ -896,15 +908,14 @@ static int block_backward_bf16(ob_layer* l, int slot, const __bf16* dout,
                       nullptr, nullptr, 0, 1, stream)))
     return 1;
   // ln1 backward accumulates into din: wait for the side's din readers
-  OB_HIP(hipStreamWaitEvent(S(stream), g_side.s2, 0));
+  OB_HIP(hipStreamWaitEvent(S(stream), ev_din_done, 0));
   if (OB_PROF(OB_PF_LN, stream,
               ob_layernorm_bwd_bf16(x, p + bp.ln1_w, st + l->o_mean1,
                                     st + l->o_rstd1, DLN, din, g + bp.ln1_w,
                                     g + bp.ln1_b, BS, H, 1, stream)))
     return 1;
   // join: the next call reuses DY4/DQKV/din workspaces on the main stream
-  OB_HIP(hipEventRecord(g_side.sf, g_side.stream));
-  OB_HIP(hipStreamWaitEvent(S(stream), g_side.sf, 0));
+  OB_SIDE_FENCE(g_side.stream, S(stream));  // join
   return 0;
 }
 
@@ -930,8 +941,7 @@ static int final_backward_bf16(ob_layer* l, int slot, const float* dout,
   // over v_pad rows, store-guard at V (pad rows of dlogits are 0 anyway,
   // but they have no grad slot).
   {
-    OB_HIP(hipEventRecord(g_side.e1, S(stream)));  // dlogits ready
-    OB_HIP(hipStreamWaitEvent(g_side.stream, g_side.e1, 0));
+    OB_SIDE_FENCE(S(stream), g_side.stream);  // dlogits ready
     void* const side = (void*)g_side.stream;
     // direct TN via hipBLASLt: M = V with lda = v_pad skips the pad
     // rows entirely (no store guard, no 826 MB logits transpose)
@@ -970,8 +980,7 @@ static int final_backward_bf16(ob_layer* l, int slot, const float* dout,
                                     BS, H, 0, stream)))
     return 1;
   // join: the next backward reuses s1/s2 and reads g on the main stream
-  OB_HIP(hipEventRecord(g_side.sf, g_side.stream));
-  OB_HIP(hipStreamWaitEvent(S(stream), g_side.sf, 0));
+  OB_SIDE_FENCE(g_side.stream, S(stream));  // join
   return 0;
 }
 

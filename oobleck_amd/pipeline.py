@@ -102,6 +102,16 @@ class PipelineExecution:
             self._s_bwd = torch.cuda.Stream()
             self._ev_fwd: dict[int, torch.cuda.Event] = {}
             self._ev_bwd: dict[int, torch.cuda.Event] = {}
+            # FRESH event per record (64-deep ring): re-recording a
+            # still-waited event serializes against the FUTURE record on
+            # ROCm, which compounded with pipeline depth
+            self._evt_ring = [torch.cuda.Event() for _ in range(64)]
+            self._evt_i = 0
+
+    def _fresh_event(self) -> "torch.cuda.Event":
+        e = self._evt_ring[self._evt_i & 63]
+        self._evt_i += 1
+        return e
 
     @property
     def pipeline(self) -> "OobleckPipeline":
@@ -169,8 +179,9 @@ class PipelineExecution:
                 if ev is not None:  # slot reuse: wait its previous backward
                     self._s_fwd.wait_event(ev)
                 self._forward_impl(buffer_id)
-                fev = self._ev_fwd.setdefault(buffer_id, torch.cuda.Event())
+                fev = self._fresh_event()
                 fev.record(self._s_fwd)
+                self._ev_fwd[buffer_id] = fev
         else:
             self._forward_impl(buffer_id)
 
@@ -214,8 +225,9 @@ class PipelineExecution:
             with torch.cuda.stream(self._s_bwd):
                 self._s_bwd.wait_event(self._ev_fwd[buffer_id])
                 self._backward_impl(buffer_id)
-                bev = self._ev_bwd.setdefault(buffer_id, torch.cuda.Event())
+                bev = self._fresh_event()
                 bev.record(self._s_bwd)
+                self._ev_bwd[buffer_id] = bev
         else:
             self._backward_impl(buffer_id)
 
