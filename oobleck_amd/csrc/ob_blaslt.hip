@@ -30,31 +30,33 @@ struct LtPlan {
   bool ok = false;
 };
 
-hipblasLtHandle_t g_lt = nullptr;
 constexpr size_t kLtWs = 64u << 20;
-// one workspace per stream: concurrent matmuls on different streams
-// (fwd/bwd/side) must not share scratch
-std::map<void*, void*> g_lt_ws_by_stream;
-
-void* lt_ws(void* stream) {
-  auto it = g_lt_ws_by_stream.find(stream);
-  if (it != g_lt_ws_by_stream.end()) return it->second;
-  void* ws = nullptr;
-  if (hipMalloc(&ws, kLtWs) != hipSuccess) return nullptr;
-  g_lt_ws_by_stream.emplace(stream, ws);
-  return ws;
-}
 // key: tA,tB (row-major semantics), M, N, K, ldc, out f32?, beta!=0?,
 // bias epilogue?, A/B f32?
 using Key = std::tuple<int, int, int64_t, int64_t, int64_t, int64_t, int,
                        int, int, int>;
-std::map<Key, LtPlan> g_plans;
 
-int lt_init() {
-  if (g_lt) return 0;
-  if (hipblasLtCreate(&g_lt) != HIPBLAS_STATUS_SUCCESS)
-    return ob_fail("hipblasLtCreate failed");
-  return 0;
+// ONE FULL CONTEXT PER STREAM — handle + workspace + plan cache.
+// Concurrent enqueue of Lt matmuls on different streams through a single
+// shared handle stalls pathologically at queue depth (measured: the pp1
+// fwd/bwd dual-stream bench never finishes a 16-microbatch step with a
+// shared handle, and runs fine with hipBLASLt disabled) — the handle
+// carries internal per-launch state that cross-stream reuse serializes
+// against.  Streams are few (main + fwd/bwd + side) and long-lived.
+struct LtCtx {
+  hipblasLtHandle_t h = nullptr;
+  void* ws = nullptr;
+  std::map<Key, LtPlan> plans;
+};
+std::map<void*, LtCtx> g_lt_by_stream;
+
+LtCtx* lt_ctx(void* stream) {
+  auto it = g_lt_by_stream.find(stream);
+  if (it != g_lt_by_stream.end()) return &it->second;
+  LtCtx ctx;
+  if (hipblasLtCreate(&ctx.h) != HIPBLAS_STATUS_SUCCESS) return nullptr;
+  if (hipMalloc(&ctx.ws, kLtWs) != hipSuccess) return nullptr;
+  return &g_lt_by_stream.emplace(stream, ctx).first->second;
 }
 
 }  // namespace
@@ -70,13 +72,14 @@ static int lt_matmul(int tA, int tB, int64_t M, int64_t N, int64_t K,
                      float alpha, const void* A, int64_t lda, const void* B,
                      int64_t ldb, float beta, void* C, int64_t ldc,
                      int c_f32, const void* bias, int ab_f32, void* stream) {
-  if (lt_init()) return 1;
+  LtCtx* ctx = lt_ctx(stream);
+  if (!ctx) return ob_fail("hipblasLt per-stream context init failed");
   const Key key{tA,     tB,           M,
                 N,      K,            ldc,
                 c_f32,  beta != 0.f,  bias != nullptr,
                 ab_f32};
-  auto it = g_plans.find(key);
-  if (it == g_plans.end()) {
+  auto it = ctx->plans.find(key);
+  if (it == ctx->plans.end()) {
     LtPlan p;
     // column-major swap: A-slot <- B bytes, B-slot <- A bytes
     const hipblasOperation_t opA = tB ? HIPBLAS_OP_T : HIPBLAS_OP_N;
@@ -116,25 +119,23 @@ static int lt_matmul(int tA, int tB, int64_t M, int64_t N, int64_t K,
         pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &ws, sizeof(ws));
     hipblasLtMatmulHeuristicResult_t res[1];
     int nres = 0;
-    hipblasLtMatmulAlgoGetHeuristic(g_lt, p.desc, p.la, p.lb, p.lc, p.lc,
+    hipblasLtMatmulAlgoGetHeuristic(ctx->h, p.desc, p.la, p.lb, p.lc, p.lc,
                                     pref, 1, res, &nres);
     hipblasLtMatmulPreferenceDestroy(pref);
     if (nres > 0 && res[0].state == HIPBLAS_STATUS_SUCCESS) {
       p.algo = res[0].algo;
       p.ok = true;
     }
-    it = g_plans.emplace(key, p).first;
+    it = ctx->plans.emplace(key, p).first;
   }
   const LtPlan& p = it->second;
   if (!p.ok) return -1;
   if (bias)
     hipblasLtMatmulDescSetAttribute(
         p.desc, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &bias, sizeof(bias));
-  void* ws = lt_ws(stream);
-  if (!ws) return ob_fail("hipBLASLt workspace alloc failed");
   const hipblasStatus_t st = hipblasLtMatmul(
-      g_lt, p.desc, &alpha, B, p.la, A, p.lb, &beta, C, p.lc, C, p.lc,
-      &p.algo, ws, kLtWs, reinterpret_cast<hipStream_t>(stream));
+      ctx->h, p.desc, &alpha, B, p.la, A, p.lb, &beta, C, p.lc, C, p.lc,
+      &p.algo, ctx->ws, kLtWs, reinterpret_cast<hipStream_t>(stream));
   if (st != HIPBLAS_STATUS_SUCCESS)
     return ob_fail("hipblasLtMatmul failed (%d)", (int)st);
   return 0;
