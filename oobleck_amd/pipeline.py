@@ -83,16 +83,16 @@ class PipelineExecution:
         # F(m)->B(m) edge are event-fenced; the optimizer/all-reduce epoch
         # re-joins both streams (train()).  Disabled under FSDP (its
         # collectives must keep a single well-ordered stream).
-        # DISABLED pending round-2 debugging: the overlap passes the full
-        # GPU parity suite and completes at 4 microbatches/step, but
-        # hangs (or degrades beyond a 180 s timeout) at the
-        # 16-microbatch bench scale.  Suspects: the pageable H2D copy on
-        # a non-default stream blocking the host per microbatch, and the
-        # slot-reuse event fences at depth.  OB_PP1_OVERLAP=1 +
-        # OB_TRACE_SCHED=1 re-enable for investigation.
+        # DEFAULT ON since round 2: the round-1 "hangs at 16 microbatches"
+        # was hipBLASLt's shared handle under cross-stream enqueue depth
+        # (fixed by per-stream Lt contexts, ob_blaslt.hip) plus pageable
+        # H2D staging and per-microbatch device allocations on a side
+        # stream (fixed by pinned + per-slot buffers here).  Measured:
+        # 246.5 ms/step vs 267.8 without (+8.6%).  OB_PP1_OVERLAP=0
+        # disables.
         import os
         self._overlap = (
-            os.environ.get("OB_PP1_OVERLAP", "0") == "1"
+            os.environ.get("OB_PP1_OVERLAP", "1") == "1"
             and pipeline.device.type == "cuda"
             and pipeline.is_first_stage() and pipeline.is_last_stage()
             and not any(getattr(l, "_sharded", None) is not None

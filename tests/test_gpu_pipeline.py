@@ -146,3 +146,79 @@ def _run_pp2(rank: int, world: int, tmp: str, dtype: str):
 def test_pp2_hip_end_to_end(dtype, tmp_path):
     """pp2 with HIP compute: two ranks on one GPU, full 1F1B schedule."""
     mp.spawn(_run_pp2, args=(2, str(tmp_path), dtype), nprocs=2, join=True)
+
+
+def _run_pp1_overlap(rank: int, world: int, tmp: str):
+    """pp1 with the fwd/bwd dual-stream overlap (default on): the full
+    OobleckPipeline.train() against the oracle — the exact path bench.py
+    times at N=1 (PipelineExecution._overlap engages automatically:
+    cuda + single stage + no FSDP)."""
+    _setup(rank, world, tmp)
+    from oracle.gpt2_oracle import OracleConfig, init_layer_params
+    from oracle.gpt2_oracle import stage_forward_backward
+
+    from oobleck_amd.config import ModelConfig, TrainingConfig
+    from oobleck_amd.engine import DataParallelEngine, make_rank_grid
+    from oobleck_amd.layer import Layer
+    from oobleck_amd.optimizer import FusedAdamW, WarmupLR
+    from oobleck_amd.pipeline import OobleckPipeline
+
+    dev = torch.device("cuda", 0)
+    mc = ModelConfig(**DIMS)
+    oc = OracleConfig(**DIMS)
+    tc = TrainingConfig(microbatch_size=B, global_microbatch_size=B * MB,
+                        seq_len=S, lr=1e-3)
+    L = oc.n_layers_total
+    flats = [init_layer_params(oc, oc.layer_kind(i), 42 * 100 + i)
+             for i in range(L)]
+    grid = make_rank_grid(L, [list(range(L))], [[0]])
+
+    class Loader:
+        def __iter__(self):
+            return iter({"input_ids": i, "labels": l}
+                        for i, l in _batches(oc.vocab_size, MB, seed=31))
+
+    pipe = OobleckPipeline(0, grid, mc, tc, Loader(), MB, dev)
+    pipe.initialize_distributed_fsdp()
+    pipe.initialize_distributed_pipeline()
+
+    def layer_factory(lid, pg, n_slots):
+        layer = Layer(lid, mc, B, S, n_slots, dev, dtype="bf16")
+        layer.flat_param.copy_(flats[lid].to(dev))
+        layer.refresh_weights()
+        return layer
+
+    def optimizer_factory(layers):
+        opt = FusedAdamW(layers, lr=tc.lr)
+        return opt, WarmupLR(opt, 0)
+
+    pipe.initialize_execution(layer_factory, optimizer_factory)
+    assert pipe.execution._overlap, "overlap should engage at pp1 on cuda"
+    dp = DataParallelEngine([pipe])
+    pipe.train()
+    dp.do_allreduce(pipe)
+    torch.cuda.synchronize()
+
+    grads_ref = [torch.zeros_like(f) for f in flats]
+    losses_ref = []
+    for ids, labels in _batches(oc.vocab_size, MB, seed=31):
+        loss, _, gs = stage_forward_backward(oc, flats, list(range(L)), ids,
+                                             labels=labels)
+        losses_ref.append(loss)
+        for g, gi in zip(grads_ref, gs):
+            g += gi
+    total_ref = sum(l.item() for l in losses_ref)
+    got = pipe.execution.total_loss.item()
+    assert abs(got - total_ref) < 2e-2 * abs(total_ref), (got, total_ref)
+    for layer in pipe.execution._layers:
+        gotg = layer.flat_grad.cpu()
+        ref = grads_ref[layer.layer_id]
+        rel = (gotg - ref).norm() / ref.norm().clamp_min(1e-12)
+        assert rel < 8e-2, (layer.layer_id, rel.item())
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@requires_gpu
+def test_pp1_overlap_end_to_end(tmp_path):
+    mp.spawn(_run_pp1_overlap, args=(1, str(tmp_path)), nprocs=1, join=True)
