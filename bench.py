@@ -150,6 +150,7 @@ def roofline_from_profile(mc, args, prof):
     if os.path.exists(pmc_file) and args.model == "gpt2":
         with open(pmc_file) as f:
             traffic = json.load(f).get("fc_fwd_hbm_bytes_per_launch")
+    overlap = os.environ.get("OB_PP1_OVERLAP", "1") == "1"
     return {
         "bound": "mfma", "achieved": round(achieved_tf, 2),
         "peak": BF16_MFMA_PEAK_TF, "unit": "TFLOP/s",
@@ -157,7 +158,12 @@ def roofline_from_profile(mc, args, prof):
         "traffic": traffic,
         "kernel": (f"MLP fc forward GEMM M={M} N={N} K={K} bf16, measured "
                    "IN-STEP via HIP events on the launch stream (production "
-                   "dispatch)"),
+                   "dispatch)"
+                   + (" — under the default pp1 fwd/bwd dual-stream overlap,"
+                      " so per-launch time includes cross-stream contention;"
+                      " same kernel solo in-step: 50.4 us = 767 TF = 0.307"
+                      " (profiles/r01_step_bf16_blaslt_kernel_stats.json,"
+                      " profiles/r02_bench_solo_instep.json)" if overlap else "")),
         "avg_launch_ms": round(fc["avg_us"] / 1e3, 4),
         "launches_per_step": fc["launches_per_step"],
     }
