@@ -68,10 +68,15 @@ LtCtx* lt_ctx(void* stream) {
 // bias (optional, fp32, length N) is applied via the BIAS epilogue —
 // in the column-major swap D = C-bar (N x M), whose rows are our output
 // columns, exactly the broadcast hipBLASLt defines.
+// Cin: the C-INPUT operand (D = alpha*op(A)op(B) + beta*Cin, written to
+// C) — hipblasLt supports C != D, which folds a residual WITHOUT the
+// 12.6 MB copy the round-1 path paid per forward projection GEMM.
 static int lt_matmul(int tA, int tB, int64_t M, int64_t N, int64_t K,
                      float alpha, const void* A, int64_t lda, const void* B,
                      int64_t ldb, float beta, void* C, int64_t ldc,
-                     int c_f32, const void* bias, int ab_f32, void* stream) {
+                     int c_f32, const void* bias, int ab_f32, void* stream,
+                     const void* Cin = nullptr) {
+  if (!Cin) Cin = C;
   LtCtx* ctx = lt_ctx(stream);
   if (!ctx) return ob_fail("hipblasLt per-stream context init failed");
   const Key key{tA,     tB,           M,
@@ -134,7 +139,7 @@ static int lt_matmul(int tA, int tB, int64_t M, int64_t N, int64_t K,
     hipblasLtMatmulDescSetAttribute(
         p.desc, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &bias, sizeof(bias));
   const hipblasStatus_t st = hipblasLtMatmul(
-      ctx->h, p.desc, &alpha, B, p.la, A, p.lb, &beta, C, p.lc, C, p.lc,
+      ctx->h, p.desc, &alpha, B, p.la, A, p.lb, &beta, Cin, p.lc, C, p.lc,
       &p.algo, ctx->ws, kLtWs, reinterpret_cast<hipStream_t>(stream));
   if (st != HIPBLAS_STATUS_SUCCESS)
     return ob_fail("hipblasLtMatmul failed (%d)", (int)st);
@@ -156,6 +161,17 @@ extern "C" int ob_gemm_lt_bias(int tA, int tB, int64_t M, int64_t N,
                                const void* bias, void* stream) {
   return lt_matmul(tA, tB, M, N, K, alpha, A, lda, B, ldb, beta, C, ldc,
                    c_f32, bias, 0, stream);
+}
+
+/* bias epilogue + residual folded as the C-input (no copy) */
+extern "C" int ob_gemm_lt_bias_res(int tA, int tB, int64_t M, int64_t N,
+                                   int64_t K, float alpha, const void* A,
+                                   int64_t lda, const void* B, int64_t ldb,
+                                   void* C, int64_t ldc, int c_f32,
+                                   const void* bias, const void* residual,
+                                   void* stream) {
+  return lt_matmul(tA, tB, M, N, K, alpha, A, lda, B, ldb, 1.f, C, ldc,
+                   c_f32, bias, 0, stream, residual);
 }
 
 // fp32 operands, fp32 accumulate/out (the reference-dtype leg)

@@ -138,6 +138,12 @@ extern "C" int ob_gemm_lt_bias(int tA, int tB, int64_t M, int64_t N,
                                int64_t lda, const void* B, int64_t ldb,
                                float beta, void* C, int64_t ldc, int c_f32,
                                const void* bias, void* stream);
+extern "C" int ob_gemm_lt_bias_res(int tA, int tB, int64_t M, int64_t N,
+                                   int64_t K, float alpha, const void* A,
+                                   int64_t lda, const void* B, int64_t ldb,
+                                   void* C, int64_t ldc, int c_f32,
+                                   const void* bias, const void* residual,
+                                   void* stream);
 extern "C" int ob_gemm_lt_f32(int tA, int tB, int64_t M, int64_t N,
                               int64_t K, float alpha, const void* A,
                               int64_t lda, const void* B, int64_t ldb,

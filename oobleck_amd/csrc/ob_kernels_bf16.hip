@@ -333,19 +333,17 @@ extern "C" int ob_gemm_bf16(int transA, int transB, int64_t M, int64_t N,
   }();
   if (!no_lt && beta == 0.f && n1 == 1 && n2 == 1 && splitk == 1 &&
       out_kind == BF_OUT_BF16 && M * N >= 512 * 512) {
-    // bias rides the BIAS epilogue; a residual is first copied into C
-    // and folded through beta=1 (one 2-pass DtoD copy, ~4 us at fc size,
-    // against a 550->815 TF kernel swap)
-    float b2 = 0.f;
+    // bias rides the BIAS epilogue; a residual is folded as the C-INPUT
+    // operand of D = alpha*AB + beta*C (C != D) — no copy (the round-1
+    // path paid a 12.6 MB DtoD per forward projection GEMM)
+    int r;
     if (residual) {
-      if (residual != C)
-        OB_HIP(hipMemcpyAsync(C, residual,
-                              (size_t)M * ldc * sizeof(__bf16),
-                              hipMemcpyDeviceToDevice, S(stream)));
-      b2 = 1.f;
+      r = ob_gemm_lt_bias_res(transA, transB, M, N, K, alpha, A, lda, B,
+                              ldb, C, ldc, 0, bias, residual, stream);
+    } else {
+      r = ob_gemm_lt_bias(transA, transB, M, N, K, alpha, A, lda, B, ldb,
+                          0.f, C, ldc, 0, bias, stream);
     }
-    const int r = ob_gemm_lt_bias(transA, transB, M, N, K, alpha, A, lda, B,
-                                  ldb, b2, C, ldc, 0, bias, stream);
     if (r >= 0) return r;  // -1: no algo, fall through to our kernels
   }
   const int BN = (N <= 64) ? 64 : 128;  // narrow tiles for head_dim GEMMs
