@@ -2655,6 +2655,143 @@ __device__ __forceinline__ int acc_row(int r, int kh) {
   return (r & 3) + 8 * (r >> 2) + 4 * kh;
 }
 
+// ---------------------------------------------------------------------------
+// dKdV v3: the paired-wave kernel below + LDS staging of the per-q-tile
+// operands (Q, dO tiles [32][64]; Q^T, dO^T tiles [64][32]), shared by
+// the block's 2 wave pairs — one coalesced 16 B/thread load replaces the
+// per-wave global fragment reads.  Two buffers, TWO barriers per q-tile
+// (the S/dP exchange splits reads across a barrier, so the single-barrier
+// fwd-v3 scheme would race stage-writes with B-fragment reads).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_v3(
+    const __bf16* __restrict__ qkv, const __bf16* __restrict__ QT,
+    const __bf16* __restrict__ dOT, const __bf16* __restrict__ dO,
+    const float* __restrict__ lse, const float* __restrict__ D,
+    __bf16* __restrict__ dqkv, int Sq, int H, int nh, float scale) {
+  __shared__ float xch[2][2][2][32 * 32];
+  __shared__ __bf16 qbuf[2][32 * 64];
+  __shared__ __bf16 obuf[2][32 * 64];   // dO tile
+  __shared__ __bf16 qtb[2][64 * 32];    // Q^T tile
+  __shared__ __bf16 otb[2][64 * 32];    // dO^T tile
+  const int z = blockIdx.z;
+  const int b = z / nh, h = z % nh;
+  const int64_t base = (int64_t)b * Sq * 3 * H + h * 64;
+  const __bf16* Qp = qkv + base;
+  const __bf16* Kp = Qp + H;
+  const __bf16* Vp = Qp + 2 * H;
+  const __bf16* dOp = dO + (int64_t)b * Sq * H + h * 64;
+  const __bf16* QTp = QT + (int64_t)z * 64 * Sq;
+  const __bf16* dOTp = dOT + (int64_t)z * 64 * Sq;
+  const float* lsep = lse + (int64_t)z * Sq;
+  const float* Dp = D + (int64_t)z * Sq;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int pair = w >> 1, role = w & 1;
+  const int il = lane & 31, kh = lane >> 5;
+  const int kv0 = blockIdx.x * 64 + pair * 32;
+  const int mykv = kv0 + il;
+
+  const __bf16* KV = role ? Vp : Kp;
+  bf16x8 of[4];
+#pragma unroll
+  for (int s = 0; s < 4; ++s)
+    of[s] = *reinterpret_cast<const bf16x8*>(
+        KV + (int64_t)mykv * 3 * H + s * 16 + kh * 8);
+
+  // staging maps
+  const int srow = tid >> 3, schunk = tid & 7;    // Q/dO: 32 rows x 8 chunks
+  const int swzq = schunk ^ (srow & 7);
+  const int trow = tid >> 2, tchunk = tid & 3;    // QT/dOT: 64 rows x 4 chunks
+  const int swzt = tchunk ^ (trow & 3);
+  auto stage = [&](int buf, int q0) {
+    *reinterpret_cast<bf16x8*>(&qbuf[buf][srow * 64 + swzq * 8]) =
+        *reinterpret_cast<const bf16x8*>(
+            Qp + (int64_t)(q0 + srow) * 3 * H + schunk * 8);
+    *reinterpret_cast<bf16x8*>(&obuf[buf][srow * 64 + swzq * 8]) =
+        *reinterpret_cast<const bf16x8*>(
+            dOp + (int64_t)(q0 + srow) * H + schunk * 8);
+    *reinterpret_cast<bf16x8*>(&qtb[buf][trow * 32 + swzt * 8]) =
+        *reinterpret_cast<const bf16x8*>(
+            QTp + (int64_t)trow * Sq + q0 + tchunk * 8);
+    *reinterpret_cast<bf16x8*>(&otb[buf][trow * 32 + swzt * 8]) =
+        *reinterpret_cast<const bf16x8*>(
+            dOTp + (int64_t)trow * Sq + q0 + tchunk * 8);
+  };
+
+  const int qt0 = (blockIdx.x * 64) / 32;
+  const int nqt = Sq / 32;
+  stage(qt0 & 1, qt0 * 32);
+  __syncthreads();
+
+  f32x16 dvh = {}, dkh = {};
+  for (int qt = qt0; qt < nqt; ++qt) {
+    const int q0 = qt * 32;
+    if (qt + 1 < nqt) stage((qt + 1) & 1, q0 + 32);
+    const float lse_t = lsep[q0 + il];
+    const float d_t = Dp[q0 + il];
+    // own tile: S (role 0, A = Q rows) or dP (role 1, A = dO rows)
+    const __bf16* ab = role ? obuf[qt & 1] : qbuf[qt & 1];
+    f32x16 own = {};
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      const int ch = s * 2 + kh;
+      const bf16x8 af = *reinterpret_cast<const bf16x8*>(
+          &ab[il * 64 + (ch ^ (il & 7)) * 8]);
+      own = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, of[s], own, 0, 0, 0);
+    }
+    float* mine = xch[qt & 1][pair][role];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) mine[acc_row(r, kh) * 32 + il] = own[r];
+    __syncthreads();
+    const float* theirs = xch[qt & 1][pair][role ^ 1];
+    float other[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) other[r] = theirs[acc_row(r, kh) * 32 + il];
+
+    const __bf16* dob = otb[qt & 1];
+    const __bf16* qtbuf = qtb[qt & 1];
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      float pv[8], dsv[8];
+#pragma unroll
+      for (int r8 = 0; r8 < 8; ++r8) {
+        const int r = t * 8 + r8;
+        const int qoff = acc_row(r, kh);
+        const bool ok = q0 + qoff >= mykv;
+        const float sv = role == 0 ? own[r] : other[r];
+        const float dpv = role == 0 ? other[r] : own[r];
+        const float lse_q = __shfl(lse_t, qoff, 64);
+        const float d_q = __shfl(d_t, qoff, 64);
+        const float p = ok ? __expf(sv * scale - lse_q) : 0.f;
+        pv[r8] = p;
+        dsv[r8] = ok ? p * (dpv - d_q) : 0.f;
+      }
+      const bf16x8 pa = bf_dance(pv);
+      const bf16x8 da = bf_dance(dsv);
+      const int row = role * 32 + il;
+      const int ch = t * 2 + kh;
+      const bf16x8 dof = *reinterpret_cast<const bf16x8*>(
+          &dob[row * 32 + ((ch ^ (row & 3)) * 8)]);
+      const bf16x8 qtf = *reinterpret_cast<const bf16x8*>(
+          &qtbuf[row * 32 + ((ch ^ (row & 3)) * 8)]);
+      dvh = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, dof, dvh, 0, 0, 0);
+      dkh = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, qtf, dkh, 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  __bf16* dKp = dqkv + base + H;
+  __bf16* dVp = dqkv + base + 2 * H;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int kv = kv0 + acc_row(r, kh);
+    dVp[(int64_t)kv * 3 * H + role * 32 + il] = (__bf16)dvh[r];
+    dKp[(int64_t)kv * 3 * H + role * 32 + il] = (__bf16)(scale * dkh[r]);
+  }
+}
+
 __global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_p(
     const __bf16* __restrict__ qkv, const __bf16* __restrict__ QT,
     const __bf16* __restrict__ dOT, const __bf16* __restrict__ dO,
@@ -2781,6 +2918,145 @@ __global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_p(
     const int kv = kv0 + acc_row(r, kh);
     dVp[(int64_t)kv * 3 * H + role * 32 + il] = (__bf16)dvh[r];
     dKp[(int64_t)kv * 3 * H + role * 32 + il] = (__bf16)(scale * dkh[r]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// dQ v3: the fwd-v3 ladder structure applied to the dQ loop — K, V and
+// K^T tiles STAGED IN LDS (double-buffered, XOR-swizzled, shared by the
+// block's 4 waves: one coalesced 16 B/thread load replaces each wave's
+// redundant global fragment reads), KVBLK=64 with two independent
+// S/dP chain pairs, tree-free in-lane masking as before.  One barrier
+// per kv tile; stage-writes target the buffer whose readers finished
+// before the previous barrier (the fwd-v3 scheme).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256, 2) void k_flash_bwd_dq_v3(
+    const __bf16* __restrict__ qkv, const __bf16* __restrict__ KT,
+    const __bf16* __restrict__ dO, const float* __restrict__ lse,
+    const float* __restrict__ D, __bf16* __restrict__ dqkv, int Sq, int H,
+    int nh, float scale) {
+  __shared__ __bf16 kbuf[2][64 * 64];
+  __shared__ __bf16 vbuf[2][64 * 64];
+  __shared__ __bf16 tbuf[2][64 * 64];  // K^T tile ([d 64][kv 64])
+  const int z = blockIdx.z;
+  const int b = z / nh, h = z % nh;
+  const int64_t base = (int64_t)b * Sq * 3 * H + h * 64;
+  const __bf16* Qp = qkv + base;
+  const __bf16* Kp = Qp + H;
+  const __bf16* Vp = Qp + 2 * H;
+  const __bf16* dOp = dO + (int64_t)b * Sq * H + h * 64;
+  const __bf16* KTp = KT + (int64_t)z * 64 * Sq;
+  const float* lsep = lse + (int64_t)z * Sq;
+  const float* Dp = D + (int64_t)z * Sq;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int il = lane & 31, kh = lane >> 5;
+  const int q0 = blockIdx.x * 128 + w * 32;
+  const int myq = q0 + il;
+  const float mylse = lsep[myq];
+  const float myD = Dp[myq];
+
+  bf16x8 qf[4], df[4];
+#pragma unroll
+  for (int s = 0; s < 4; ++s) {
+    qf[s] = *reinterpret_cast<const bf16x8*>(
+        Qp + (int64_t)myq * 3 * H + s * 16 + kh * 8);
+    df[s] = *reinterpret_cast<const bf16x8*>(
+        dOp + (int64_t)myq * H + s * 16 + kh * 8);
+  }
+
+  // staging map: 256 threads x 2 rows each (row r and r+32), 16 B chunks
+  const int srow = tid >> 3, schunk = tid & 7;  // rows 0..31
+  const int swz0 = schunk ^ (srow & 7);
+  const int swz1 = schunk ^ ((srow + 32) & 7);
+  auto stage = [&](int buf, int kv0) {
+    // K and V rows kv0+srow / kv0+srow+32; K^T rows d=srow / d=srow+32
+    *reinterpret_cast<bf16x8*>(&kbuf[buf][srow * 64 + swz0 * 8]) =
+        *reinterpret_cast<const bf16x8*>(
+            Kp + (int64_t)(kv0 + srow) * 3 * H + schunk * 8);
+    *reinterpret_cast<bf16x8*>(&kbuf[buf][(srow + 32) * 64 + swz1 * 8]) =
+        *reinterpret_cast<const bf16x8*>(
+            Kp + (int64_t)(kv0 + srow + 32) * 3 * H + schunk * 8);
+    *reinterpret_cast<bf16x8*>(&vbuf[buf][srow * 64 + swz0 * 8]) =
+        *reinterpret_cast<const bf16x8*>(
+            Vp + (int64_t)(kv0 + srow) * 3 * H + schunk * 8);
+    *reinterpret_cast<bf16x8*>(&vbuf[buf][(srow + 32) * 64 + swz1 * 8]) =
+        *reinterpret_cast<const bf16x8*>(
+            Vp + (int64_t)(kv0 + srow + 32) * 3 * H + schunk * 8);
+    *reinterpret_cast<bf16x8*>(&tbuf[buf][srow * 64 + swz0 * 8]) =
+        *reinterpret_cast<const bf16x8*>(
+            KTp + (int64_t)srow * Sq + kv0 + schunk * 8);
+    *reinterpret_cast<bf16x8*>(&tbuf[buf][(srow + 32) * 64 + swz1 * 8]) =
+        *reinterpret_cast<const bf16x8*>(
+            KTp + (int64_t)(srow + 32) * Sq + kv0 + schunk * 8);
+  };
+
+  const int ntiles = (blockIdx.x * 128 + 128) / 64;  // block-uniform
+  stage(0, 0);
+  __syncthreads();
+
+  f32x16 dq0 = {}, dq1 = {};
+  for (int kvt = 0; kvt < ntiles; ++kvt) {
+    const int kv0 = kvt * 64;
+    if (kvt + 1 < ntiles) stage((kvt + 1) & 1, kv0 + 64);
+    const bool active = kv0 <= q0 + 31;  // wave-uniform causal skip
+    if (active) {
+      const __bf16* kb = kbuf[kvt & 1];
+      const __bf16* vb = vbuf[kvt & 1];
+      const __bf16* tb = tbuf[kvt & 1];
+      // two kv sub-tiles: independent S and dP chains
+      f32x16 s0 = {}, s1 = {}, p0 = {}, p1 = {};
+#pragma unroll
+      for (int s = 0; s < 4; ++s) {
+        const int ch = s * 2 + kh;
+        const bf16x8 ka = *reinterpret_cast<const bf16x8*>(
+            &kb[il * 64 + (ch ^ (il & 7)) * 8]);
+        const bf16x8 kb2 = *reinterpret_cast<const bf16x8*>(
+            &kb[(32 + il) * 64 + (ch ^ ((32 + il) & 7)) * 8]);
+        const bf16x8 va = *reinterpret_cast<const bf16x8*>(
+            &vb[il * 64 + (ch ^ (il & 7)) * 8]);
+        const bf16x8 vb2 = *reinterpret_cast<const bf16x8*>(
+            &vb[(32 + il) * 64 + (ch ^ ((32 + il) & 7)) * 8]);
+        s0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qf[s], s0, 0, 0, 0);
+        s1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kb2, qf[s], s1, 0, 0,
+                                                     0);
+        p0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, df[s], p0, 0, 0, 0);
+        p1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vb2, df[s], p1, 0, 0,
+                                                     0);
+      }
+      float dsv[32];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kv = kv0 + (r & 3) + 8 * (r >> 2) + 4 * kh;
+        const bool ok0 = kv <= myq;
+        const bool ok1 = kv + 32 <= myq;
+        const float pa = ok0 ? __expf(s0[r] * scale - mylse) : 0.f;
+        const float pb = ok1 ? __expf(s1[r] * scale - mylse) : 0.f;
+        dsv[r] = ok0 ? pa * (p0[r] - myD) : 0.f;
+        dsv[16 + r] = ok1 ? pb * (p1[r] - myD) : 0.f;
+      }
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        const bf16x8 da = bf_dance(dsv + t * 8);
+        const int ch = t * 2 + kh;
+        const bf16x8 kt0 = *reinterpret_cast<const bf16x8*>(
+            &tb[il * 64 + (ch ^ (il & 7)) * 8]);
+        const bf16x8 kt1 = *reinterpret_cast<const bf16x8*>(
+            &tb[(32 + il) * 64 + (ch ^ ((32 + il) & 7)) * 8]);
+        dq0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, kt0, dq0, 0, 0, 0);
+        dq1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, kt1, dq1, 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+  __bf16* dQp = dqkv + base;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int q = q0 + (r & 3) + 8 * (r >> 2) + 4 * kh;
+    dQp[(int64_t)q * 3 * H + il] = (__bf16)(scale * dq0[r]);
+    dQp[(int64_t)q * 3 * H + 32 + il] = (__bf16)(scale * dq1[r]);
   }
 }
 
@@ -2913,12 +3189,19 @@ extern "C" int ob_flash_bwd_bf16(const void* qkv, const void* QT,
     OB_LAUNCH_CHECK();
     return 0;
   }
-  // default: the paired-wave 4-occupancy kernel (64 kv rows/block);
-  // OB_FLASH_PAIR=0 reverts to the round-1 2-occupancy kernel (read
-  // per call so tools/flash_probe.py can A/B in one process)
+  // dkdv kernel choice: "3" (default) = v3 LDS-staged paired-wave;
+  // "p" = round-2 register-pipelined paired; anything else = round 1.
+  // Read per call so tools/flash_probe.py can A/B in one process.
   const char* pe = getenv("OB_FLASH_PAIR");
-  const bool pair_off = pe && pe[0] == '0';
-  if (!pair_off) {
+  const char dsel = pe ? pe[0] : '3';
+  if (dsel == '3') {
+    dim3 gridp((unsigned)(Sq / 64), 1, (unsigned)(B * nh));
+    k_flash_bwd_dkdv_v3<<<gridp, 256, 0, S(stream)>>>(
+        (const __bf16*)qkv, (const __bf16*)QT, (const __bf16*)dOT,
+        (const __bf16*)dO, (const float*)lse, (const float*)D,
+        (__bf16*)dqkv, (int)Sq, (int)H, (int)nh, scale);
+    OB_LAUNCH_CHECK();
+  } else if (dsel == 'p') {
     dim3 gridp((unsigned)(Sq / 64), 1, (unsigned)(B * nh));
     k_flash_bwd_dkdv_p<<<gridp, 256, 0, S(stream)>>>(
         (const __bf16*)qkv, (const __bf16*)QT, (const __bf16*)dOT,
@@ -2932,10 +3215,19 @@ extern "C" int ob_flash_bwd_bf16(const void* qkv, const void* QT,
         (int)Sq, (int)H, (int)nh, scale);
     OB_LAUNCH_CHECK();
   }
-  k_flash_bwd_dq<<<grid, 256, 0, S(stream)>>>(
-      (const __bf16*)qkv, (const __bf16*)KT, (const __bf16*)dO,
-      (const float*)lse, (const float*)D, (__bf16*)dqkv, (int)Sq, (int)H,
-      (int)nh, scale);
+  // dq kernel choice: v3 LDS-staged ladder by default; OB_FLASH_DQ=1
+  // reverts to the register-pipelined round-2 kernel
+  const char* dqe = getenv("OB_FLASH_DQ");
+  if (dqe && dqe[0] == '1')
+    k_flash_bwd_dq<<<grid, 256, 0, S(stream)>>>(
+        (const __bf16*)qkv, (const __bf16*)KT, (const __bf16*)dO,
+        (const float*)lse, (const float*)D, (__bf16*)dqkv, (int)Sq, (int)H,
+        (int)nh, scale);
+  else
+    k_flash_bwd_dq_v3<<<grid, 256, 0, S(stream)>>>(
+        (const __bf16*)qkv, (const __bf16*)KT, (const __bf16*)dO,
+        (const float*)lse, (const float*)D, (__bf16*)dqkv, (int)Sq, (int)H,
+        (int)nh, scale);
   OB_LAUNCH_CHECK();
   return 0;
 }

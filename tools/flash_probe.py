@@ -79,13 +79,17 @@ check(ext.ob_flash_dsum_bf16(ptr(O), ptr(dO), ptr(D), B, S, H, nh, stream()))
 
 fl_bwd = 4 * 2 * B * nh * S * S * hd / 2 + 3 * 2 * B * nh * S * S * hd / 2
 results = {}
-for mode in ["1", "0"]:
-    os.environ["OB_FLASH_PAIR"] = mode
+for dkdv, dq, name in [("3", "", "v3+v3   "), ("p", "", "pair+v3 "),
+                       ("p", "1", "pair+pf "), ("0", "1", "round1  ")]:
+    os.environ["OB_FLASH_PAIR"] = dkdv
+    if dq:
+        os.environ["OB_FLASH_DQ"] = dq
+    else:
+        os.environ.pop("OB_FLASH_DQ", None)
     ms = timeit(lambda: check(ext.ob_flash_bwd_bf16(
         ptr(qkv), ptr(QT), ptr(KT), ptr(dOT), ptr(dO), ptr(lse), ptr(D),
         ptr(dqkv), B, S, H, nh, scale, stream())))
-    name = "paired(4occ)" if mode == "1" else "round1(2occ)"
     results[name] = ms
     print(f"flash_bwd {name} {ms*1e3:8.1f} us  {fl_bwd/ms/1e9:6.0f} TF "
           f"(dkdv+dq together)")
-print(f"speedup: {results['round1(2occ)']/results['paired(4occ)']:.3f}x")
+print(f"speedup v3 vs round1: {results['round1  ']/results['v3+v3   ']:.3f}x")
