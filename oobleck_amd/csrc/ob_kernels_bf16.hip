@@ -3189,11 +3189,14 @@ extern "C" int ob_flash_bwd_bf16(const void* qkv, const void* QT,
     OB_LAUNCH_CHECK();
     return 0;
   }
-  // dkdv kernel choice: "3" (default) = v3 LDS-staged paired-wave;
-  // "p" = round-2 register-pipelined paired; anything else = round 1.
+  // dkdv kernel choice: "p" (default) = register-pipelined paired-wave
+  // (measured FASTER than the LDS-staged v3: the S/dP exchange already
+  // forces one barrier; v3's second staging barrier costs more than the
+  // shared loads save — 259 vs 302 us with dq-v3, profiles/
+  // r02_flash_probe.log); "3" = LDS-staged v3; anything else = round 1.
   // Read per call so tools/flash_probe.py can A/B in one process.
   const char* pe = getenv("OB_FLASH_PAIR");
-  const char dsel = pe ? pe[0] : '3';
+  const char dsel = pe ? pe[0] : 'p';
   if (dsel == '3') {
     dim3 gridp((unsigned)(Sq / 64), 1, (unsigned)(B * nh));
     k_flash_bwd_dkdv_v3<<<gridp, 256, 0, S(stream)>>>(
