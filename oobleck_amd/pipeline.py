@@ -434,9 +434,17 @@ class OobleckPipeline:
         self.my_stage_index = next(i for i, (rs, lids) in enumerate(stages)
                                    if my_layers[0] in lids)
 
+        import os as _os
+        min_buf = 2
+        if (self.num_stages == 1 and self.device.type == "cuda"
+                and _os.environ.get("OB_PP1_OVERLAP", "1") == "1"):
+            # deeper slot rotation lets the overlap's forward stream run
+            # further ahead of the backward stream (slot reuse is the
+            # F(m+k) -> B(m) fence); costs one extra stash slot per layer
+            min_buf = int(_os.environ.get("OB_PP1_SLOTS", "3"))
         self.train_schedule = OobleckPipelineSchedule(
             micro_batches=self.num_microbatches, stages=self.num_stages,
-            stage_id=self.my_stage_index)
+            stage_id=self.my_stage_index, min_pipe_buffers=min_buf)
         n_slots = self.train_schedule.num_pipe_buffers()
         self.pipe_buffers = {
             "inputs": [None] * n_slots,

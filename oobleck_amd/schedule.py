@@ -41,7 +41,12 @@ class OobleckPipelineSchedule:
     compute command.  Allreduce + optimizer step are deliberately excluded
     (the engine drives them per training step, engine.py:645-649)."""
 
-    def __init__(self, micro_batches: int, stages: int, stage_id: int):
+    def __init__(self, micro_batches: int, stages: int, stage_id: int,
+                 min_pipe_buffers: int = 2):
+        # min_pipe_buffers > 2 deepens the slot rotation (used by the pp1
+        # fwd/bwd overlap so the forward stream can run further ahead);
+        # the command SEQUENCE is unchanged, only the buffer assignment.
+        self.min_pipe_buffers = min_pipe_buffers
         assert 0 <= stage_id < stages
         self.micro_batches = micro_batches
         self.stages = stages
@@ -64,7 +69,7 @@ class OobleckPipelineSchedule:
 
     def num_pipe_buffers(self) -> int:
         buffers = min(self.stages - self.stage_id, self.micro_batches)
-        return max(2, buffers)
+        return max(getattr(self, "min_pipe_buffers", 2), buffers)
 
     def _buffer_idx(self, micro_batch_id: int) -> int:
         assert self._valid_micro_batch(micro_batch_id)
