@@ -438,10 +438,11 @@ class OobleckPipeline:
         min_buf = 2
         if (self.num_stages == 1 and self.device.type == "cuda"
                 and _os.environ.get("OB_PP1_OVERLAP", "1") == "1"):
-            # deeper slot rotation lets the overlap's forward stream run
-            # further ahead of the backward stream (slot reuse is the
-            # F(m+k) -> B(m) fence); costs one extra stash slot per layer
-            min_buf = int(_os.environ.get("OB_PP1_SLOTS", "3"))
+            # deeper slot rotation would let the overlap's forward stream
+            # run further ahead of the backward stream (slot reuse is the
+            # F(m+k) -> B(m) fence); measured flat at this workload
+            # (230.8 / 232.8 / 232.1 ms at 2/3/4 slots), so default 2
+            min_buf = int(_os.environ.get("OB_PP1_SLOTS", "2"))
         self.train_schedule = OobleckPipelineSchedule(
             micro_batches=self.num_microbatches, stages=self.num_stages,
             stage_id=self.my_stage_index, min_pipe_buffers=min_buf)
