@@ -737,7 +737,7 @@ __global__ __launch_bounds__(256) void k_ln_bwd_bf16_w(
     const __bf16* __restrict__ x, const float* __restrict__ w,
     const float* __restrict__ mean, const float* __restrict__ rstd,
     const __bf16* __restrict__ dy, __bf16* __restrict__ dx,
-    float* __restrict__ dwp, float* __restrict__ dbp, int64_t rows, int H) {
+    float* __restrict__ dw, float* __restrict__ db, int64_t rows, int H) {
   __shared__ float redw[4][1024];
   __shared__ float redb[4][1024];
   const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
@@ -856,11 +856,7 @@ __global__ __launch_bounds__(256) void k_ln_bwd_bf16_w(
             *reinterpret_cast<const uint4*>(&oxb[8]);
     }
   }
-  // dw/db: lane accs -> LDS (per wave) -> per-BLOCK partial rows.
-  // (The round-1 tail atomicAdd'ed every column from every block: 256
-  // blocks x 2x768 atomics hammering 3 KB — cross-block serialization
-  // measured as the kernel's fixed cost.  A tiny column-reduce kernel
-  // folds the partials instead.)
+  // dw/db: lane accs -> LDS (per wave) -> wave 0 folds -> one atomic/col
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     redw[wid][c0 + j] = accw[0][j];
@@ -872,26 +868,9 @@ __global__ __launch_bounds__(256) void k_ln_bwd_bf16_w(
   }
   __syncthreads();
   for (int c = threadIdx.x; c < H; c += 256) {
-    dwp[(int64_t)blockIdx.x * H + c] =
-        redw[0][c] + redw[1][c] + redw[2][c] + redw[3][c];
-    dbp[(int64_t)blockIdx.x * H + c] =
-        redb[0][c] + redb[1][c] + redb[2][c] + redb[3][c];
+    atomicAdd(&dw[c], redw[0][c] + redw[1][c] + redw[2][c] + redw[3][c]);
+    atomicAdd(&db[c], redb[0][c] + redb[1][c] + redb[2][c] + redb[3][c]);
   }
-}
-
-// fold the per-block dw/db partials: dw[c] += sum_b dwp[b*H + c]
-__global__ __launch_bounds__(256) void k_ln_colreduce(
-    const float* __restrict__ dwp, const float* __restrict__ dbp,
-    float* __restrict__ dw, float* __restrict__ db, int nb, int H) {
-  const int c = blockIdx.x * 256 + threadIdx.x;
-  if (c >= H) return;
-  float sw = 0.f, sb = 0.f;
-  for (int b = 0; b < nb; ++b) {
-    sw += dwp[(int64_t)b * H + c];
-    sb += dbp[(int64_t)b * H + c];
-  }
-  dw[c] += sw;
-  db[c] += sb;
 }
 
 template <bool DX_ACCUM>
@@ -955,33 +934,16 @@ extern "C" int ob_layernorm_bwd_bf16(const void* x, const void* w,
   if (H % 8) return ob_fail("ln_bwd_bf16: H must be a multiple of 8");
   if (H >= 512 && H <= 1024) {
     const int grid = (int)((rows + 4 * BLN_WROWS - 1) / (4 * BLN_WROWS));
-    // per-block dw/db partial workspace (single driving stream for
-    // backward by ABI contract, so one cached buffer suffices)
-    static float* lnws = nullptr;
-    static int64_t lnws_sz = 0;
-    const int64_t need = 2 * (int64_t)grid * H;
-    if (need > lnws_sz) {
-      if (lnws) OB_HIP(hipFree(lnws));
-      lnws = nullptr;
-      lnws_sz = 0;
-      OB_HIP(hipMalloc(&lnws, need * sizeof(float)));
-      lnws_sz = need;
-    }
-    float* dwp = lnws;
-    float* dbp = lnws + (int64_t)grid * H;
     if (dx_accum)
       k_ln_bwd_bf16_w<true><<<grid, 256, 0, S(stream)>>>(
           (const __bf16*)x, (const float*)w, (const float*)mean,
-          (const float*)rstd, (const __bf16*)dy, (__bf16*)dx, dwp, dbp,
-          rows, (int)H);
+          (const float*)rstd, (const __bf16*)dy, (__bf16*)dx, (float*)dw,
+          (float*)db, rows, (int)H);
     else
       k_ln_bwd_bf16_w<false><<<grid, 256, 0, S(stream)>>>(
           (const __bf16*)x, (const float*)w, (const float*)mean,
-          (const float*)rstd, (const __bf16*)dy, (__bf16*)dx, dwp, dbp,
-          rows, (int)H);
-    OB_LAUNCH_CHECK();
-    k_ln_colreduce<<<(int)((H + 255) / 256), 256, 0, S(stream)>>>(
-        dwp, dbp, (float*)dw, (float*)db, grid, (int)H);
+          (const float*)rstd, (const __bf16*)dy, (__bf16*)dx, (float*)dw,
+          (float*)db, rows, (int)H);
     OB_LAUNCH_CHECK();
     return 0;
   }
