@@ -2799,6 +2799,14 @@ __global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_p(
     __bf16* __restrict__ dqkv, int Sq, int H, int nh, float scale) {
   // [buffer][pair][role][32 rows x 32 cols]
   __shared__ float xch[2][2][2][32 * 32];
+  // hybrid LDS staging (round-2c): the q-tile's Q and dO rows — the
+  // A-fragments, read BEFORE the exchange barrier, so the single-barrier
+  // fwd-v3 buffering scheme applies — staged once per block instead of
+  // per wave pair.  B-fragments (Q^T/dO^T) stay register-pipelined from
+  // global: LDS-staging them too needs a second barrier, which measured
+  // slower (dkdv_v3, profiles/r02_flash_probe.log).
+  __shared__ __bf16 qbuf[2][32 * 64];
+  __shared__ __bf16 obuf[2][32 * 64];
   const int z = blockIdx.z;
   const int b = z / nh, h = z % nh;
   const int64_t base = (int64_t)b * Sq * 3 * H + h * 64;
@@ -2811,12 +2819,23 @@ __global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_p(
   const float* lsep = lse + (int64_t)z * Sq;
   const float* Dp = D + (int64_t)z * Sq;
 
-  const int lane = threadIdx.x & 63;
-  const int w = threadIdx.x >> 6;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
   const int pair = w >> 1, role = w & 1;
   const int il = lane & 31, kh = lane >> 5;
   const int kv0 = blockIdx.x * 64 + pair * 32;
   const int mykv = kv0 + il;
+  const int srow = tid >> 3, schunk = tid & 7;
+  const int swzq = schunk ^ (srow & 7);
+  auto stage = [&](int buf, int q0s) {
+    *reinterpret_cast<bf16x8*>(&qbuf[buf][srow * 64 + swzq * 8]) =
+        *reinterpret_cast<const bf16x8*>(
+            Qp + (int64_t)(q0s + srow) * 3 * H + schunk * 8);
+    *reinterpret_cast<bf16x8*>(&obuf[buf][srow * 64 + swzq * 8]) =
+        *reinterpret_cast<const bf16x8*>(
+            dOp + (int64_t)(q0s + srow) * H + schunk * 8);
+  };
 
   // one resident operand per role (K for S, V for dP)
   const __bf16* KV = role ? Vp : Kp;
@@ -2829,42 +2848,39 @@ __global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_p(
   f32x16 dvh = {}, dkh = {};
   const int nqt = Sq / 32;
   const int qt0 = (blockIdx.x * 64) / 32;
-  // A-operand rows for this wave's role: Q rows (S tile) or dO rows (dP)
-  const __bf16* Arow = role ? dOp : Qp;
-  const int64_t Astride = role ? H : 3 * H;
-  // software pipeline: fragments + lse/D for tile t+1 prefetched during
-  // tile t's compute (the barrier kept the compiler from doing this)
-  bf16x8 fA[4], fB[4];
-#pragma unroll
-  for (int s = 0; s < 4; ++s)
-    fA[s] = *reinterpret_cast<const bf16x8*>(
-        Arow + (int64_t)(qt0 * 32 + il) * Astride + s * 16 + kh * 8);
+  stage(qt0 & 1, qt0 * 32);
+  __syncthreads();
   float lseA = lsep[qt0 * 32 + il], dA = Dp[qt0 * 32 + il];
   float lseB, dB;
 
-  auto tile = [&](int qt, bf16x8 (&fcur)[4], bf16x8 (&fnxt)[4],
-                  float lse_t, float d_t, float& lse_n, float& d_n) {
+  auto tile = [&](int qt, float lse_t, float d_t, float& lse_n,
+                  float& d_n) {
     const int q0 = qt * 32;
+    // stage tile t+1's Q/dO rows (single-barrier scheme: this buffer's
+    // readers finished before the PREVIOUS tile's barrier)
+    if (qt + 1 < nqt) stage((qt + 1) & 1, q0 + 32);
     // early: B-operands for this tile's dv/dk (consumed after the barrier)
     const bf16x8 dof0 = *reinterpret_cast<const bf16x8*>(
         dOTp + (int64_t)(role * 32 + il) * Sq + q0 + kh * 8);
     const bf16x8 qtf0 = *reinterpret_cast<const bf16x8*>(
         QTp + (int64_t)(role * 32 + il) * Sq + q0 + kh * 8);
-    // prefetch tile t+1's A-fragments + lse/D
+    // prefetch tile t+1's lse/D
     if (qt + 1 < nqt) {
-#pragma unroll
-      for (int s = 0; s < 4; ++s)
-        fnxt[s] = *reinterpret_cast<const bf16x8*>(
-            Arow + (int64_t)(q0 + 32 + il) * Astride + s * 16 + kh * 8);
       lse_n = lsep[q0 + 32 + il];
       d_n = Dp[q0 + 32 + il];
     }
-    // own tile: S (role 0) or dP (role 1)
+    // own tile: S (role 0, A = Q rows) or dP (role 1, A = dO rows),
+    // A-fragments from the staged LDS image
+    const __bf16* ab = role ? obuf[qt & 1] : qbuf[qt & 1];
     f32x16 own = {};
 #pragma unroll
-    for (int s = 0; s < 4; ++s)
-      own = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fcur[s], of[s], own, 0,
-                                                    0, 0);
+    for (int s = 0; s < 4; ++s) {
+      const int ch = s * 2 + kh;
+      const bf16x8 af = *reinterpret_cast<const bf16x8*>(
+          &ab[il * 64 + (ch ^ (il & 7)) * 8]);
+      own = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, of[s], own, 0, 0,
+                                                    0);
+    }
     // exchange: write own tile, barrier, read the partner's
     float* mine = xch[qt & 1][pair][role];
 #pragma unroll
@@ -2907,8 +2923,8 @@ __global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_p(
   };
 
   for (int qt = qt0; qt < nqt; qt += 2) {
-    tile(qt, fA, fB, lseA, dA, lseB, dB);
-    if (qt + 1 < nqt) tile(qt + 1, fB, fA, lseB, dB, lseA, dA);
+    tile(qt, lseA, dA, lseB, dB);
+    if (qt + 1 < nqt) tile(qt + 1, lseB, dB, lseA, dA);
   }
 
   __bf16* dKp = dqkv + base + H;
