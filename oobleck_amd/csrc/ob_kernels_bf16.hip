@@ -2864,6 +2864,10 @@ __global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_p(
         dOTp + (int64_t)(role * 32 + il) * Sq + q0 + kh * 8);
     const bf16x8 qtf0 = *reinterpret_cast<const bf16x8*>(
         QTp + (int64_t)(role * 32 + il) * Sq + q0 + kh * 8);
+    const bf16x8 dof1 = *reinterpret_cast<const bf16x8*>(
+        dOTp + (int64_t)(role * 32 + il) * Sq + q0 + 16 + kh * 8);
+    const bf16x8 qtf1 = *reinterpret_cast<const bf16x8*>(
+        QTp + (int64_t)(role * 32 + il) * Sq + q0 + 16 + kh * 8);
     // prefetch tile t+1's lse/D
     if (qt + 1 < nqt) {
       lse_n = lsep[q0 + 32 + il];
@@ -2909,14 +2913,8 @@ __global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_p(
       }
       const bf16x8 pa = bf_dance(pv);
       const bf16x8 da = bf_dance(dsv);
-      const bf16x8 dof =
-          t ? *reinterpret_cast<const bf16x8*>(
-                  dOTp + (int64_t)(role * 32 + il) * Sq + q0 + 16 + kh * 8)
-            : dof0;
-      const bf16x8 qtf =
-          t ? *reinterpret_cast<const bf16x8*>(
-                  QTp + (int64_t)(role * 32 + il) * Sq + q0 + 16 + kh * 8)
-            : qtf0;
+      const bf16x8 dof = t ? dof1 : dof0;
+      const bf16x8 qtf = t ? qtf1 : qtf0;
       dvh = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, dof, dvh, 0, 0, 0);
       dkh = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, qtf, dkh, 0, 0, 0);
     }
