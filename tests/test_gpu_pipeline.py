@@ -222,3 +222,84 @@ def _run_pp1_overlap(rank: int, world: int, tmp: str):
 @requires_gpu
 def test_pp1_overlap_end_to_end(tmp_path):
     mp.spawn(_run_pp1_overlap, args=(1, str(tmp_path)), nprocs=1, join=True)
+
+
+def _run_pp2dp2(rank: int, world: int, tmp: str):
+    """2 stages x 2 DP replicas with HIP compute, four ranks sharing one
+    GPU — the single-GPU analog of BASELINE config[2]'s topology
+    (stages x replicas), composing p2p + per-layer cross-pipeline DP
+    all-reduce + HIP layers + fused AdamW in one run."""
+    _setup(rank, world, tmp)
+    from oracle.gpt2_oracle import OracleConfig, init_layer_params
+    from oracle.gpt2_oracle import stage_forward_backward
+
+    from oobleck_amd.config import ModelConfig, TrainingConfig
+    from oobleck_amd.engine import DataParallelEngine, make_rank_grid
+    from oobleck_amd.layer import Layer
+    from oobleck_amd.optimizer import FusedAdamW, WarmupLR
+    from oobleck_amd.pipeline import OobleckPipeline
+
+    dev = torch.device("cuda", 0)
+    mc = ModelConfig(**DIMS)
+    oc = OracleConfig(**DIMS)
+    tc = TrainingConfig(microbatch_size=B,
+                        global_microbatch_size=B * MB * 2, seq_len=S)
+    L = oc.n_layers_total
+    flats = [init_layer_params(oc, oc.layer_kind(i), 42 * 100 + i)
+             for i in range(L)]
+
+    def loader_for(pid):
+        class Loader:
+            def __iter__(self):
+                return iter({"input_ids": i, "labels": l}
+                            for i, l in _batches(oc.vocab_size, MB,
+                                                 seed=7 + pid))
+        return Loader()
+
+    pipelines, my_pipeline = [], None
+    for pid in range(2):
+        ranks = [pid * 2, pid * 2 + 1]
+        grid = make_rank_grid(L, [[0, 1], [2, 3]], [[ranks[0]], [ranks[1]]])
+        p = OobleckPipeline(pid, grid, mc, tc, loader_for(pid), MB, dev)
+        p.initialize_distributed_fsdp()
+        p.initialize_distributed_pipeline()
+        pipelines.append(p)
+    for p in pipelines:
+        if p.my_pipeline:
+            def layer_factory(lid, pg, n_slots):
+                layer = Layer(lid, mc, B, S, n_slots, dev, dtype="bf16")
+                layer.flat_param.copy_(flats[lid].to(dev))
+                layer.refresh_weights()
+                return layer
+
+            def optimizer_factory(layers):
+                opt = FusedAdamW(layers, lr=tc.lr)
+                return opt, WarmupLR(opt, 0)
+            p.initialize_execution(layer_factory, optimizer_factory)
+            my_pipeline = p
+    dp = DataParallelEngine(pipelines)
+    my_pipeline.train()
+    dp.do_allreduce(my_pipeline)
+    torch.cuda.synchronize()
+
+    # reference: per-layer grads summed over BOTH replicas' microbatches
+    grads_ref = [torch.zeros_like(f) for f in flats]
+    for pid in range(2):
+        for ids, labels in _batches(oc.vocab_size, MB, seed=7 + pid):
+            _, _, gs = stage_forward_backward(oc, flats, list(range(L)),
+                                              ids, labels=labels)
+            for g, gi in zip(grads_ref, gs):
+                g += gi
+    for layer in my_pipeline.execution._layers:
+        got = layer.flat_grad.cpu()
+        ref = grads_ref[layer.layer_id]
+        rel = (got - ref).norm() / ref.norm().clamp_min(1e-12)
+        assert rel < 8e-2, (layer.layer_id, rel.item())
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@requires_gpu
+def test_pp2dp2_hip_end_to_end(tmp_path):
+    """config[2]'s stages-x-replicas composition on one GPU (4 ranks)."""
+    mp.spawn(_run_pp2dp2, args=(4, str(tmp_path)), nprocs=4, join=True)
