@@ -39,6 +39,8 @@ def _setup(rank: int, world: int, tmp: str):
     dist.init_process_group(
         "gloo", init_method=f"file://{tmp}/rdzv", rank=rank, world_size=world)
     torch.cuda.set_device(0)
+    # each rank recomputes the CPU oracle reference: don't oversubscribe
+    torch.set_num_threads(max(2, (os.cpu_count() or 8) // (2 * world)))
 
 
 def _batches(vocab, n, seed):
