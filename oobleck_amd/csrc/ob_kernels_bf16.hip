@@ -2848,10 +2848,33 @@ __global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_p(
   f32x16 dvh = {}, dkh = {};
   const int nqt = Sq / 32;
   const int qt0 = (blockIdx.x * 64) / 32;
+  stage(qt0 & 1, qt0 * 32);
+  __syncthreads();
+  float lseA = lsep[qt0 * 32 + il], dA = Dp[qt0 * 32 + il];
+  float lseB, dB;
 
-  // A-phase: S (role 0) or dP (role 1) MFMAs from the staged LDS image
-  // + write the tile into the pair's exchange buffer
-  auto aphase = [&](int qt, f32x16& own_out) {
+  auto tile = [&](int qt, float lse_t, float d_t, float& lse_n,
+                  float& d_n) {
+    const int q0 = qt * 32;
+    // stage tile t+1's Q/dO rows (single-barrier scheme: this buffer's
+    // readers finished before the PREVIOUS tile's barrier)
+    if (qt + 1 < nqt) stage((qt + 1) & 1, q0 + 32);
+    // early: B-operands for this tile's dv/dk (consumed after the barrier)
+    const bf16x8 dof0 = *reinterpret_cast<const bf16x8*>(
+        dOTp + (int64_t)(role * 32 + il) * Sq + q0 + kh * 8);
+    const bf16x8 qtf0 = *reinterpret_cast<const bf16x8*>(
+        QTp + (int64_t)(role * 32 + il) * Sq + q0 + kh * 8);
+    const bf16x8 dof1 = *reinterpret_cast<const bf16x8*>(
+        dOTp + (int64_t)(role * 32 + il) * Sq + q0 + 16 + kh * 8);
+    const bf16x8 qtf1 = *reinterpret_cast<const bf16x8*>(
+        QTp + (int64_t)(role * 32 + il) * Sq + q0 + 16 + kh * 8);
+    // prefetch tile t+1's lse/D
+    if (qt + 1 < nqt) {
+      lse_n = lsep[q0 + 32 + il];
+      d_n = Dp[q0 + 32 + il];
+    }
+    // own tile: S (role 0, A = Q rows) or dP (role 1, A = dO rows),
+    // A-fragments from the staged LDS image
     const __bf16* ab = role ? obuf[qt & 1] : qbuf[qt & 1];
     f32x16 own = {};
 #pragma unroll
@@ -2862,29 +2885,16 @@ __global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_p(
       own = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, of[s], own, 0, 0,
                                                     0);
     }
+    // exchange: write own tile, barrier, read the partner's
     float* mine = xch[qt & 1][pair][role];
 #pragma unroll
     for (int r = 0; r < 16; ++r) mine[acc_row(r, kh) * 32 + il] = own[r];
-    own_out = own;
-  };
-
-  // B-phase: partner's tile + exp/dance + the dV/dK MFMAs
-  auto bphase = [&](int qt, const f32x16& own) {
-    const int q0 = qt * 32;
-    const bf16x8 dof0 = *reinterpret_cast<const bf16x8*>(
-        dOTp + (int64_t)(role * 32 + il) * Sq + q0 + kh * 8);
-    const bf16x8 qtf0 = *reinterpret_cast<const bf16x8*>(
-        QTp + (int64_t)(role * 32 + il) * Sq + q0 + kh * 8);
-    const bf16x8 dof1 = *reinterpret_cast<const bf16x8*>(
-        dOTp + (int64_t)(role * 32 + il) * Sq + q0 + 16 + kh * 8);
-    const bf16x8 qtf1 = *reinterpret_cast<const bf16x8*>(
-        QTp + (int64_t)(role * 32 + il) * Sq + q0 + 16 + kh * 8);
-    const float lse_t = lsep[q0 + il];
-    const float d_t = Dp[q0 + il];
+    __syncthreads();
     const float* theirs = xch[qt & 1][pair][role ^ 1];
     float other[16];
 #pragma unroll
     for (int r = 0; r < 16; ++r) other[r] = theirs[acc_row(r, kh) * 32 + il];
+
 #pragma unroll
     for (int t = 0; t < 2; ++t) {
       float pv[8], dsv[8];
@@ -2910,24 +2920,9 @@ __global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_p(
     }
   };
 
-  // SOFTWARE-PIPELINED EXCHANGE (one barrier per tile): region R(t) holds
-  // B-phase(t) AND A-phase(t+1) — two independent instruction streams, so
-  // the exp/shuffle/dance chain of tile t interleaves with tile t+1's
-  // MFMAs instead of serializing behind the exchange barrier.
-  stage(qt0 & 1, qt0 * 32);
-  __syncthreads();  // staged(qt0) visible
-  f32x16 ownC, ownN;
-  aphase(qt0, ownC);
-  if (qt0 + 1 < nqt) stage((qt0 + 1) & 1, qt0 * 32 + 32);
-  __syncthreads();  // xch(qt0) + staged(qt0+1) visible
-  for (int qt = qt0; qt < nqt; ++qt) {
-    if (qt + 1 < nqt) {
-      aphase(qt + 1, ownN);
-      if (qt + 2 < nqt) stage((qt + 2) & 1, (qt + 2) * 32);
-    }
-    bphase(qt, ownC);
-    ownC = ownN;
-    __syncthreads();
+  for (int qt = qt0; qt < nqt; qt += 2) {
+    tile(qt, lseA, dA, lseB, dB);
+    if (qt + 1 < nqt) tile(qt + 1, lseB, dB, lseA, dA);
   }
 
   __bf16* dKp = dqkv + base + H;
