@@ -245,15 +245,20 @@ def agent_main(agent_id: int, master_port: int, worker_fn) -> None:
         pipe.send(("dist", dist_info))
 
     # worker -> master port forwarding (any worker may elect itself rank 0
-    # after a reconfiguration, so watch every pipe)
+    # after a reconfiguration, so watch every pipe).  One lock around the
+    # two-part send: concurrent watcher threads must not interleave the
+    # request byte with another message's length-prefixed payload.
+    send_lock = threading.Lock()
+
     def pipe_watcher(pipe):
         try:
             while True:
                 msg = pipe.recv()
                 if msg[0] == "port_out":
-                    sock_send_request_type(sock,
-                                           RequestType.FORWARD_RANK0_PORT)
-                    sock_send(sock, msg[1])
+                    with send_lock:
+                        sock_send_request_type(
+                            sock, RequestType.FORWARD_RANK0_PORT)
+                        sock_send(sock, msg[1])
         except (EOFError, OSError):
             pass
     for pipe, _ in workers:
