@@ -2196,13 +2196,22 @@ __global__ __launch_bounds__(512, 2) void k_flash_fwd_bf16_v3(
         sacc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kb2, qf[s], sacc1,
                                                         0, 0, 0);
       }
-      // masked scale + tree max over the 32 kv entries of this lane
+      // masked scale + tree max over the 32 kv entries of this lane;
+      // interior tiles (kv0+63 < q0 <= every myq) skip the masking
       float sv[32];
+      if (kv0 + 63 < q0) {
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int kvr = kv0 + (r & 3) + 8 * (r >> 2) + 4 * kh;
-        sv[r] = (kvr <= myq) ? sacc0[r] * scale : -INFINITY;
-        sv[16 + r] = (kvr + 32 <= myq) ? sacc1[r] * scale : -INFINITY;
+        for (int r = 0; r < 16; ++r) {
+          sv[r] = sacc0[r] * scale;
+          sv[16 + r] = sacc1[r] * scale;
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int kvr = kv0 + (r & 3) + 8 * (r >> 2) + 4 * kh;
+          sv[r] = (kvr <= myq) ? sacc0[r] * scale : -INFINITY;
+          sv[16 + r] = (kvr + 32 <= myq) ? sacc1[r] * scale : -INFINITY;
+        }
       }
       float mx[8];
 #pragma unroll
@@ -2895,6 +2904,9 @@ __global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_p(
 #pragma unroll
     for (int r = 0; r < 16; ++r) other[r] = theirs[acc_row(r, kh) * 32 + il];
 
+    // interior q-tiles (q0 > every kv row of this wave) need no causal
+    // mask: drop the compare + two selects per element (~80% of tiles)
+    const bool full = q0 >= kv0 + 32;
 #pragma unroll
     for (int t = 0; t < 2; ++t) {
       float pv[8], dsv[8];
@@ -2902,14 +2914,20 @@ __global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_p(
       for (int r8 = 0; r8 < 8; ++r8) {
         const int r = t * 8 + r8;
         const int qoff = acc_row(r, kh);
-        const bool ok = q0 + qoff >= mykv;
         const float sv = role == 0 ? own[r] : other[r];
         const float dpv = role == 0 ? other[r] : own[r];
         const float lse_q = __shfl(lse_t, qoff, 64);
         const float d_q = __shfl(d_t, qoff, 64);
-        const float p = ok ? __expf(sv * scale - lse_q) : 0.f;
-        pv[r8] = p;
-        dsv[r8] = ok ? p * (dpv - d_q) : 0.f;
+        if (full) {
+          const float p = __expf(sv * scale - lse_q);
+          pv[r8] = p;
+          dsv[r8] = p * (dpv - d_q);
+        } else {
+          const bool ok = q0 + qoff >= mykv;
+          const float p = ok ? __expf(sv * scale - lse_q) : 0.f;
+          pv[r8] = p;
+          dsv[r8] = ok ? p * (dpv - d_q) : 0.f;
+        }
       }
       const bf16x8 pa = bf_dance(pv);
       const bf16x8 da = bf_dance(dsv);
