@@ -2904,9 +2904,6 @@ __global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_p(
 #pragma unroll
     for (int r = 0; r < 16; ++r) other[r] = theirs[acc_row(r, kh) * 32 + il];
 
-    // interior q-tiles (q0 > every kv row of this wave) need no causal
-    // mask: drop the compare + two selects per element (~80% of tiles)
-    const bool full = q0 >= kv0 + 32;
 #pragma unroll
     for (int t = 0; t < 2; ++t) {
       float pv[8], dsv[8];
@@ -2914,20 +2911,14 @@ __global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_p(
       for (int r8 = 0; r8 < 8; ++r8) {
         const int r = t * 8 + r8;
         const int qoff = acc_row(r, kh);
+        const bool ok = q0 + qoff >= mykv;
         const float sv = role == 0 ? own[r] : other[r];
         const float dpv = role == 0 ? other[r] : own[r];
         const float lse_q = __shfl(lse_t, qoff, 64);
         const float d_q = __shfl(d_t, qoff, 64);
-        if (full) {
-          const float p = __expf(sv * scale - lse_q);
-          pv[r8] = p;
-          dsv[r8] = p * (dpv - d_q);
-        } else {
-          const bool ok = q0 + qoff >= mykv;
-          const float p = ok ? __expf(sv * scale - lse_q) : 0.f;
-          pv[r8] = p;
-          dsv[r8] = ok ? p * (dpv - d_q) : 0.f;
-        }
+        const float p = ok ? __expf(sv * scale - lse_q) : 0.f;
+        pv[r8] = p;
+        dsv[r8] = ok ? p * (dpv - d_q) : 0.f;
       }
       const bf16x8 pa = bf_dance(pv);
       const bf16x8 da = bf_dance(dsv);
