@@ -36,7 +36,6 @@ import enum
 import multiprocessing
 import pickle
 import socket
-import struct
 import threading
 from dataclasses import dataclass, field
 

@@ -10,7 +10,7 @@
 # content as the RCCL path.
 from __future__ import annotations
 
-import json
+
 import os
 import pathlib
 import sys
