@@ -2102,6 +2102,161 @@ __global__ __launch_bounds__(256, 2) void k_flash_fwd_bf16_pf(
 //   * one barrier per tile (stage t+1 into the buffer whose readers
 //     finished at the previous barrier).
 // ---------------------------------------------------------------------------
+// v4: the v3 structure at 4 waves / 128 q-rows per block — halves the
+// intra-block causal diagonal spread the per-tile barrier serializes on
+// (waves idle while the block's longest diagonal finishes), doubles the
+// block count.  Same LDS staging, shared by 4 waves; each thread stages
+// two 16 B chunks.
+__global__ __launch_bounds__(256, 3) void k_flash_fwd_bf16_v4(
+    const __bf16* __restrict__ qkv, const __bf16* __restrict__ VT,
+    __bf16* __restrict__ Obase, float* __restrict__ lse, int Sq, int H,
+    int nh, float scale) {
+  __shared__ __bf16 kbuf[2][64 * 64];
+  __shared__ __bf16 vbuf[2][64 * 64];
+  __shared__ float bcast[128];
+  const int z = blockIdx.z;
+  const int b = z / nh, h = z % nh;
+  const int64_t qoff = (int64_t)b * Sq * 3 * H + h * 64;
+  const __bf16* Qp = qkv + qoff;
+  const __bf16* Kp = Qp + H;
+  const __bf16* VTp = VT + (int64_t)z * 64 * Sq;
+  __bf16* Op = Obase + (int64_t)b * Sq * H + h * 64;
+  float* lsep = lse + (int64_t)z * Sq;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int il = lane & 31, kh = lane >> 5;
+  const int q0b = blockIdx.x * 128;
+  const int q0 = q0b + w * 32;
+  const int myq = q0 + il;
+
+  // staging: thread covers chunks tid and tid+256 of the 512-slot tile
+  const int sr0 = tid >> 3, sc0 = tid & 7;
+  const int sr1 = (tid + 256) >> 3, sc1 = (tid + 256) & 7;
+  const int sw0 = sc0 ^ (sr0 & 7), sw1 = sc1 ^ (sr1 & 7);
+  auto stage = [&](int buf, int kv0) {
+    *reinterpret_cast<bf16x8*>(&kbuf[buf][sr0 * 64 + sw0 * 8]) =
+        *reinterpret_cast<const bf16x8*>(
+            Kp + (int64_t)(kv0 + sr0) * 3 * H + sc0 * 8);
+    *reinterpret_cast<bf16x8*>(&kbuf[buf][sr1 * 64 + sw1 * 8]) =
+        *reinterpret_cast<const bf16x8*>(
+            Kp + (int64_t)(kv0 + sr1) * 3 * H + sc1 * 8);
+    *reinterpret_cast<bf16x8*>(&vbuf[buf][sr0 * 64 + sw0 * 8]) =
+        *reinterpret_cast<const bf16x8*>(
+            VTp + (int64_t)sr0 * Sq + kv0 + sc0 * 8);
+    *reinterpret_cast<bf16x8*>(&vbuf[buf][sr1 * 64 + sw1 * 8]) =
+        *reinterpret_cast<const bf16x8*>(
+            VTp + (int64_t)sr1 * Sq + kv0 + sc1 * 8);
+  };
+
+  bf16x8 qf[4];
+#pragma unroll
+  for (int s = 0; s < 4; ++s)
+    qf[s] = *reinterpret_cast<const bf16x8*>(
+        Qp + (int64_t)myq * 3 * H + s * 16 + kh * 8);
+
+  const int ntiles = (q0b + 128) / 64;
+  stage(0, 0);
+  __syncthreads();
+
+  f32x16 o0 = {}, o1 = {};
+  float m = -INFINITY, l = 0.f;
+  for (int kvt = 0; kvt < ntiles; ++kvt) {
+    const int kv0 = kvt * 64;
+    if (kvt + 1 < ntiles) stage((kvt + 1) & 1, kv0 + 64);
+    const bool active = kv0 <= q0 + 31;
+    if (active) {
+      const __bf16* kb = kbuf[kvt & 1];
+      const __bf16* vb = vbuf[kvt & 1];
+      f32x16 sacc0 = {}, sacc1 = {};
+#pragma unroll
+      for (int s = 0; s < 4; ++s) {
+        const int ch = (s * 2 + kh);
+        const bf16x8 ka = *reinterpret_cast<const bf16x8*>(
+            &kb[il * 64 + (ch ^ (il & 7)) * 8]);
+        const bf16x8 kb2 = *reinterpret_cast<const bf16x8*>(
+            &kb[(32 + il) * 64 + (ch ^ ((32 + il) & 7)) * 8]);
+        sacc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qf[s], sacc0, 0,
+                                                        0, 0);
+        sacc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kb2, qf[s], sacc1,
+                                                        0, 0, 0);
+      }
+      float sv[32];
+      if (kv0 + 63 < q0) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          sv[r] = sacc0[r] * scale;
+          sv[16 + r] = sacc1[r] * scale;
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int kvr = kv0 + (r & 3) + 8 * (r >> 2) + 4 * kh;
+          sv[r] = (kvr <= myq) ? sacc0[r] * scale : -INFINITY;
+          sv[16 + r] = (kvr + 32 <= myq) ? sacc1[r] * scale : -INFINITY;
+        }
+      }
+      float mx[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        mx[j] = fmaxf(fmaxf(sv[j], sv[8 + j]),
+                      fmaxf(sv[16 + j], sv[24 + j]));
+      float mt = fmaxf(fmaxf(fmaxf(mx[0], mx[1]), fmaxf(mx[2], mx[3])),
+                       fmaxf(fmaxf(mx[4], mx[5]), fmaxf(mx[6], mx[7])));
+      mt = fmaxf(mt, __shfl_xor(mt, 32, 64));
+      const bool rescale = !__all(mt - m <= 8.f);
+      if (rescale) {
+        const float mnew = fmaxf(m, mt);
+        const float af = __expf(m - mnew);
+        m = mnew;
+        if (kh == 0) bcast[w * 32 + il] = af;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const float a = bcast[w * 32 + (r & 3) + 8 * (r >> 2) + 4 * kh];
+          o0[r] *= a;
+          o1[r] *= a;
+        }
+        l *= af;
+      }
+      float ps[8] = {};
+#pragma unroll
+      for (int r = 0; r < 32; ++r) {
+        sv[r] = (sv[r] == -INFINITY) ? 0.f : __expf(sv[r] - m);
+        ps[r & 7] += sv[r];
+      }
+      float psum = ((ps[0] + ps[1]) + (ps[2] + ps[3])) +
+                   ((ps[4] + ps[5]) + (ps[6] + ps[7]));
+      psum += __shfl_xor(psum, 32, 64);
+      l += psum;
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        const bf16x8 pa = bf_dance(sv + t * 8);
+        const int ch = t * 2 + kh;
+        const bf16x8 v0f = *reinterpret_cast<const bf16x8*>(
+            &vb[il * 64 + (ch ^ (il & 7)) * 8]);
+        const bf16x8 v1f = *reinterpret_cast<const bf16x8*>(
+            &vb[(32 + il) * 64 + (ch ^ ((32 + il) & 7)) * 8]);
+        o0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, v0f, o0, 0, 0, 0);
+        o1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, v1f, o1, 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  if (kh == 0) {
+    bcast[w * 32 + il] = 1.f / l;
+    lsep[myq] = m + __logf(l);
+  }
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = (r & 3) + 8 * (r >> 2) + 4 * kh;
+    const float inv = bcast[w * 32 + row];
+    Op[(int64_t)(q0 + row) * H + il] = (__bf16)(o0[r] * inv);
+    Op[(int64_t)(q0 + row) * H + 32 + il] = (__bf16)(o1[r] * inv);
+  }
+}
+
 __global__ __launch_bounds__(512, 2) void k_flash_fwd_bf16_v3(
     const __bf16* __restrict__ qkv, const __bf16* __restrict__ VT,
     __bf16* __restrict__ Obase, float* __restrict__ lse, int Sq, int H,
@@ -2414,7 +2569,15 @@ extern "C" int ob_flash_fwd_bf16(const void* qkv, const void* VT, void* O,
   // ladder, needs Sq % 256 == 0 — other shapes fall back to v1), "pf"
   // (register-pipelined), "v1" (round 1)
   const char* fsel = getenv("OB_FLASH_FWD");
-  const char sel = fsel ? fsel[0] : '3';  // "3" v3 / "pf" / anything else v1
+  const char sel = fsel ? fsel[0] : '3';  // 3/4 = LDS ladder, pf, else v1
+  if (sel == '4' && Sq % 128 == 0) {
+    dim3 grid4((unsigned)(Sq / 128), 1, (unsigned)(B * nh));
+    k_flash_fwd_bf16_v4<<<grid4, 256, 0, S(stream)>>>(
+        (const __bf16*)qkv, (const __bf16*)VT, (__bf16*)O, (float*)lse,
+        (int)Sq, (int)H, (int)nh, scale);
+    OB_LAUNCH_CHECK();
+    return 0;
+  }
   if (sel == '3' && Sq % 256 == 0) {
     dim3 grid3((unsigned)(Sq / 256), 1, (unsigned)(B * nh));
     k_flash_fwd_bf16_v3<<<grid3, 512, 0, S(stream)>>>(

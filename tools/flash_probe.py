@@ -60,8 +60,8 @@ check(ext.ob_transpose_bf16_b(ptr(qkv) if False else
                               stream()))
 # causal attention fwd flops: 2 matmuls, ~half masked
 fl_fwd = 2 * 2 * B * nh * S * S * hd / 2
-for mode, name in [("3", "v3-ladder"), ("pf", "pipelined"),
-                   ("v1", "round1   ")]:
+for mode, name in [("4", "v4-4wave "), ("3", "v3-ladder"),
+                   ("pf", "pipelined"), ("v1", "round1   ")]:
     os.environ["OB_FLASH_FWD"] = mode
     ms = timeit(lambda: check(ext.ob_flash_fwd_bf16(
         ptr(qkv), ptr(VT), ptr(O), ptr(lse), B, S, H, nh, scale, stream())))
