@@ -2967,6 +2967,134 @@ __global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_v3(
   }
 }
 
+// Split-duty paired dkdv (round-2d): the paired kernel below exchanges
+// BOTH tiles (S and dP) and every wave runs the full exp + lse/D fan-out.
+// Here role 0 owns the whole P chain (S -> P -> BOTH d-halves of dV) and
+// role 1 owns the whole dS chain (dP -> dS -> BOTH d-halves of dK):
+// only P crosses the pair (one-way, layout-preserving: lane L element r
+// of role 0 is exactly what lane L element r of role 1 multiplies), role
+// 1 skips the exp + causal mask entirely (masked P is already 0), and
+// each wave runs HALF the lse/D shuffle fan-out.
+__global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_s(
+    const __bf16* __restrict__ qkv, const __bf16* __restrict__ QT,
+    const __bf16* __restrict__ dOT, const __bf16* __restrict__ dO,
+    const float* __restrict__ lse, const float* __restrict__ D,
+    __bf16* __restrict__ dqkv, int Sq, int H, int nh, float scale) {
+  __shared__ float xch[2][2][32 * 32];  // [buffer][pair][P tile]
+  __shared__ __bf16 qbuf[2][32 * 64];
+  __shared__ __bf16 obuf[2][32 * 64];
+  const int z = blockIdx.z;
+  const int b = z / nh, h = z % nh;
+  const int64_t base = (int64_t)b * Sq * 3 * H + h * 64;
+  const __bf16* Qp = qkv + base;
+  const __bf16* Kp = Qp + H;
+  const __bf16* Vp = Qp + 2 * H;
+  const __bf16* dOp = dO + (int64_t)b * Sq * H + h * 64;
+  const __bf16* QTp = QT + (int64_t)z * 64 * Sq;
+  const __bf16* dOTp = dOT + (int64_t)z * 64 * Sq;
+  const float* lsep = lse + (int64_t)z * Sq;
+  const float* Dp = D + (int64_t)z * Sq;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int pair = w >> 1, role = w & 1;
+  const int il = lane & 31, kh = lane >> 5;
+  const int kv0 = blockIdx.x * 64 + pair * 32;
+  const int mykv = kv0 + il;
+
+  const __bf16* KV = role ? Vp : Kp;
+  bf16x8 of[4];
+#pragma unroll
+  for (int s = 0; s < 4; ++s)
+    of[s] = *reinterpret_cast<const bf16x8*>(
+        KV + (int64_t)mykv * 3 * H + s * 16 + kh * 8);
+
+  const int srow = tid >> 3, schunk = tid & 7;
+  const int swzq = schunk ^ (srow & 7);
+  auto stage = [&](int buf, int q0s) {
+    *reinterpret_cast<bf16x8*>(&qbuf[buf][srow * 64 + swzq * 8]) =
+        *reinterpret_cast<const bf16x8*>(
+            Qp + (int64_t)(q0s + srow) * 3 * H + schunk * 8);
+    *reinterpret_cast<bf16x8*>(&obuf[buf][srow * 64 + swzq * 8]) =
+        *reinterpret_cast<const bf16x8*>(
+            dOp + (int64_t)(q0s + srow) * H + schunk * 8);
+  };
+
+  f32x16 acc0 = {}, acc1 = {};  // role 0: dV d-halves; role 1: dK d-halves
+  const int nqt = Sq / 32;
+  const int qt0 = (blockIdx.x * 64) / 32;
+  stage(qt0 & 1, qt0 * 32);
+  __syncthreads();
+
+  for (int qt = qt0; qt < nqt; ++qt) {
+    const int q0 = qt * 32;
+    if (qt + 1 < nqt) stage((qt + 1) & 1, q0 + 32);
+    // B-operands early: BOTH d-halves of this role's output operand
+    const __bf16* Bp = role ? QTp : dOTp;
+    const bf16x8 b00 = *reinterpret_cast<const bf16x8*>(
+        Bp + (int64_t)il * Sq + q0 + kh * 8);
+    const bf16x8 b01 = *reinterpret_cast<const bf16x8*>(
+        Bp + (int64_t)(32 + il) * Sq + q0 + kh * 8);
+    const bf16x8 b10 = *reinterpret_cast<const bf16x8*>(
+        Bp + (int64_t)il * Sq + q0 + 16 + kh * 8);
+    const bf16x8 b11 = *reinterpret_cast<const bf16x8*>(
+        Bp + (int64_t)(32 + il) * Sq + q0 + 16 + kh * 8);
+    // own tile: S (role 0, A = Q rows) or dP (role 1, A = dO rows)
+    const __bf16* ab = role ? obuf[qt & 1] : qbuf[qt & 1];
+    f32x16 own = {};
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      const int ch = s * 2 + kh;
+      const bf16x8 af = *reinterpret_cast<const bf16x8*>(
+          &ab[il * 64 + (ch ^ (il & 7)) * 8]);
+      own = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, of[s], own, 0, 0,
+                                                    0);
+    }
+    float pv[16];
+    float* ptile = xch[qt & 1][pair];
+    if (role == 0) {
+      const float lse_t = lsep[q0 + il];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qoff = acc_row(r, kh);
+        const bool ok = q0 + qoff >= mykv;
+        const float lse_q = __shfl(lse_t, qoff, 64);
+        pv[r] = ok ? __expf(own[r] * scale - lse_q) : 0.f;
+        ptile[qoff * 32 + il] = pv[r];
+      }
+    }
+    __syncthreads();
+    if (role == 1) {
+      const float d_t = Dp[q0 + il];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qoff = acc_row(r, kh);
+        const float d_q = __shfl(d_t, qoff, 64);
+        // dS = P * (dP - D); masked entries have P == 0 already
+        pv[r] = ptile[qoff * 32 + il] * (own[r] - d_q);
+      }
+    }
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      const bf16x8 a = bf_dance(pv + t * 8);
+      const bf16x8 bl = t ? b10 : b00;
+      const bf16x8 bh = t ? b11 : b01;
+      acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, bl, acc0, 0, 0, 0);
+      acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, bh, acc1, 0, 0, 0);
+    }
+  }
+
+  __bf16* outp = dqkv + base + (role ? H : 2 * H);  // dK or dV slice
+  const float os = role ? scale : 1.f;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int kv = kv0 + acc_row(r, kh);
+    outp[(int64_t)kv * 3 * H + il] = (__bf16)(os * acc0[r]);
+    outp[(int64_t)kv * 3 * H + 32 + il] = (__bf16)(os * acc1[r]);
+  }
+}
+
 __global__ __launch_bounds__(256, 3) void k_flash_bwd_dkdv_p(
     const __bf16* __restrict__ qkv, const __bf16* __restrict__ QT,
     const __bf16* __restrict__ dOT, const __bf16* __restrict__ dO,
@@ -3385,7 +3513,15 @@ extern "C" int ob_flash_bwd_bf16(const void* qkv, const void* QT,
   // r02_flash_probe.log); "3" = LDS-staged v3; anything else = round 1.
   // Read per call so tools/flash_probe.py can A/B in one process.
   const char* pe = getenv("OB_FLASH_PAIR");
-  const char dsel = pe ? pe[0] : 'p';
+  const char dsel = pe ? pe[0] : 'p';  // "s" = split-duty variant
+  if (dsel == 's') {
+    dim3 gridp((unsigned)(Sq / 64), 1, (unsigned)(B * nh));
+    k_flash_bwd_dkdv_s<<<gridp, 256, 0, S(stream)>>>(
+        (const __bf16*)qkv, (const __bf16*)QT, (const __bf16*)dOT,
+        (const __bf16*)dO, (const float*)lse, (const float*)D,
+        (__bf16*)dqkv, (int)Sq, (int)H, (int)nh, scale);
+    OB_LAUNCH_CHECK();
+  } else
   if (dsel == '3') {
     dim3 gridp((unsigned)(Sq / 64), 1, (unsigned)(B * nh));
     k_flash_bwd_dkdv_v3<<<gridp, 256, 0, S(stream)>>>(

@@ -79,8 +79,9 @@ check(ext.ob_flash_dsum_bf16(ptr(O), ptr(dO), ptr(D), B, S, H, nh, stream()))
 
 fl_bwd = 4 * 2 * B * nh * S * S * hd / 2 + 3 * 2 * B * nh * S * S * hd / 2
 results = {}
-for dkdv, dq, name in [("3", "", "v3+v3   "), ("p", "", "pair+v3 "),
-                       ("p", "1", "pair+pf "), ("0", "1", "round1  ")]:
+for dkdv, dq, name in [("s", "", "split+v3"), ("3", "", "v3+v3   "),
+                       ("p", "", "pair+v3 "), ("p", "1", "pair+pf "),
+                       ("0", "1", "round1  ")]:
     os.environ["OB_FLASH_PAIR"] = dkdv
     if dq:
         os.environ["OB_FLASH_DQ"] = dq
