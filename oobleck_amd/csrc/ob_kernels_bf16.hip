@@ -3513,7 +3513,9 @@ extern "C" int ob_flash_bwd_bf16(const void* qkv, const void* QT,
   // r02_flash_probe.log); "3" = LDS-staged v3; anything else = round 1.
   // Read per call so tools/flash_probe.py can A/B in one process.
   const char* pe = getenv("OB_FLASH_PAIR");
-  const char dsel = pe ? pe[0] : 'p';  // "s" = split-duty variant
+  // default "s": the split-duty pair (208.6 us whole-backward vs 231.2
+  // for the two-way exchange, profiles/r02_flash_probe.log)
+  const char dsel = pe ? pe[0] : 's';
   if (dsel == 's') {
     dim3 gridp((unsigned)(Sq / 64), 1, (unsigned)(B * nh));
     k_flash_bwd_dkdv_s<<<gridp, 256, 0, S(stream)>>>(
