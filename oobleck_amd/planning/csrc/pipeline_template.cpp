@@ -216,6 +216,9 @@ class PipelineTemplate {
   int num_gpus_per_node_;
 };
 
+// NOTE: like the reference's dc_cache_ (execution_result.h:213), the memo
+// table is keyed by (num_stages, layer range, num_nodes, num_gpus) WITHOUT
+// the profile's identity — a generator instance serves ONE profile.
 class PipelineTemplateGenerator {
  public:
   std::vector<PipelineTemplate> create_pipeline_templates(

@@ -116,6 +116,49 @@ def test_rank_grid_semantics(pt, profile):
                        for i in range(s._num_gpus))
 
 
+def test_iteration_time_model_numeric(pt):
+    """Numeric pin of the 1F1B latency model (execution_result.h:114-204):
+    for a 2-layer profile split into 2 single-GPU stages,
+      t1 = (f1+b1) + (f2+b2)
+      kstar = index of the slower stage (left wins ties)
+      nk = 2*(1+1) + kstar + 1
+      t2 = nk * (f+b)[kstar]
+      t3 = sum of (f+b) from kstar to the end (both stages when kstar=0)
+    and iteration_time = t1 + t2 + t3."""
+    mk = lambda i, f, bwd: pt.LayerExecutionResult(
+        layer_index=i, forward=f, backward=bwd,
+        allreduce_in_node={j + 1: 0.0 for j in range(8)},
+        allreduce_across_nodes={j + 1: 0.0 for j in range(8)},
+        mem_required=(1, 1))
+    # layer costs: L0 f=2,b=4 (fb=6); L1 f=1,b=2 (fb=3)
+    lers = pt.LayerExecutionResults([mk(0, 2.0, 4.0), mk(1, 1.0, 2.0)])
+    gen = pt.PipelineTemplateGenerator()
+    ts = gen.create_pipeline_templates(lers, (2, 2), 1)
+    assert len(ts) == 1
+    stages = ts[0].get_stages()
+    assert [s._layer_indices for s in stages] == [[0], [1]]
+    # left stage fb=6 > right fb=3 -> kstar=0
+    t1 = 6.0 + 3.0
+    nk = 2 * (1 + 1) + 0 + 1
+    t2 = nk * 6.0
+    t3 = 6.0 + 3.0
+    assert ts[0]._iteration_time == pytest.approx(t1 + t2 + t3)
+
+    # kstar on the RIGHT: swap the weights; right index shifts by left
+    # size.  NOTE: a generator's memo cache is keyed by (stages, layer
+    # range, nodes, gpus) WITHOUT profile identity — exactly like the
+    # reference's dc_cache_ (execution_result.h:213) — so one generator
+    # serves ONE profile; use a fresh one here.
+    lers2 = pt.LayerExecutionResults([mk(0, 1.0, 2.0), mk(1, 2.0, 4.0)])
+    gen2 = pt.PipelineTemplateGenerator()
+    ts2 = gen2.create_pipeline_templates(lers2, (2, 2), 1)
+    t1 = 3.0 + 6.0
+    nk = 2 * 2 + 1 + 1  # kstar = 1
+    t2 = nk * 6.0
+    t3 = 6.0  # only from kstar to the end of the right side
+    assert ts2[0]._iteration_time == pytest.approx(t1 + t2 + t3)
+
+
 def test_get_profile_results_json(pt, tmp_path, monkeypatch):
     """get_profile_results loads the profiler's JSON cache layout
     (pipeline_template.cpp:29-80; profiler.py:290-319 writes it)."""
